@@ -1,0 +1,11 @@
+from sparkflow_amd.models.mlp import ClassifierMLP, Autoencoder, RegressorMLP  # noqa: F401
+from sparkflow_amd.models.cnn import MNISTCNN, ResNet18Shape  # noqa: F401
+from sparkflow_amd.models.zoo import (  # noqa: F401
+    mnist_mlp,
+    mnist_cnn,
+    mnist_autoencoder,
+    tabular_autoencoder,
+    resnet18_shape,
+    xor_mlp,
+    gaussian_mlp,
+)

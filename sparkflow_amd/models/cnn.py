@@ -1,0 +1,106 @@
+"""CNN model families.
+
+MNISTCNN mirrors reference examples/cnn_example.py:13-21 (5x5x32 conv ->
+2x2 pool -> 3x3x64 conv -> 2x2 pool -> flatten -> dense 10).
+
+ResNet18Shape is the BASELINE.json config-5 model: a ResNet-18-shaped CNN
+for synthetic 3x224x224 inputs (large-batch PS stress).  Convs/GEMMs run on
+the CDNA4 HIP kernels; batch-norm uses torch-ROCm natives.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from sparkflow_amd import nn as snn
+from sparkflow_amd.ops import functional as F
+
+
+class MNISTCNN(nn.Module):
+    """Input: flat [B, 784] (Spark vector rows), reshaped to NCHW."""
+
+    def __init__(self, num_classes: int = 10):
+        super().__init__()
+        self.conv1 = snn.Conv2d(1, 32, 5, padding="same", activation="relu")
+        self.pool1 = snn.MaxPool2d(2, 2)
+        self.conv2 = snn.Conv2d(32, 64, 3, padding="same", activation="relu")
+        self.pool2 = snn.MaxPool2d(2, 2)
+        self.fc = snn.Linear(7 * 7 * 64, num_classes)
+        self.loss_fn = snn.SoftmaxCrossEntropy()
+
+    def forward(self, x):
+        if x.dim() == 2:
+            x = x.reshape(-1, 1, 28, 28)
+        x = self.pool1(self.conv1(x))
+        x = self.pool2(self.conv2(x))
+        x = x.reshape(x.shape[0], -1)
+        return self.fc(x)
+
+    def compute_loss(self, out, target):
+        return self.loss_fn(out, target)
+
+    @torch.no_grad()
+    def predict(self, x):
+        return F.argmax_rows(self.forward(x))
+
+
+class _BasicBlock(nn.Module):
+    def __init__(self, cin, cout, stride):
+        super().__init__()
+        self.conv1 = snn.Conv2d(cin, cout, 3, stride=stride,
+                                padding=1, activation="none", bias=False)
+        self.bn1 = nn.BatchNorm2d(cout)
+        self.conv2 = snn.Conv2d(cout, cout, 3, stride=1, padding=1,
+                                activation="none", bias=False)
+        self.bn2 = nn.BatchNorm2d(cout)
+        if stride != 1 or cin != cout:
+            self.down_conv = snn.Conv2d(cin, cout, 1, stride=stride,
+                                        padding=0, activation="none", bias=False)
+            self.down_bn = nn.BatchNorm2d(cout)
+        else:
+            self.down_conv = None
+
+    def forward(self, x):
+        idt = x
+        y = torch.relu(self.bn1(self.conv1(x)))
+        y = self.bn2(self.conv2(y))
+        if self.down_conv is not None:
+            idt = self.down_bn(self.down_conv(x))
+        return torch.relu(y + idt)
+
+
+class ResNet18Shape(nn.Module):
+    """ResNet-18 layer shapes for 3x224x224 synthetic inputs."""
+
+    def __init__(self, num_classes: int = 1000):
+        super().__init__()
+        self.stem_conv = snn.Conv2d(3, 64, 7, stride=2, padding=3,
+                                    activation="none", bias=False)
+        self.stem_bn = nn.BatchNorm2d(64)
+        self.stem_pool = snn.MaxPool2d(3, 2)  # 112 -> 55 (valid pooling)
+        stages = []
+        cin = 64
+        for cout, stride in [(64, 1), (64, 1), (128, 2), (128, 1),
+                             (256, 2), (256, 1), (512, 2), (512, 1)]:
+            stages.append(_BasicBlock(cin, cout, stride))
+            cin = cout
+        self.stages = nn.ModuleList(stages)
+        self.fc = snn.Linear(512, num_classes)
+        self.loss_fn = snn.SoftmaxCrossEntropy()
+
+    def forward(self, x):
+        if x.dim() == 2:
+            x = x.reshape(-1, 3, 224, 224)
+        x = torch.relu(self.stem_bn(self.stem_conv(x)))
+        x = self.stem_pool(x)
+        for block in self.stages:
+            x = block(x)
+        x = x.mean(dim=(2, 3))
+        return self.fc(x)
+
+    def compute_loss(self, out, target):
+        return self.loss_fn(out, target)
+
+    @torch.no_grad()
+    def predict(self, x):
+        return F.argmax_rows(self.forward(x))

@@ -1,0 +1,9 @@
+from sparkflow_amd.nn.modules import (  # noqa: F401
+    Linear,
+    Conv2d,
+    MaxPool2d,
+    Flatten,
+    Dropout,
+    SoftmaxCrossEntropy,
+    MSE,
+)

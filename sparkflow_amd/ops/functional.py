@@ -1,0 +1,212 @@
+"""Autograd wrappers over the CDNA4 HIP kernels (fp32).
+
+Each op is a ``torch.autograd.Function`` whose GPU path calls the gfx950
+extension and whose CPU path implements identical fp32 math with plain
+torch ops (used by the no-GPU tests; numerics tests compare the HIP path
+against this path).
+
+Replaces the reference's implicit TF ops: dense
+GEMM+bias+activation, conv2d, maxpool, softmax-cross-entropy, MSE, argmax.
+"""
+from __future__ import annotations
+
+import torch
+
+from sparkflow_amd import ops
+
+ACT_NONE, ACT_RELU, ACT_SIGMOID = 0, 1, 2
+ACT_CODES = {"none": ACT_NONE, "linear": ACT_NONE, "relu": ACT_RELU,
+             "sigmoid": ACT_SIGMOID}
+
+
+def _act_fwd(pre: torch.Tensor, act: int) -> torch.Tensor:
+    if act == ACT_RELU:
+        return torch.relu(pre)
+    if act == ACT_SIGMOID:
+        return torch.sigmoid(pre)
+    return pre
+
+
+def _act_bwd_from_out(y: torch.Tensor, dy: torch.Tensor, act: int) -> torch.Tensor:
+    # activation grads recomputed from the *output* (saves the preactivation)
+    if act == ACT_RELU:
+        return dy * (y > 0).to(dy.dtype)
+    if act == ACT_SIGMOID:
+        return dy * y * (1.0 - y)
+    return dy
+
+
+class LinearAct(torch.autograd.Function):
+    """y = act(x @ W + b) with x:[M,K], W:[K,N] (TF dense layout), b:[N]."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, act: int):
+        if ops.use_hip(x):
+            y = ops.hip.gemm_act(x, w, b if b is not None else torch.empty(0, device=x.device), act)
+        else:
+            pre = x @ w
+            if b is not None:
+                pre = pre + b
+            y = _act_fwd(pre, act)
+        ctx.save_for_backward(x, w, y)
+        ctx.act = act
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        act = ctx.act
+        dy = dy.contiguous()
+        if ops.use_hip(x):
+            dpre = ops.hip.act_bwd(y, dy, act) if act != ACT_NONE else dy
+            dx = ops.hip.gemm_dgrad(dpre, w)
+            dw = ops.hip.gemm_wgrad(x, dpre)
+            db = ops.hip.bias_grad(dpre) if ctx.has_bias else None
+        else:
+            dpre = _act_bwd_from_out(y, dy, act)
+            dx = dpre @ w.t()
+            dw = x.t() @ dpre
+            db = dpre.sum(dim=0) if ctx.has_bias else None
+        return dx, dw, db, None
+
+
+def linear_act(x, w, b=None, act="none"):
+    return LinearAct.apply(x, w, b, ACT_CODES[act] if isinstance(act, str) else act)
+
+
+class SoftmaxXent(torch.autograd.Function):
+    """Mean softmax-cross-entropy over rows; labels are one-hot [M,C]."""
+
+    @staticmethod
+    def forward(ctx, logits, onehot):
+        if ops.use_hip(logits):
+            loss, probs = ops.hip.softmax_xent_fwd(logits, onehot)
+        else:
+            lse = torch.logsumexp(logits, dim=1, keepdim=True)
+            logp = logits - lse
+            loss = -(onehot * logp).sum(dim=1).mean()
+            probs = torch.exp(logp)
+        ctx.save_for_backward(probs, onehot)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        probs, onehot = ctx.saved_tensors
+        m = probs.shape[0]
+        if ops.use_hip(probs):
+            dlogits = ops.hip.softmax_xent_bwd(probs, onehot, float(dloss) / m)
+        else:
+            dlogits = (probs - onehot) * (dloss / m)
+        return dlogits, None
+
+
+def softmax_xent(logits, labels):
+    """labels: one-hot float [M,C] or class-index long [M]."""
+    if labels.dtype in (torch.int64, torch.int32):
+        onehot = torch.zeros_like(logits)
+        onehot.scatter_(1, labels.view(-1, 1).long(), 1.0)
+        labels = onehot
+    elif labels.dim() == 1:
+        labels = labels.view(-1, 1)
+    return SoftmaxXent.apply(logits, labels.to(logits.dtype))
+
+
+class MseLoss(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, pred, target):
+        ctx.save_for_backward(pred, target)
+        if ops.use_hip(pred):
+            return ops.hip.mse_fwd(pred, target)
+        return ((pred - target) ** 2).mean()
+
+    @staticmethod
+    def backward(ctx, dloss):
+        pred, target = ctx.saved_tensors
+        n = pred.numel()
+        if ops.use_hip(pred):
+            dpred = ops.hip.mse_bwd(pred, target, 2.0 * float(dloss) / n)
+        else:
+            dpred = (pred - target) * (2.0 * dloss / n)
+        return dpred, None
+
+
+def mse_loss(pred, target):
+    return MseLoss.apply(pred, target)
+
+
+class Conv2dAct(torch.autograd.Function):
+    """NCHW conv2d + bias + activation. w: [Cout, Cin, Kh, Kw]."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, stride: int, pad: int, act: int):
+        if ops.use_hip(x):
+            y = ops.hip.conv2d_act(
+                x, w, b if b is not None else torch.empty(0, device=x.device),
+                stride, pad, act)
+        else:
+            pre = torch.nn.functional.conv2d(x, w, b, stride=stride, padding=pad)
+            y = _act_fwd(pre, act)
+        ctx.save_for_backward(x, w, y)
+        ctx.conf = (stride, pad, act, b is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        stride, pad, act, has_bias = ctx.conf
+        dy = dy.contiguous()
+        if ops.use_hip(x):
+            dpre = ops.hip.act_bwd(y, dy, act) if act != ACT_NONE else dy
+            dx = ops.hip.conv2d_dgrad(dpre, w, x.shape[2], x.shape[3], stride, pad)
+            dw = ops.hip.conv2d_wgrad(x, dpre, w.shape[2], w.shape[3], stride, pad)
+            db = ops.hip.conv_bias_grad(dpre) if has_bias else None
+        else:
+            dpre = _act_bwd_from_out(y, dy, act)
+            dx = torch.nn.grad.conv2d_input(x.shape, w, dpre, stride=stride,
+                                            padding=pad)
+            dw = torch.nn.grad.conv2d_weight(x, w.shape, dpre, stride=stride,
+                                             padding=pad)
+            db = dpre.sum(dim=(0, 2, 3)) if has_bias else None
+        return dx, dw, db, None, None, None
+
+
+def conv2d_act(x, w, b=None, stride=1, pad=0, act="none"):
+    return Conv2dAct.apply(x, w, b, stride, pad,
+                           ACT_CODES[act] if isinstance(act, str) else act)
+
+
+class MaxPool2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k: int, stride: int):
+        if ops.use_hip(x):
+            y, idx = ops.hip.maxpool2d_fwd(x, k, stride)
+        else:
+            y, idx = torch.nn.functional.max_pool2d(x, k, stride,
+                                                    return_indices=True)
+        ctx.save_for_backward(idx)
+        ctx.conf = (x.shape, k, stride)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        shape, k, stride = ctx.conf
+        dy = dy.contiguous()
+        if ops.use_hip(dy):
+            dx = ops.hip.maxpool2d_bwd(dy, idx, shape[2], shape[3])
+        else:
+            dx = torch.nn.functional.max_unpool2d(dy, idx, k, stride,
+                                                  output_size=shape[2:])
+        return dx, None, None
+
+
+def maxpool2d(x, k, stride=None):
+    return MaxPool2d.apply(x, k, stride or k)
+
+
+def argmax_rows(y: torch.Tensor) -> torch.Tensor:
+    """Rowwise argmax prediction head (reference tf.argmax(out, 1))."""
+    if ops.use_hip(y):
+        return ops.hip.argmax_rows(y)
+    return torch.argmax(y, dim=1).to(torch.float32)

@@ -1,0 +1,312 @@
+"""Pull/push transports between workers and the parameter server.
+
+The reference moves weights/grads as HTTP + pickle (reference
+HogwildSparkModel.py:22-35).  Here every message is ONE
+contiguous fp32 buffer (all trainable tensors flattened), in three
+transports behind one interface:
+
+* ``LocalTransport``   — PS in the same process (1-GPU jobs, tests): direct
+  calls into the PS core; Hogwild raciness comes from PS service threads
+  sharing the weight buffer.
+* ``TcpTransport``     — raw-socket length-prefixed fp32 frames against a
+  threaded PS server.  This is the Spark-executor / CPU-plumbing path
+  (BASELINE.json config 1) — same pull/push semantics as the reference's
+  GET /parameters / POST /update, minus Flask and pickle.
+* ``DistTransport``    — torch.distributed point-to-point: per-worker
+  {PS, worker} pair process-groups, payload over the ``nccl`` backend
+  (= RCCL over xGMI on an MI355X node) or ``gloo`` on CPU, control
+  messages over a per-worker gloo pair group.  The PS star over p2p xGMI
+  links gives each worker<->PS pair its own ~153 GB/s link.
+"""
+from __future__ import annotations
+
+import socket
+import struct
+import threading
+import time
+from typing import Optional
+
+import numpy as np
+import torch
+
+# control ops (DistTransport control channel + TCP op bytes)
+OP_PULL, OP_PUSH, OP_DONE, OP_PING = 1, 2, 3, 4
+
+
+class WorkerTransport:
+    def pull(self) -> torch.Tensor:
+        raise NotImplementedError
+
+    def push(self, flat_grads: torch.Tensor) -> None:
+        raise NotImplementedError
+
+    def close(self) -> None:
+        pass
+
+
+class LocalTransport(WorkerTransport):
+    """Direct in-process handle to the PS core."""
+
+    def __init__(self, ps_core, device="cpu"):
+        self.ps = ps_core
+        self.device = torch.device(device)
+
+    def pull(self) -> torch.Tensor:
+        return self.ps.read_weights().to(self.device, non_blocking=True)
+
+    def push(self, flat_grads: torch.Tensor) -> None:
+        self.ps.apply_gradients(flat_grads.to(self.ps.device, non_blocking=True))
+
+
+# ---------------------------------------------------------------------------
+# TCP transport
+# ---------------------------------------------------------------------------
+
+_HDR = struct.Struct("<Q")
+
+
+def _recv_exact(sock: socket.socket, n: int) -> bytes:
+    chunks = []
+    while n:
+        b = sock.recv(min(n, 1 << 20))
+        if not b:
+            raise ConnectionError("peer closed")
+        chunks.append(b)
+        n -= len(b)
+    return b"".join(chunks)
+
+
+class TcpTransport(WorkerTransport):
+    def __init__(self, master_url: str, device="cpu", timeout: float = 120.0):
+        host, port = master_url.rsplit(":", 1)
+        self.device = torch.device(device)
+        self.sock = socket.create_connection((host, int(port)), timeout=timeout)
+        self.sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+
+    def pull(self) -> torch.Tensor:
+        self.sock.sendall(bytes([OP_PULL]))
+        (nbytes,) = _HDR.unpack(_recv_exact(self.sock, _HDR.size))
+        buf = _recv_exact(self.sock, nbytes)
+        arr = np.frombuffer(buf, dtype=np.float32)
+        return torch.from_numpy(arr.copy()).to(self.device)
+
+    def push(self, flat_grads: torch.Tensor) -> None:
+        data = flat_grads.detach().to("cpu", torch.float32).numpy().tobytes()
+        self.sock.sendall(bytes([OP_PUSH]) + _HDR.pack(len(data)) + data)
+        status = _recv_exact(self.sock, 1)[0]
+        if status != 0:
+            raise RuntimeError("parameter server rejected update (error budget)")
+
+    def ping(self) -> bool:
+        self.sock.sendall(bytes([OP_PING]))
+        return _recv_exact(self.sock, 1)[0] == 0
+
+    def close(self) -> None:
+        try:
+            self.sock.sendall(bytes([OP_DONE]))
+            self.sock.close()
+        except OSError:
+            pass
+
+
+class TcpPSServer:
+    """Threaded socket front-end over a PS core — one service thread per
+    worker connection (the Hogwild concurrency surface, reference
+    HogwildSparkModel.py:244 ``threaded=True``)."""
+
+    def __init__(self, ps_core, host: str = "0.0.0.0", port: int = 5000):
+        self.ps = ps_core
+        self.sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        self.sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        self.sock.bind((host, port))
+        self.port = self.sock.getsockname()[1]
+        self.sock.listen(128)
+        self._stop = threading.Event()
+        self._threads = []
+        self._accept_thread = threading.Thread(target=self._accept_loop,
+                                               daemon=True)
+
+    def start(self):
+        self._accept_thread.start()
+        return self
+
+    def _accept_loop(self):
+        self.sock.settimeout(0.2)
+        while not self._stop.is_set():
+            try:
+                conn, _ = self.sock.accept()
+            except socket.timeout:
+                continue
+            except OSError:
+                break
+            t = threading.Thread(target=self._serve, args=(conn,), daemon=True)
+            t.start()
+            self._threads.append(t)
+
+    def _serve(self, conn: socket.socket):
+        conn.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        try:
+            while not self._stop.is_set():
+                op = conn.recv(1)
+                if not op:
+                    return
+                op = op[0]
+                if op == OP_PULL:
+                    data = self.ps.read_weights().to("cpu", torch.float32) \
+                        .numpy().tobytes()
+                    conn.sendall(_HDR.pack(len(data)) + data)
+                elif op == OP_PUSH:
+                    (nbytes,) = _HDR.unpack(_recv_exact(conn, _HDR.size))
+                    buf = _recv_exact(conn, nbytes)
+                    grads = torch.from_numpy(
+                        np.frombuffer(buf, dtype=np.float32).copy())
+                    try:
+                        self.ps.apply_gradients(grads.to(self.ps.device))
+                        conn.sendall(b"\x00")
+                    except Exception:
+                        conn.sendall(b"\x01")
+                        raise
+                elif op == OP_PING:
+                    conn.sendall(b"\x00")
+                elif op == OP_DONE:
+                    return
+        except (ConnectionError, OSError):
+            pass
+        finally:
+            conn.close()
+
+    def stop(self):
+        self._stop.set()
+        try:
+            self.sock.close()
+        except OSError:
+            pass
+
+
+def wait_for_server(master_url: str, timeout: float = 60.0) -> None:
+    """Readiness handshake replacing the reference's fixed ``sleep(8)``
+    (HogwildSparkModel.py:117,135)."""
+    deadline = time.monotonic() + timeout
+    last_err: Optional[Exception] = None
+    while time.monotonic() < deadline:
+        try:
+            t = TcpTransport(master_url, timeout=5.0)
+            ok = t.ping()
+            t.close()
+            if ok:
+                return
+        except (OSError, ConnectionError) as e:
+            last_err = e
+        time.sleep(0.1)
+    raise TimeoutError(f"parameter server at {master_url} not ready: {last_err}")
+
+
+# ---------------------------------------------------------------------------
+# torch.distributed transport (RCCL p2p over xGMI on GPU, gloo on CPU)
+# ---------------------------------------------------------------------------
+
+
+class DistPairs:
+    """Pair process-groups between rank 0 (PS) and each worker rank.
+
+    Every rank must construct this collectively (``new_group`` is a
+    collective call).  Payload groups use the default backend of the world
+    (nccl on GPU = one RCCL communicator per worker<->PS xGMI link); control
+    groups are always gloo (small CPU int tensors carry PULL/PUSH/DONE)."""
+
+    def __init__(self, payload_backend: Optional[str] = None):
+        import torch.distributed as dist
+
+        self.world = dist.get_world_size()
+        self.rank = dist.get_rank()
+        self.payload = {}
+        self.ctrl = {}
+        for w in range(1, self.world):
+            self.payload[w] = dist.new_group([0, w], backend=payload_backend)
+            self.ctrl[w] = dist.new_group([0, w], backend="gloo")
+
+
+class DistTransport(WorkerTransport):
+    """Worker side of the RCCL star."""
+
+    def __init__(self, pairs: DistPairs, numel: int, device):
+        import torch.distributed as dist
+
+        self.dist = dist
+        self.rank = pairs.rank
+        self.pg = pairs.payload[self.rank]
+        self.cg = pairs.ctrl[self.rank]
+        self.device = torch.device(device)
+        self.buf = torch.empty(numel, dtype=torch.float32, device=self.device)
+        self._ctrl_buf = torch.zeros(1, dtype=torch.int64)
+
+    def _ctrl(self, op: int):
+        self._ctrl_buf.fill_(op)
+        self.dist.send(self._ctrl_buf, dst=0, group=self.cg)
+
+    def pull(self) -> torch.Tensor:
+        self._ctrl(OP_PULL)
+        self.dist.recv(self.buf, src=0, group=self.pg)
+        return self.buf
+
+    def push(self, flat_grads: torch.Tensor) -> None:
+        self._ctrl(OP_PUSH)
+        self.dist.send(flat_grads.contiguous(), dst=0, group=self.pg)
+
+    def close(self) -> None:
+        self._ctrl(OP_DONE)
+
+
+class DistPSService:
+    """PS side of the RCCL star: one host thread per worker, each driving
+    its own pair communicator and its own HIP stream, so one slow worker
+    never blocks another (SURVEY.md §7 hard-part 1).  Hogwild mode lets the
+    per-thread optimizer applies race on the weight buffer; lock mode gates
+    them through the PS core's RWLock."""
+
+    def __init__(self, ps_core, pairs: DistPairs):
+        import torch.distributed as dist
+
+        self.dist = dist
+        self.ps = ps_core
+        self.pairs = pairs
+        self._threads = []
+
+    def start(self):
+        for w in range(1, self.pairs.world):
+            t = threading.Thread(target=self._serve, args=(w,), daemon=True)
+            t.start()
+            self._threads.append(t)
+        return self
+
+    def _serve(self, w: int):
+        dist = self.dist
+        pg = self.pairs.payload[w]
+        cg = self.pairs.ctrl[w]
+        dev = self.ps.device
+        use_cuda = dev.type == "cuda"
+        stream = torch.cuda.Stream(device=dev) if use_cuda else None
+        grad_buf = torch.empty(self.ps.numel, dtype=torch.float32, device=dev)
+        ctrl = torch.zeros(1, dtype=torch.int64)
+        while True:
+            dist.recv(ctrl, src=w, group=cg)
+            op = int(ctrl.item())
+            if op == OP_DONE:
+                return
+            if use_cuda:
+                with torch.cuda.stream(stream):
+                    if op == OP_PULL:
+                        self.ps.send_weights(lambda buf: dist.send(buf, dst=w, group=pg))
+                    elif op == OP_PUSH:
+                        dist.recv(grad_buf, src=w, group=pg)
+                        self.ps.apply_gradients(grad_buf)
+            else:
+                if op == OP_PULL:
+                    self.ps.send_weights(lambda buf: dist.send(buf, dst=w, group=pg))
+                elif op == OP_PUSH:
+                    dist.recv(grad_buf, src=w, group=pg)
+                    self.ps.apply_gradients(grad_buf)
+
+    def join(self):
+        for t in self._threads:
+            t.join()

@@ -1,0 +1,172 @@
+"""Worker training loop — parity with reference ``handle_model``
+(HogwildSparkModel.py:38-100), GPU-resident.
+
+Per outer iteration: pull the flat weight buffer from the PS, load it into
+the module, optionally shuffle, then one of three batching modes:
+
+* stochastic minibatches (``mini_stochastic_iters >= 1``): msi random
+  minibatches per epoch, one push each (weights pulled once per epoch) —
+  reference HogwildSparkModel.py:62-71;
+* sequential minibatches (``mini_batch >= 1``): walk the partition in
+  order, RE-PULLING weights before every batch — reference 72-83 (that
+  per-batch re-pull is reference behavior, preserved);
+* full batch — reference 84-92.
+
+Gradients are packed into one flat fp32 buffer and pushed; push failures
+print and continue (reference 68-71).  ``verbose``/``loss_callback`` get a
+full-partition loss per outer iteration (reference 94-100).
+"""
+from __future__ import annotations
+
+import uuid
+from typing import Callable, Optional
+
+import numpy as np
+import torch
+
+from sparkflow_amd.parallel.ps import TrainConfig
+from sparkflow_amd.parallel.transport import WorkerTransport
+from sparkflow_amd.utils.serialization import (
+    module_from_json,
+    pack_grads,
+    unpack_params,
+)
+
+
+class Worker:
+    def __init__(self, graph_json: str, transport: WorkerTransport,
+                 config: TrainConfig, device="cpu",
+                 loss_callback: Optional[Callable] = None):
+        self.device = torch.device(device)
+        self.module = module_from_json(graph_json).to(self.device).train()
+        self.transport = transport
+        self.config = config
+        self.loss_callback = loss_callback
+        self.partition_id = uuid.uuid4().hex
+        self.rng = np.random.default_rng(config.seed)
+        self._grad_buf = None
+        self.steps_done = 0  # pushes performed (for benchmarking)
+
+    # -- data staging --------------------------------------------------------
+
+    def stage(self, features: np.ndarray, labels: Optional[np.ndarray]):
+        """Move the partition to the device once (the reference re-feeds
+        host ndarrays every session.run; HBM is 288 GB — keep it resident)."""
+        self.x = torch.as_tensor(np.ascontiguousarray(features),
+                                 dtype=torch.float32, device=self.device)
+        if labels is not None:
+            if np.issubdtype(np.asarray(labels).dtype, np.floating):
+                self.y = torch.as_tensor(labels, dtype=torch.float32,
+                                         device=self.device)
+            else:
+                self.y = torch.as_tensor(labels, device=self.device)
+        else:
+            self.y = None
+
+    # -- core step pieces ----------------------------------------------------
+
+    def _pull_and_load(self):
+        flat = self.transport.pull()
+        unpack_params(self.module, flat)
+
+    def _grad_step(self, xb: torch.Tensor, yb: Optional[torch.Tensor]):
+        self.module.zero_grad(set_to_none=True)
+        out = self.module(xb)
+        target = yb if yb is not None else xb
+        loss = self.module.compute_loss(out, target)
+        loss.backward()
+        self._grad_buf = pack_grads(self.module, self._grad_buf)
+        return self._grad_buf
+
+    def _push(self, grads: torch.Tensor):
+        try:
+            self.transport.push(grads)
+            self.steps_done += 1
+        except Exception:
+            print("Timeout error from partition %s" % self.partition_id)
+
+    def _shuffle_indices(self, n: int) -> torch.Tensor:
+        return torch.as_tensor(self.rng.permutation(n), device=self.device)
+
+    def _minibatch_indices(self, n: int, mbs: int) -> torch.Tensor:
+        return torch.as_tensor(self.rng.choice(n, mbs, replace=False),
+                               device=self.device)
+
+    # -- the loop ------------------------------------------------------------
+
+    def run(self, features: np.ndarray, labels: Optional[np.ndarray]) -> None:
+        self.stage(features, labels)
+        cfg = self.config
+        n = self.x.shape[0]
+        mbs = cfg.mini_batch
+        if mbs > n:
+            mbs = n - 1  # reference clamp (ml_util.py:105-106)
+
+        for it in range(cfg.iters):
+            self._pull_and_load()
+
+            if cfg.shuffle:
+                order = self._shuffle_indices(n)
+                self.x = self.x[order]
+                if self.y is not None:
+                    self.y = self.y[order]
+
+            if cfg.mini_stochastic_iters >= 1 and mbs >= 1:
+                for _ in range(cfg.mini_stochastic_iters):
+                    pick = self._minibatch_indices(n, mbs)
+                    yb = self.y[pick] if self.y is not None else None
+                    self._push(self._grad_step(self.x[pick], yb))
+            elif mbs >= 1:
+                for r in range(0, n, mbs):
+                    self._pull_and_load()  # reference re-pulls per batch
+                    xb = self.x[r:r + mbs]
+                    yb = self.y[r:r + mbs] if self.y is not None else None
+                    self._push(self._grad_step(xb, yb))
+            else:
+                self._push(self._grad_step(self.x, self.y))
+
+            if cfg.verbose or self.loss_callback:
+                with torch.no_grad():
+                    out = self.module(self.x)
+                    target = self.y if self.y is not None else self.x
+                    loss = float(self.module.compute_loss(out, target))
+                if cfg.verbose:
+                    print("Partition Id: %s, Iteration: %i, Loss: %f"
+                          % (self.partition_id, it, loss))
+                if self.loss_callback:
+                    self.loss_callback(loss, it, self.partition_id)
+
+
+    def run_steps(self, k: int, mini_batch: int) -> int:
+        """Benchmark-shaped loop: k steps of {pull -> one stochastic
+        minibatch grad -> push}.  Returns samples processed."""
+        n = self.x.shape[0]
+        mbs = min(mini_batch, n)
+        for _ in range(k):
+            self._pull_and_load()
+            if mbs < n:
+                pick = self._minibatch_indices(n, mbs)
+                xb = self.x[pick]
+                yb = self.y[pick] if self.y is not None else None
+            else:
+                xb, yb = self.x, self.y
+            self._push(self._grad_step(xb, yb))
+        return k * mbs
+
+
+def run_partition(data_iter, graph_json: str, transport: WorkerTransport,
+                  config: TrainConfig, is_supervised: bool, device="cpu",
+                  loss_callback=None):
+    """Entry used per partition (Spark executor closure / local worker)."""
+    from sparkflow_amd.ml_util import handle_features
+
+    data = list(data_iter)
+    if not data:
+        return
+    features, labels = handle_features(data, is_supervised)
+    w = Worker(graph_json, transport, config, device=device,
+               loss_callback=loss_callback)
+    try:
+        w.run(features, labels)
+    finally:
+        transport.close()

@@ -1,0 +1,93 @@
+"""torch.distributed engine tests over gloo (CPU stand-in for the RCCL
+star used on the MI355X node; same code path, different backend)."""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from tests.conftest import free_port
+
+
+def _dist_worker(rank, world, port, iters):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["LOCAL_RANK"] = str(rank)
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import gaussian_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    graph_json = build_graph(gaussian_mlp)
+    cfg = TrainConfig(iters=iters, mini_batch=16, mini_stochastic_iters=1,
+                      shuffle=True)
+    engine = build_engine(graph_json, OptimizerSpec("adam", 0.01), cfg,
+                          device="cpu")
+
+    rng = np.random.default_rng(100 + rank)
+    n = 64
+    x = np.concatenate([rng.normal(-2, 0.7, (n // 2, 10)),
+                        rng.normal(2, 0.7, (n // 2, 10))]).astype(np.float32)
+    y = np.concatenate([np.zeros((n // 2, 1)), np.ones((n // 2, 1))]) \
+        .astype(np.float32)
+
+    w_before = engine.ps.weights.clone() if engine.rank == 0 else None
+    engine.barrier()
+    engine.worker.run(x, y)
+    engine.close()
+    engine.barrier()
+
+    if engine.rank == 0:
+        assert engine.ps.update_count >= world * iters
+        assert not torch.equal(w_before, engine.ps.weights)
+        assert torch.isfinite(engine.ps.weights).all()
+
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.parametrize("world", [1, 2, 3])
+def test_dist_engine_gloo(world):
+    port = free_port()
+    if world == 1:
+        _dist_worker(0, 1, port, iters=4)
+    else:
+        mp.spawn(_dist_worker, args=(world, port, 4), nprocs=world,
+                 join=True)
+
+
+def test_dist_engine_lock_mode():
+    port = free_port()
+    mp.spawn(_dist_lock_worker, args=(2, port), nprocs=2, join=True)
+
+
+def _dist_lock_worker(rank, world, port):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["LOCAL_RANK"] = str(rank)
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import gaussian_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    graph_json = build_graph(gaussian_mlp)
+    cfg = TrainConfig(iters=3, acquire_lock=True)
+    engine = build_engine(graph_json, OptimizerSpec("gradient_descent", 0.01),
+                          cfg, device="cpu")
+    rng = np.random.default_rng(rank)
+    x = rng.normal(size=(32, 10)).astype(np.float32)
+    y = rng.random((32, 1)).astype(np.float32)
+    engine.barrier()
+    engine.worker.run(x, y)
+    engine.close()
+    engine.barrier()
+    if engine.rank == 0:
+        assert engine.ps.update_count == world * 3
+    torch.distributed.destroy_process_group()

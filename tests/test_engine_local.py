@@ -1,0 +1,156 @@
+"""End-to-end async-PS training on the local (threaded workers + spawned
+PS process over TCP) path — the no-Spark analog of the reference's
+Spark-local[2] integration tests (reference tests/dl_runner.py)."""
+import numpy as np
+import pytest
+
+from sparkflow_amd.graph_utils import build_graph
+from sparkflow_amd.ml_util import convert_weights_to_json, torch_set_weights
+from sparkflow_amd.models import gaussian_mlp, tabular_autoencoder, xor_mlp
+from sparkflow_amd.parallel.hogwild import HogwildSparkModel
+from sparkflow_amd.parallel.ps import OptimizerSpec
+from sparkflow_amd.utils.serialization import module_from_json
+
+import torch
+
+
+def _gaussian_data(n=400, seed=12345):
+    rng = np.random.default_rng(seed)
+    half = n // 2
+    x0 = rng.normal(loc=-2.0, scale=0.7, size=(half, 10))
+    x1 = rng.normal(loc=2.0, scale=0.7, size=(half, 10))
+    rows = [(x0[i], 0) for i in range(half)] + [(x1[i], 1) for i in range(half)]
+    rng.shuffle(rows)
+    return rows
+
+
+def _partitions(rows, k):
+    return [rows[i::k] for i in range(k)]
+
+
+def _accuracy(graph_json, weights, rows):
+    module = module_from_json(graph_json)
+    torch_set_weights(module, weights)
+    module.eval()
+    x = torch.tensor(np.asarray([r[0] for r in rows]), dtype=torch.float32)
+    y = np.asarray([r[1] for r in rows])
+    with torch.no_grad():
+        pred = (module(x).numpy().reshape(-1) > 0.5).astype(int)
+    return (pred == y).mean()
+
+
+def _train(rows, graph_json, tcp_port, **kw):
+    defaults = dict(
+        tensorflowGraph=graph_json,
+        iters=30,
+        tfInput="x",
+        tfLabel="y",
+        optimizer=OptimizerSpec("adam", 0.01),
+        master_url="127.0.0.1:%d" % tcp_port,
+        port=tcp_port,
+        worker_device="cpu",
+        ps_device="cpu",
+    )
+    defaults.update(kw)
+    model = HogwildSparkModel(**defaults)
+    return model.train(_partitions(rows, 2))
+
+
+def test_hogwild_gaussians_converge(tcp_port):
+    rows = _gaussian_data()
+    graph_json = build_graph(gaussian_mlp)
+    weights = _train(rows, graph_json, tcp_port)
+    assert len(weights) > 0
+    acc = _accuracy(graph_json, weights, rows)
+    assert acc > 0.8, acc
+
+
+def test_lock_mode_converges(tcp_port):
+    rows = _gaussian_data(n=200)
+    graph_json = build_graph(gaussian_mlp)
+    weights = _train(rows, graph_json, tcp_port, acquire_lock=True, iters=20)
+    acc = _accuracy(graph_json, weights, rows)
+    assert acc > 0.8, acc
+
+
+def test_stochastic_minibatch_mode(tcp_port):
+    rows = _gaussian_data(n=200)
+    graph_json = build_graph(gaussian_mlp)
+    weights = _train(rows, graph_json, tcp_port, iters=15,
+                     mini_batch=32, mini_stochastic_iters=3)
+    acc = _accuracy(graph_json, weights, rows)
+    assert acc > 0.75, acc
+
+
+def test_sequential_minibatch_mode(tcp_port):
+    rows = _gaussian_data(n=200)
+    graph_json = build_graph(gaussian_mlp)
+    weights = _train(rows, graph_json, tcp_port, iters=10, mini_batch=50)
+    acc = _accuracy(graph_json, weights, rows)
+    assert acc > 0.75, acc
+
+
+def test_partition_shuffles(tcp_port):
+    rows = _gaussian_data(n=200)
+    graph_json = build_graph(gaussian_mlp)
+    weights = _train(rows, graph_json, tcp_port, iters=10,
+                     partition_shuffles=2)
+    acc = _accuracy(graph_json, weights, rows)
+    assert acc > 0.75, acc
+
+
+def test_autoencoder_unsupervised(tcp_port):
+    rng = np.random.default_rng(0)
+    rows = [rng.random(784).astype(np.float32) for _ in range(64)]
+    graph_json = build_graph(tabular_autoencoder)
+
+    model = HogwildSparkModel(
+        tensorflowGraph=graph_json,
+        iters=10,
+        tfInput="x",
+        tfLabel=None,  # unsupervised
+        optimizer=OptimizerSpec("adam", 0.01),
+        master_url="127.0.0.1:%d" % tcp_port,
+        port=tcp_port,
+        mini_batch=16,
+        mini_stochastic_iters=2,
+        worker_device="cpu",
+        ps_device="cpu",
+    )
+    weights = model.train(_partitions(rows, 2))
+    assert len(weights) == 4  # 2 layers x (W, b)
+    for w in weights:
+        assert np.isfinite(w).all()
+
+
+def test_verbose_and_loss_callback(tcp_port, capsys):
+    rows = _gaussian_data(n=100)
+    graph_json = build_graph(gaussian_mlp)
+    seen = []
+
+    model = HogwildSparkModel(
+        tensorflowGraph=graph_json,
+        iters=3,
+        tfInput="x",
+        tfLabel="y",
+        optimizer=OptimizerSpec("gradient_descent", 0.01),
+        master_url="127.0.0.1:%d" % tcp_port,
+        port=tcp_port,
+        verbose=1,
+        loss_callback=lambda loss, it, pid: seen.append((loss, it, pid)),
+        worker_device="cpu",
+        ps_device="cpu",
+    )
+    model.train(_partitions(rows, 2))
+    assert len(seen) == 6  # 2 partitions x 3 iters
+    out = capsys.readouterr().out
+    assert "Partition Id" in out
+
+
+def test_final_weights_json_roundtrip(tcp_port):
+    rows = _gaussian_data(n=100)
+    graph_json = build_graph(xor_mlp)
+    rows2 = [(r[0][:2], r[1]) for r in rows]
+    weights = _train(rows2, graph_json, tcp_port, iters=5)
+    js = convert_weights_to_json(weights)
+    assert isinstance(js, str) and js.startswith("[")
