@@ -1,0 +1,142 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: MNIST 784-256-256-10 MLP, async-PS Hogwild,
+samples/sec (BASELINE.json headline metric).
+
+Single node, one rank per GPU:
+  python bench.py --gpus 1 --steps 200 --warmup 50
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 hosts the GPU parameter server (flat fp32 weights in HBM, fused
+Adam applies) plus a local worker; ranks 1..N-1 are workers over RCCL p2p
+pair communicators (xGMI star).  A "step" is one worker iteration:
+pull weights -> forward/backward on a 300-sample minibatch via the CDNA4
+HIP kernels -> push gradients.  Weak scaling: per-GPU work is fixed.
+
+Data is synthetic (random MNIST-shaped tensors, random-init weights);
+dtype fp32 (the reference's TF-1.x compute dtype).
+"""
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=50)
+    ap.add_argument("--mbs", type=int, default=300,
+                    help="minibatch per step (reference config: 300)")
+    ap.add_argument("--model", default="mnist_mlp",
+                    choices=["mnist_mlp", "mnist_cnn", "tabular_autoencoder",
+                             "resnet18_shape"])
+    ap.add_argument("--optimizer", default="adam")
+    ap.add_argument("--lr", type=float, default=0.001)
+    ap.add_argument("--lock", action="store_true",
+                    help="async-with-lock mode instead of Hogwild")
+    args = ap.parse_args()
+
+    from sparkflow_amd import models
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+
+    builder = getattr(models, args.model)
+    graph_json = build_graph(builder)
+
+    cfg = TrainConfig(iters=args.steps, mini_batch=args.mbs,
+                      mini_stochastic_iters=1, shuffle=False,
+                      acquire_lock=args.lock)
+    engine = build_engine(graph_json, OptimizerSpec(args.optimizer, args.lr),
+                          cfg)
+    device = engine.device
+
+    # synthetic data of the benchmark config's shape
+    rng = np.random.default_rng(1234 + rank)
+    n_rows = max(2 * args.mbs, 2048)
+    if args.model == "resnet18_shape":
+        n_rows = max(2 * args.mbs, 64)
+        x = rng.random((n_rows, 3 * 224 * 224), dtype=np.float32)
+        y = rng.integers(0, 1000, n_rows).astype(np.float32).reshape(-1, 1)
+    elif args.model == "mnist_cnn" or args.model == "mnist_mlp":
+        x = rng.random((n_rows, 784), dtype=np.float32)
+        y = rng.integers(0, 10, n_rows).astype(np.float32).reshape(-1, 1)
+    else:  # autoencoder: unsupervised
+        x = rng.random((n_rows, 784), dtype=np.float32)
+        y = None
+    engine.worker.stage(x, y)
+
+    # warmup (also initializes the RCCL pair communicators)
+    engine.worker.run_steps(args.warmup, args.mbs)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    engine.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    engine.worker.run_steps(args.steps, args.mbs)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    engine.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if device.type == "cuda" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    mbs_eff = min(args.mbs, x.shape[0])
+    total_samples = args.steps * mbs_eff * world
+    value = total_samples / elapsed
+
+    if rank == 0:
+        result = {
+            "metric": "samples_per_sec",
+            "value": value,
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": {"mnist_mlp": "mnist_mlp_784_256_256_10",
+                          "mnist_cnn": "mnist_cnn_5x5x32_3x3x64",
+                          "tabular_autoencoder": "autoencoder_784_32_784",
+                          "resnet18_shape": "resnet18_shape_3x224x224"}[args.model],
+                "global_batch": mbs_eff * world,
+                "seq_len": None,
+                "parallelism": ("async-ps-%s dp%d"
+                                % ("lock" if args.lock else "hogwild", world)),
+            },
+        }
+        print(json.dumps(result))
+
+    engine.close()
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

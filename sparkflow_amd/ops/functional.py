@@ -102,14 +102,15 @@ class SoftmaxXent(torch.autograd.Function):
 
 
 def softmax_xent(logits, labels):
-    """labels: one-hot float [M,C] or class-index long [M]."""
-    if labels.dtype in (torch.int64, torch.int32):
+    """labels: one-hot [M,C], class-index [M], or class-index [M,1]
+    (the engine's ``handle_features`` wraps scalar labels to [y])."""
+    if labels.dim() == 2 and labels.size(1) == 1 and logits.size(1) != 1:
+        labels = labels.reshape(-1)
+    if labels.dim() == 1:
         onehot = torch.zeros_like(logits)
-        onehot.scatter_(1, labels.view(-1, 1).long(), 1.0)
+        onehot.scatter_(1, labels.long().view(-1, 1), 1.0)
         labels = onehot
-    elif labels.dim() == 1:
-        labels = labels.view(-1, 1)
-    return SoftmaxXent.apply(logits, labels.to(logits.dtype))
+    return SoftmaxXent.apply(logits, labels.to(logits.dtype).contiguous())
 
 
 class MseLoss(torch.autograd.Function):

@@ -1,0 +1,71 @@
+// Python bindings for the sparkflow_amd gfx950 kernel extension.
+#include <torch/extension.h>
+
+#include <vector>
+
+// gemm.hip
+torch::Tensor gemm_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                       int64_t act);
+torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w);
+torch::Tensor gemm_wgrad(torch::Tensor x, torch::Tensor dy);
+torch::Tensor bias_grad(torch::Tensor dy);
+
+// elementwise.hip
+torch::Tensor act_bwd(torch::Tensor y, torch::Tensor dy, int64_t act);
+torch::Tensor argmax_rows(torch::Tensor x);
+
+// losses.hip
+std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
+                                            torch::Tensor onehot);
+torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,
+                               double gscale);
+torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target);
+torch::Tensor mse_bwd(torch::Tensor pred, torch::Tensor target, double gscale);
+
+// optim.hip
+void fused_sgd(torch::Tensor p, torch::Tensor g, double lr);
+void fused_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor acc,
+                    double lr, double mom, bool nesterov);
+void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, double lr_t, double b1, double b2,
+                double eps);
+void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
+                   torch::Tensor mom, double lr, double decay,
+                   double momentum, double eps);
+
+// conv.hip
+torch::Tensor conv2d_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                         int64_t stride, int64_t pad, int64_t act);
+torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t H,
+                           int64_t W, int64_t stride, int64_t pad);
+torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
+                           int64_t Kw, int64_t stride, int64_t pad);
+torch::Tensor conv_bias_grad(torch::Tensor dy);
+std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
+                                         int64_t stride);
+torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
+                            int64_t W);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "sparkflow_amd gfx950 (CDNA4) kernels";
+  m.def("gemm_act", &gemm_act, "fused fp32 MFMA GEMM + bias + activation");
+  m.def("gemm_dgrad", &gemm_dgrad, "dx = dy @ W^T");
+  m.def("gemm_wgrad", &gemm_wgrad, "dW = x^T @ dy");
+  m.def("bias_grad", &bias_grad, "column-sum bias gradient");
+  m.def("act_bwd", &act_bwd, "activation backward from output");
+  m.def("argmax_rows", &argmax_rows, "rowwise argmax head");
+  m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax+xent fwd");
+  m.def("softmax_xent_bwd", &softmax_xent_bwd, "softmax+xent bwd");
+  m.def("mse_fwd", &mse_fwd, "fused MSE loss");
+  m.def("mse_bwd", &mse_bwd, "MSE grad");
+  m.def("fused_sgd", &fused_sgd, "single-launch SGD apply");
+  m.def("fused_momentum", &fused_momentum, "single-launch momentum apply");
+  m.def("fused_adam", &fused_adam, "single-launch Adam apply");
+  m.def("fused_rmsprop", &fused_rmsprop, "single-launch RMSProp apply");
+  m.def("conv2d_act", &conv2d_act, "conv2d + bias + activation fwd");
+  m.def("conv2d_dgrad", &conv2d_dgrad, "conv2d input grad");
+  m.def("conv2d_wgrad", &conv2d_wgrad, "conv2d weight grad");
+  m.def("conv_bias_grad", &conv_bias_grad, "conv bias grad");
+  m.def("maxpool2d_fwd", &maxpool2d_fwd, "maxpool fwd with argmax indices");
+  m.def("maxpool2d_bwd", &maxpool2d_bwd, "maxpool bwd scatter");
+}

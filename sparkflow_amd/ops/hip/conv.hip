@@ -1,0 +1,297 @@
+// conv2d (NCHW, fp32) + maxpool kernels — the reference's implicit TF
+// conv surface (reference examples/cnn_example.py:13-21).
+//
+// v1 strategy: direct convolution with fused bias+activation (one thread
+// per output element, grid-stride), weight-element-per-block reductions
+// for wgrad, argmax-index maxpool.  Correctness-first; the MFMA
+// implicit-GEMM path is the planned optimization for the ResNet-18-shape
+// config (BASELINE.json config 5).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+// ---- forward: y[n,co,ho,wo] = act(sum_{ci,kh,kw} x * w + b) ----
+__global__ void conv2d_fwd_kernel(const float* __restrict__ x,
+                                  const float* __restrict__ w,
+                                  const float* __restrict__ bias,
+                                  float* __restrict__ y, int N, int Cin,
+                                  int H, int W, int Cout, int Kh, int Kw,
+                                  int Ho, int Wo, int stride, int pad,
+                                  int act, int has_bias) {
+  const long total = (long)N * Cout * Ho * Wo;
+  const long gstride = (long)gridDim.x * blockDim.x;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += gstride) {
+    const int wo = idx % Wo;
+    const int ho = (idx / Wo) % Ho;
+    const int co = (idx / ((long)Wo * Ho)) % Cout;
+    const int n = idx / ((long)Wo * Ho * Cout);
+
+    float s = has_bias ? bias[co] : 0.0f;
+    const int h0 = ho * stride - pad;
+    const int w0 = wo * stride - pad;
+    for (int ci = 0; ci < Cin; ++ci) {
+      const float* xp = x + (((long)n * Cin + ci) * H) * W;
+      const float* wp = w + (((long)co * Cin + ci) * Kh) * Kw;
+      for (int kh = 0; kh < Kh; ++kh) {
+        const int h = h0 + kh;
+        if (h < 0 || h >= H) continue;
+        for (int kw = 0; kw < Kw; ++kw) {
+          const int ww = w0 + kw;
+          if (ww < 0 || ww >= W) continue;
+          s += xp[(long)h * W + ww] * wp[kh * Kw + kw];
+        }
+      }
+    }
+    y[idx] = act_apply(s, act);
+  }
+}
+
+// ---- dgrad: dx[n,ci,h,w] = sum_{co,kh,kw aligned} dy * w ----
+__global__ void conv2d_dgrad_kernel(const float* __restrict__ dy,
+                                    const float* __restrict__ w,
+                                    float* __restrict__ dx, int N, int Cin,
+                                    int H, int W, int Cout, int Kh, int Kw,
+                                    int Ho, int Wo, int stride, int pad) {
+  const long total = (long)N * Cin * H * W;
+  const long gstride = (long)gridDim.x * blockDim.x;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += gstride) {
+    const int ww = idx % W;
+    const int h = (idx / W) % H;
+    const int ci = (idx / ((long)W * H)) % Cin;
+    const int n = idx / ((long)W * H * Cin);
+
+    float s = 0.0f;
+    for (int kh = 0; kh < Kh; ++kh) {
+      const int hp = h + pad - kh;
+      if (hp < 0 || hp % stride) continue;
+      const int ho = hp / stride;
+      if (ho >= Ho) continue;
+      for (int kw = 0; kw < Kw; ++kw) {
+        const int wp = ww + pad - kw;
+        if (wp < 0 || wp % stride) continue;
+        const int wo = wp / stride;
+        if (wo >= Wo) continue;
+        for (int co = 0; co < Cout; ++co) {
+          s += dy[(((long)n * Cout + co) * Ho + ho) * Wo + wo] *
+               w[(((long)co * Cin + ci) * Kh + kh) * Kw + kw];
+        }
+      }
+    }
+    dx[idx] = s;
+  }
+}
+
+// ---- wgrad: one 256-thread block per filter element, reduction over
+// (n, ho, wo) staged through wave shuffles + LDS ----
+__global__ void conv2d_wgrad_kernel(const float* __restrict__ x,
+                                    const float* __restrict__ dy,
+                                    float* __restrict__ dw, int N, int Cin,
+                                    int H, int W, int Cout, int Kh, int Kw,
+                                    int Ho, int Wo, int stride, int pad) {
+  const long widx = blockIdx.x;  // (co, ci, kh, kw)
+  const int kw = widx % Kw;
+  const int kh = (widx / Kw) % Kh;
+  const int ci = (widx / (Kw * Kh)) % Cin;
+  const int co = widx / ((long)Kw * Kh * Cin);
+
+  const long total = (long)N * Ho * Wo;
+  float s = 0.0f;
+  for (long i = threadIdx.x; i < total; i += blockDim.x) {
+    const int wo = i % Wo;
+    const int ho = (i / Wo) % Ho;
+    const int n = i / ((long)Wo * Ho);
+    const int h = ho * stride - pad + kh;
+    const int ww = wo * stride - pad + kw;
+    if (h < 0 || h >= H || ww < 0 || ww >= W) continue;
+    s += x[(((long)n * Cin + ci) * H + h) * W + ww] *
+         dy[(((long)n * Cout + co) * Ho + ho) * Wo + wo];
+  }
+  s = wave_sum(s);
+  __shared__ float partials[4];
+  if ((threadIdx.x & (WAVE - 1)) == 0) partials[threadIdx.x / WAVE] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float b = 0.0f;
+    for (int i = 0; i < (int)(blockDim.x / WAVE); ++i) b += partials[i];
+    dw[widx] = b;
+  }
+}
+
+__global__ void conv_bias_grad_kernel(const float* __restrict__ dy,
+                                      float* __restrict__ db, int N, int C,
+                                      int HW) {
+  const int c = blockIdx.x;
+  float s = 0.0f;
+  for (long i = threadIdx.x; i < (long)N * HW; i += blockDim.x) {
+    const int hw = i % HW;
+    const int n = i / HW;
+    s += dy[((long)n * C + c) * HW + hw];
+  }
+  s = wave_sum(s);
+  __shared__ float partials[4];
+  if ((threadIdx.x & (WAVE - 1)) == 0) partials[threadIdx.x / WAVE] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float b = 0.0f;
+    for (int i = 0; i < (int)(blockDim.x / WAVE); ++i) b += partials[i];
+    db[c] = b;
+  }
+}
+
+// ---- maxpool with argmax indices (bwd scatters through them) ----
+__global__ void maxpool_fwd_kernel(const float* __restrict__ x,
+                                   float* __restrict__ y,
+                                   long* __restrict__ idx, int N, int C,
+                                   int H, int W, int K, int stride, int Ho,
+                                   int Wo) {
+  const long total = (long)N * C * Ho * Wo;
+  const long gstride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gstride) {
+    const int wo = i % Wo;
+    const int ho = (i / Wo) % Ho;
+    const int c = (i / ((long)Wo * Ho)) % C;
+    const int n = i / ((long)Wo * Ho * C);
+    const float* xp = x + ((long)n * C + c) * H * W;
+    float best = -INFINITY;
+    long best_hw = 0;
+    for (int kh = 0; kh < K; ++kh) {
+      const int h = ho * stride + kh;
+      if (h >= H) break;
+      for (int kw = 0; kw < K; ++kw) {
+        const int ww = wo * stride + kw;
+        if (ww >= W) break;
+        const float v = xp[(long)h * W + ww];
+        if (v > best) {
+          best = v;
+          best_hw = (long)h * W + ww;
+        }
+      }
+    }
+    y[i] = best;
+    idx[i] = best_hw;  // torch max_pool2d index convention: within the plane
+  }
+}
+
+__global__ void maxpool_bwd_kernel(const float* __restrict__ dy,
+                                   const long* __restrict__ idx,
+                                   float* __restrict__ dx, int N, int C,
+                                   int HW, int HoWo) {
+  const long total = (long)N * C * HoWo;
+  const long gstride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gstride) {
+    const int c = (i / HoWo) % C;
+    const int n = i / ((long)HoWo * C);
+    atomicAdd(&dx[((long)n * C + c) * HW + idx[i]], dy[i]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+
+static inline void check_f32c(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                  t.is_contiguous(),
+              name, " must be contiguous fp32 on GPU");
+}
+
+static inline int out_dim(int in, int k, int stride, int pad) {
+  return (in + 2 * pad - k) / stride + 1;
+}
+
+torch::Tensor conv2d_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                         int64_t stride, int64_t pad, int64_t act) {
+  check_f32c(x, "x");
+  check_f32c(w, "w");
+  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = w.size(0), Kh = w.size(2), Kw = w.size(3);
+  TORCH_CHECK(w.size(1) == Cin, "conv channel mismatch");
+  const int Ho = out_dim(H, Kh, stride, pad), Wo = out_dim(W, Kw, stride, pad);
+  auto y = torch::empty({N, Cout, Ho, Wo}, x.options());
+  const long total = y.numel();
+  const int blocks = (int)std::min<long>(8192, (total + 255) / 256);
+  hipLaunchKernelGGL(conv2d_fwd_kernel, dim3(blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
+                     w.data_ptr<float>(),
+                     bias.numel() ? bias.data_ptr<float>() : nullptr,
+                     y.data_ptr<float>(), N, Cin, H, W, Cout, Kh, Kw, Ho, Wo,
+                     (int)stride, (int)pad, (int)act, bias.numel() > 0);
+  return y;
+}
+
+torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t H,
+                           int64_t W, int64_t stride, int64_t pad) {
+  check_f32c(dy, "dy");
+  check_f32c(w, "w");
+  const int N = dy.size(0), Cout = dy.size(1), Ho = dy.size(2),
+            Wo = dy.size(3);
+  const int Cin = w.size(1), Kh = w.size(2), Kw = w.size(3);
+  auto dx = torch::empty({N, Cin, H, W}, dy.options());
+  const long total = dx.numel();
+  const int blocks = (int)std::min<long>(8192, (total + 255) / 256);
+  hipLaunchKernelGGL(conv2d_dgrad_kernel, dim3(blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
+                     w.data_ptr<float>(), dx.data_ptr<float>(), N, Cin,
+                     (int)H, (int)W, Cout, Kh, Kw, Ho, Wo, (int)stride,
+                     (int)pad);
+  return dx;
+}
+
+torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
+                           int64_t Kw, int64_t stride, int64_t pad) {
+  check_f32c(x, "x");
+  check_f32c(dy, "dy");
+  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = dy.size(1), Ho = dy.size(2), Wo = dy.size(3);
+  auto dw = torch::empty({Cout, Cin, Kh, Kw}, x.options());
+  hipLaunchKernelGGL(conv2d_wgrad_kernel, dim3((int)dw.numel()), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
+                     dy.data_ptr<float>(), dw.data_ptr<float>(), N, Cin, H, W,
+                     Cout, (int)Kh, (int)Kw, Ho, Wo, (int)stride, (int)pad);
+  return dw;
+}
+
+torch::Tensor conv_bias_grad(torch::Tensor dy) {
+  check_f32c(dy, "dy");
+  const int N = dy.size(0), C = dy.size(1), HW = dy.size(2) * dy.size(3);
+  auto db = torch::empty({C}, dy.options());
+  hipLaunchKernelGGL(conv_bias_grad_kernel, dim3(C), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
+                     db.data_ptr<float>(), N, C, HW);
+  return db;
+}
+
+std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
+                                         int64_t stride) {
+  check_f32c(x, "x");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int Ho = (H - k) / stride + 1, Wo = (W - k) / stride + 1;
+  auto y = torch::empty({N, C, Ho, Wo}, x.options());
+  auto idx = torch::empty({N, C, Ho, Wo},
+                          x.options().dtype(torch::kInt64));
+  const long total = y.numel();
+  const int blocks = (int)std::min<long>(8192, (total + 255) / 256);
+  hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
+                     y.data_ptr<float>(), idx.data_ptr<long>(), N, C, H, W,
+                     (int)k, (int)stride, Ho, Wo);
+  return {y, idx};
+}
+
+torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
+                            int64_t W) {
+  check_f32c(dy, "dy");
+  const int N = dy.size(0), C = dy.size(1);
+  const int HoWo = dy.size(2) * dy.size(3);
+  auto dx = torch::zeros({N, C, H, W}, dy.options());
+  const long total = dy.numel();
+  const int blocks = (int)std::min<long>(8192, (total + 255) / 256);
+  hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
+                     idx.data_ptr<long>(), dx.data_ptr<float>(), N, C,
+                     (int)(H * W), HoWo);
+  return dx;
+}

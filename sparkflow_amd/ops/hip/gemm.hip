@@ -1,0 +1,244 @@
+// fp32 MFMA GEMM with fused bias + activation epilogue, plus the dgrad /
+// wgrad / bias-grad companions — the MI355X replacement for the
+// reference's TF dense layers (reference examples/simple_dnn.py:16-18;
+// SURVEY.md §2.2 "Dense layer: GEMM + bias add").
+//
+// Design: 64x64 output tile per 256-thread workgroup (4 waves, each a
+// 32x32 subtile of 2x2 v_mfma_f32_16x16x4_f32 fragments — exact fp32 at
+// the 157 TF matrix rate, cdna_hip_programming.md §3), K staged through
+// LDS in 16-deep chunks, bias+activation fused into the accumulator
+// store.  Transpose handling for the backward GEMMs is done at the LDS
+// staging step (coalesced global reads either way), so dgrad
+// (dy @ W^T) and wgrad (x^T @ dy) reuse the same MFMA pipeline.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+#define BM 64
+#define BN 64
+#define BK 16
+
+// Logical C[M,N] = act(A'[M,K] @ B'[K,N] + bias).
+// TA=0: A' = A_phys[M,K] (lda=K);  TA=1: A' = A_phys[K,M]^T (lda=M).
+// TB=0: B' = B_phys[K,N] (ldb=N);  TB=1: B' = B_phys[N,K]^T (ldb=K).
+template <bool TA, bool TB, bool HAS_BIAS>
+__global__ __launch_bounds__(256) void gemm_f32_kernel(
+    const float* __restrict__ A, const float* __restrict__ B,
+    const float* __restrict__ bias, float* __restrict__ C, int M, int N,
+    int K, int lda, int ldb, int act) {
+  __shared__ float As[BM][BK + 1];
+  __shared__ float Bs[BK][BN + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int w = t >> 6;       // wave 0..3
+  const int wm = (w >> 1) * 32;
+  const int wn = (w & 1) * 32;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+
+  f32x4 acc[2][2] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A tile: As[m][k] = A'(m0+m, k0+k) ----
+    if (!TA) {
+      const int row = t >> 2, c4 = (t & 3) * 4;  // 64 rows x 4 float4
+      const int gm = m0 + row;
+      if (gm < M && k0 + BK <= K) {
+        const float4 v = *reinterpret_cast<const float4*>(
+            &A[(long)gm * lda + k0 + c4]);
+        As[row][c4 + 0] = v.x;
+        As[row][c4 + 1] = v.y;
+        As[row][c4 + 2] = v.z;
+        As[row][c4 + 3] = v.w;
+      } else {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int gk = k0 + c4 + i;
+          As[row][c4 + i] =
+              (gm < M && gk < K) ? A[(long)gm * lda + gk] : 0.0f;
+        }
+      }
+    } else {
+      const int kk = t >> 4, m4 = (t & 15) * 4;  // 16 k-rows x 16 float4
+      const int gk = k0 + kk;
+      if (gk < K && m0 + BM <= M) {
+        const float4 v = *reinterpret_cast<const float4*>(
+            &A[(long)gk * lda + m0 + m4]);
+        As[m4 + 0][kk] = v.x;
+        As[m4 + 1][kk] = v.y;
+        As[m4 + 2][kk] = v.z;
+        As[m4 + 3][kk] = v.w;
+      } else {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int gm = m0 + m4 + i;
+          As[m4 + i][kk] =
+              (gk < K && gm < M) ? A[(long)gk * lda + gm] : 0.0f;
+        }
+      }
+    }
+
+    // ---- stage B tile: Bs[k][n] = B'(k0+k, n0+n) ----
+    if (!TB) {
+      const int kk = t >> 4, n4 = (t & 15) * 4;  // 16 k-rows x 16 float4
+      const int gk = k0 + kk;
+      if (gk < K && n0 + BN <= N) {
+        const float4 v = *reinterpret_cast<const float4*>(
+            &B[(long)gk * ldb + n0 + n4]);
+        Bs[kk][n4 + 0] = v.x;
+        Bs[kk][n4 + 1] = v.y;
+        Bs[kk][n4 + 2] = v.z;
+        Bs[kk][n4 + 3] = v.w;
+      } else {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int gn = n0 + n4 + i;
+          Bs[kk][n4 + i] =
+              (gk < K && gn < N) ? B[(long)gk * ldb + gn] : 0.0f;
+        }
+      }
+    } else {
+      const int row = t >> 2, c4 = (t & 3) * 4;  // 64 n-rows x 4 float4
+      const int gn = n0 + row;
+      if (gn < N && k0 + BK <= K) {
+        const float4 v = *reinterpret_cast<const float4*>(
+            &B[(long)gn * ldb + k0 + c4]);
+        Bs[c4 + 0][row] = v.x;
+        Bs[c4 + 1][row] = v.y;
+        Bs[c4 + 2][row] = v.z;
+        Bs[c4 + 3][row] = v.w;
+      } else {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int gk = k0 + c4 + i;
+          Bs[c4 + i][row] =
+              (gn < N && gk < K) ? B[(long)gn * ldb + gk] : 0.0f;
+        }
+      }
+    }
+
+    __syncthreads();
+
+    // ---- 16x16x4 MFMA over the 16-deep chunk ----
+    const int fr = lane & 15;   // fragment row/col position
+    const int fk = lane >> 4;   // k sub-index 0..3
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      const float a0 = As[wm + fr][kk * 4 + fk];
+      const float a1 = As[wm + 16 + fr][kk * 4 + fk];
+      const float b0 = Bs[kk * 4 + fk][wn + fr];
+      const float b1 = Bs[kk * 4 + fk][wn + 16 + fr];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: bias + activation fused into the store ----
+  // C/D fragment map (16x16 shapes): col = lane&15, row = (lane>>4)*4 + i.
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int col = n0 + wn + ni * 16 + fcol;
+      if (col >= N) continue;
+      const float bv = HAS_BIAS ? bias[col] : 0.0f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = m0 + wm + mi * 16 + frow + i;
+        if (row < M) {
+          C[(long)row * N + col] = act_apply(acc[mi][ni][i] + bv, act);
+        }
+      }
+    }
+  }
+}
+
+__global__ void bias_grad_kernel(const float* __restrict__ dy,
+                                 float* __restrict__ db, int M, int N) {
+  const int n = blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= N) return;
+  float s = 0.0f;
+  for (int m = 0; m < M; ++m) s += dy[(long)m * N + n];
+  db[n] = s;
+}
+
+// ---------------------------------------------------------------------------
+
+static inline void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+template <bool TA, bool TB>
+static void launch_gemm(const torch::Tensor& A, const torch::Tensor& B,
+                        const torch::Tensor& bias, torch::Tensor& C, int M,
+                        int N, int K, int lda, int ldb, int act) {
+  dim3 grid(ceil_div(N, BN), ceil_div(M, BM));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bool has_bias = bias.numel() > 0;
+  if (has_bias) {
+    hipLaunchKernelGGL((gemm_f32_kernel<TA, TB, true>), grid, dim3(256), 0,
+                       stream, A.data_ptr<float>(), B.data_ptr<float>(),
+                       bias.data_ptr<float>(), C.data_ptr<float>(), M, N, K,
+                       lda, ldb, act);
+  } else {
+    hipLaunchKernelGGL((gemm_f32_kernel<TA, TB, false>), grid, dim3(256), 0,
+                       stream, A.data_ptr<float>(), B.data_ptr<float>(),
+                       nullptr, C.data_ptr<float>(), M, N, K, lda, ldb, act);
+  }
+}
+
+torch::Tensor gemm_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                       int64_t act) {
+  check_f32(x, "x");
+  check_f32(w, "w");
+  const int M = x.size(0), K = x.size(1), N = w.size(1);
+  TORCH_CHECK(w.size(0) == K, "shape mismatch x@w");
+  auto y = torch::empty({M, N}, x.options());
+  launch_gemm<false, false>(x, w, bias, y, M, N, K, K, N, (int)act);
+  return y;
+}
+
+torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w) {
+  check_f32(dy, "dy");
+  check_f32(w, "w");
+  const int M = dy.size(0), Nout = dy.size(1), Kin = w.size(0);
+  TORCH_CHECK(w.size(1) == Nout, "shape mismatch dgrad");
+  auto dx = torch::empty({M, Kin}, dy.options());
+  auto empty = torch::empty({0}, dy.options());
+  // dx[M,Kin] = dy[M,Nout] @ w^T  (B' = w^T via TB=1)
+  launch_gemm<false, true>(dy, w, empty, dx, M, Kin, Nout, Nout, Nout,
+                           ACT_NONE);
+  return dx;
+}
+
+torch::Tensor gemm_wgrad(torch::Tensor x, torch::Tensor dy) {
+  check_f32(x, "x");
+  check_f32(dy, "dy");
+  const int Mb = x.size(0), Kin = x.size(1), Nout = dy.size(1);
+  TORCH_CHECK(dy.size(0) == Mb, "shape mismatch wgrad");
+  auto dw = torch::empty({Kin, Nout}, x.options());
+  auto empty = torch::empty({0}, x.options());
+  // dw[Kin,Nout] = x^T @ dy  (A' = x^T via TA=1)
+  launch_gemm<true, false>(x, dy, empty, dw, Kin, Nout, Mb, Kin, Nout,
+                           ACT_NONE);
+  return dw;
+}
+
+torch::Tensor bias_grad(torch::Tensor dy) {
+  check_f32(dy, "dy");
+  const int M = dy.size(0), N = dy.size(1);
+  auto db = torch::empty({N}, dy.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(bias_grad_kernel, dim3(ceil_div(N, 256)), dim3(256), 0,
+                     stream, dy.data_ptr<float>(), db.data_ptr<float>(), M, N);
+  return db;
+}

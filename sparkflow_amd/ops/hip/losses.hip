@@ -1,0 +1,151 @@
+// Fused loss kernels (reference SURVEY.md §2.2: softmax_cross_entropy and
+// mean_squared_error fwd+bwd as fused CDNA4 kernels).
+//
+// softmax-xent forward: one wave per row computes max / sum-exp with
+// wave-wide __shfl_xor reductions (no LDS round trip), writes the probs
+// tile (saved for backward) and atomically accumulates the summed loss —
+// one kernel for softmax + xent + loss reduction.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+__global__ void softmax_xent_fwd_kernel(const float* __restrict__ logits,
+                                        const float* __restrict__ onehot,
+                                        float* __restrict__ probs,
+                                        float* __restrict__ loss_sum, int M,
+                                        int C) {
+  const int row = blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (row >= M) return;
+  const float* xr = logits + (long)row * C;
+  const float* yr = onehot + (long)row * C;
+
+  float mx = -INFINITY;
+  for (int c = lane; c < C; c += WAVE) mx = fmaxf(mx, xr[c]);
+  mx = wave_max(mx);
+
+  float se = 0.0f, dot = 0.0f;
+  for (int c = lane; c < C; c += WAVE) {
+    const float e = __expf(xr[c] - mx);
+    se += e;
+    dot += yr[c] * xr[c];
+  }
+  se = wave_sum(se);
+  dot = wave_sum(dot);
+
+  const float inv = 1.0f / se;
+  for (int c = lane; c < C; c += WAVE) {
+    probs[(long)row * C + c] = __expf(xr[c] - mx) * inv;
+  }
+  if (lane == 0) {
+    const float lse = mx + __logf(se);
+    atomicAdd(loss_sum, lse - dot);
+  }
+}
+
+__global__ void softmax_xent_bwd_kernel(const float* __restrict__ probs,
+                                        const float* __restrict__ onehot,
+                                        float* __restrict__ dlogits,
+                                        float gscale, long n) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    dlogits[i] = (probs[i] - onehot[i]) * gscale;
+  }
+}
+
+__global__ void mse_fwd_kernel(const float* __restrict__ p,
+                               const float* __restrict__ t,
+                               float* __restrict__ out, long n) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  float s = 0.0f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    const float d = p[i] - t[i];
+    s += d * d;
+  }
+  s = wave_sum(s);
+  __shared__ float partials[4];
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) partials[wid] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float b = 0.0f;
+    for (int i = 0; i < (int)(blockDim.x / WAVE); ++i) b += partials[i];
+    atomicAdd(out, b);
+  }
+}
+
+__global__ void mse_bwd_kernel(const float* __restrict__ p,
+                               const float* __restrict__ t,
+                               float* __restrict__ dp, float gscale, long n) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    dp[i] = (p[i] - t[i]) * gscale;
+  }
+}
+
+static inline void check_f32l(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                  t.is_contiguous(),
+              name, " must be contiguous fp32 on GPU");
+}
+
+std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
+                                            torch::Tensor onehot) {
+  check_f32l(logits, "logits");
+  check_f32l(onehot, "onehot");
+  const int M = logits.size(0), C = logits.size(1);
+  auto probs = torch::empty_like(logits);
+  auto loss = torch::zeros({}, logits.options());
+  const int wpb = 4;
+  hipLaunchKernelGGL(softmax_xent_fwd_kernel, dim3(ceil_div(M, wpb)),
+                     dim3(wpb * WAVE), 0, at::cuda::getCurrentCUDAStream(),
+                     logits.data_ptr<float>(), onehot.data_ptr<float>(),
+                     probs.data_ptr<float>(), loss.data_ptr<float>(), M, C);
+  loss.div_((double)M);  // mean over rows
+  return {loss, probs};
+}
+
+torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,
+                               double gscale) {
+  check_f32l(probs, "probs");
+  check_f32l(onehot, "onehot");
+  auto d = torch::empty_like(probs);
+  const long n = probs.numel();
+  const int blocks = (int)std::min<long>(4096, (n + 255) / 256);
+  hipLaunchKernelGGL(softmax_xent_bwd_kernel, dim3(blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     probs.data_ptr<float>(), onehot.data_ptr<float>(),
+                     d.data_ptr<float>(), (float)gscale, n);
+  return d;
+}
+
+torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target) {
+  check_f32l(pred, "pred");
+  check_f32l(target, "target");
+  auto out = torch::zeros({}, pred.options());
+  const long n = pred.numel();
+  const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
+  hipLaunchKernelGGL(mse_fwd_kernel, dim3(blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), pred.data_ptr<float>(),
+                     target.data_ptr<float>(), out.data_ptr<float>(), n);
+  out.div_((double)n);
+  return out;
+}
+
+torch::Tensor mse_bwd(torch::Tensor pred, torch::Tensor target,
+                      double gscale) {
+  check_f32l(pred, "pred");
+  check_f32l(target, "target");
+  auto d = torch::empty_like(pred);
+  const long n = pred.numel();
+  const int blocks = (int)std::min<long>(4096, (n + 255) / 256);
+  hipLaunchKernelGGL(mse_bwd_kernel, dim3(blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), pred.data_ptr<float>(),
+                     target.data_ptr<float>(), d.data_ptr<float>(),
+                     (float)gscale, n);
+  return d;
+}

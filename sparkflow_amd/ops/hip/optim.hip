@@ -1,0 +1,128 @@
+// Fused single-launch optimizer applies over the PS's flat fp32 weight
+// buffer (the north-star "fused multi-tensor SGD and Adam HIP kernels on
+// the rank-0 PS GPU" — BASELINE.json; reference applies TF optimizers per
+// incoming POST, HogwildSparkModel.py:219-242).
+//
+// Because the canonical weights are ONE contiguous tensor, a whole-model
+// apply is a single grid-stride float4 sweep: one kernel launch per
+// gradient push regardless of how many variables the model has (vs ~5
+// launches for a torch optimizer even on a single tensor).  These kernels
+// are what makes concurrent Hogwild applies cheap: each PS service thread
+// enqueues one launch on its own HIP stream.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+#define GS_LOOP(i, n)                                                         \
+  const long _stride = (long)gridDim.x * blockDim.x;                          \
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;           \
+       i += _stride)
+
+__global__ void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
+                           float lr, long n) {
+  GS_LOOP(i, n) { p[i] -= lr * g[i]; }
+}
+
+__global__ void momentum_kernel(float* __restrict__ p,
+                                const float* __restrict__ g,
+                                float* __restrict__ acc, float lr, float mom,
+                                int nesterov, long n) {
+  GS_LOOP(i, n) {
+    const float a = mom * acc[i] + g[i];
+    acc[i] = a;
+    p[i] -= nesterov ? lr * (g[i] + mom * a) : lr * a;
+  }
+}
+
+__global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v,
+                            float lr_t, float b1, float b2, float eps,
+                            long n) {
+  GS_LOOP(i, n) {
+    const float gi = g[i];
+    const float mi = b1 * m[i] + (1.0f - b1) * gi;
+    const float vi = b2 * v[i] + (1.0f - b2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    p[i] -= lr_t * mi / (__fsqrt_rn(vi) + eps);
+  }
+}
+
+__global__ void rmsprop_kernel(float* __restrict__ p,
+                               const float* __restrict__ g,
+                               float* __restrict__ ms,
+                               float* __restrict__ mom, float lr, float decay,
+                               float momentum, float eps, long n) {
+  GS_LOOP(i, n) {
+    const float gi = g[i];
+    const float m2 = decay * ms[i] + (1.0f - decay) * gi * gi;
+    ms[i] = m2;
+    // TF rmsprop: epsilon inside the sqrt
+    const float mo = momentum * mom[i] + lr * gi / __fsqrt_rn(m2 + eps);
+    mom[i] = mo;
+    p[i] -= mo;
+  }
+}
+
+static inline void check_flat(const torch::Tensor& t, const char* name,
+                              long n) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                  t.is_contiguous() && t.numel() == n,
+              name, " must be contiguous fp32 on GPU with matching numel");
+}
+
+static inline int opt_blocks(long n) {
+  return (int)std::min<long>(2048, (n + 255) / 256);
+}
+
+void fused_sgd(torch::Tensor p, torch::Tensor g, double lr) {
+  const long n = p.numel();
+  check_flat(p, "p", n);
+  check_flat(g, "g", n);
+  hipLaunchKernelGGL(sgd_kernel, dim3(opt_blocks(n)), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), p.data_ptr<float>(),
+                     g.data_ptr<float>(), (float)lr, n);
+}
+
+void fused_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor acc,
+                    double lr, double mom, bool nesterov) {
+  const long n = p.numel();
+  check_flat(p, "p", n);
+  check_flat(g, "g", n);
+  check_flat(acc, "acc", n);
+  hipLaunchKernelGGL(momentum_kernel, dim3(opt_blocks(n)), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), p.data_ptr<float>(),
+                     g.data_ptr<float>(), acc.data_ptr<float>(), (float)lr,
+                     (float)mom, (int)nesterov, n);
+}
+
+void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, double lr_t, double b1, double b2,
+                double eps) {
+  const long n = p.numel();
+  check_flat(p, "p", n);
+  check_flat(g, "g", n);
+  check_flat(m, "m", n);
+  check_flat(v, "v", n);
+  hipLaunchKernelGGL(adam_kernel, dim3(opt_blocks(n)), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), p.data_ptr<float>(),
+                     g.data_ptr<float>(), m.data_ptr<float>(),
+                     v.data_ptr<float>(), (float)lr_t, (float)b1, (float)b2,
+                     (float)eps, n);
+}
+
+void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
+                   torch::Tensor mom, double lr, double decay,
+                   double momentum, double eps) {
+  const long n = p.numel();
+  check_flat(p, "p", n);
+  check_flat(g, "g", n);
+  check_flat(ms, "ms", n);
+  check_flat(mom, "mom", n);
+  hipLaunchKernelGGL(rmsprop_kernel, dim3(opt_blocks(n)), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), p.data_ptr<float>(),
+                     g.data_ptr<float>(), ms.data_ptr<float>(),
+                     mom.data_ptr<float>(), (float)lr, (float)decay,
+                     (float)momentum, (float)eps, n);
+}
