@@ -101,15 +101,43 @@ class SoftmaxXent(torch.autograd.Function):
         return dlogits, None
 
 
+class SoftmaxXentIdx(torch.autograd.Function):
+    """Class-index label path — no one-hot materialization (the fused
+    kernel reads int64 labels directly)."""
+
+    @staticmethod
+    def forward(ctx, logits, labels):
+        if ops.use_hip(logits):
+            loss, probs = ops.hip.softmax_xent_idx_fwd(logits, labels)
+        else:
+            lse = torch.logsumexp(logits, dim=1, keepdim=True)
+            logp = logits - lse
+            loss = -logp.gather(1, labels.view(-1, 1)).mean()
+            probs = torch.exp(logp)
+        ctx.save_for_backward(probs, labels)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        probs, labels = ctx.saved_tensors
+        m = probs.shape[0]
+        if ops.use_hip(probs):
+            dlogits = ops.hip.softmax_xent_idx_bwd(probs, labels,
+                                                   float(dloss) / m)
+        else:
+            dlogits = probs.clone()
+            dlogits[torch.arange(m, device=probs.device), labels] -= 1.0
+            dlogits.mul_(dloss / m)
+        return dlogits, None
+
+
 def softmax_xent(logits, labels):
     """labels: one-hot [M,C], class-index [M], or class-index [M,1]
     (the engine's ``handle_features`` wraps scalar labels to [y])."""
     if labels.dim() == 2 and labels.size(1) == 1 and logits.size(1) != 1:
         labels = labels.reshape(-1)
     if labels.dim() == 1:
-        onehot = torch.zeros_like(logits)
-        onehot.scatter_(1, labels.long().view(-1, 1), 1.0)
-        labels = onehot
+        return SoftmaxXentIdx.apply(logits, labels.long().contiguous())
     return SoftmaxXent.apply(logits, labels.to(logits.dtype).contiguous())
 
 

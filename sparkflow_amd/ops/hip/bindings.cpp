@@ -19,6 +19,10 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
                                             torch::Tensor onehot);
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,
                                double gscale);
+std::vector<torch::Tensor> softmax_xent_idx_fwd(torch::Tensor logits,
+                                                torch::Tensor labels);
+torch::Tensor softmax_xent_idx_bwd(torch::Tensor probs, torch::Tensor labels,
+                                   double gscale);
 torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target);
 torch::Tensor mse_bwd(torch::Tensor pred, torch::Tensor target, double gscale);
 
@@ -56,6 +60,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("argmax_rows", &argmax_rows, "rowwise argmax head");
   m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax+xent fwd");
   m.def("softmax_xent_bwd", &softmax_xent_bwd, "softmax+xent bwd");
+  m.def("softmax_xent_idx_fwd", &softmax_xent_idx_fwd,
+        "fused softmax+xent fwd, class-index labels");
+  m.def("softmax_xent_idx_bwd", &softmax_xent_idx_bwd,
+        "softmax+xent bwd, class-index labels");
   m.def("mse_fwd", &mse_fwd, "fused MSE loss");
   m.def("mse_bwd", &mse_bwd, "MSE grad");
   m.def("fused_sgd", &fused_sgd, "single-launch SGD apply");

@@ -160,13 +160,23 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
   }
 }
 
+// Column sums with 2D parallelism: blockIdx.y owns a row chunk so narrow
+// layers (N=10..256) still fill more than one CU; partial sums combine
+// with one atomicAdd per (col, chunk).  db must arrive zeroed.
+#define BG_ROWS 32
 __global__ void bias_grad_kernel(const float* __restrict__ dy,
                                  float* __restrict__ db, int M, int N) {
   const int n = blockIdx.x * blockDim.x + threadIdx.x;
   if (n >= N) return;
+  const int m0 = blockIdx.y * BG_ROWS;
+  const int m1 = min(M, m0 + BG_ROWS);
   float s = 0.0f;
-  for (int m = 0; m < M; ++m) s += dy[(long)m * N + n];
-  db[n] = s;
+  for (int m = m0; m < m1; ++m) s += dy[(long)m * N + n];
+  if (gridDim.y == 1) {
+    db[n] = s;
+  } else {
+    atomicAdd(&db[n], s);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -236,9 +246,12 @@ torch::Tensor gemm_wgrad(torch::Tensor x, torch::Tensor dy) {
 torch::Tensor bias_grad(torch::Tensor dy) {
   check_f32(dy, "dy");
   const int M = dy.size(0), N = dy.size(1);
-  auto db = torch::empty({N}, dy.options());
+  const int chunks = ceil_div(M, BG_ROWS);
+  auto db = chunks > 1 ? torch::zeros({N}, dy.options())
+                       : torch::empty({N}, dy.options());
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(bias_grad_kernel, dim3(ceil_div(N, 256)), dim3(256), 0,
-                     stream, dy.data_ptr<float>(), db.data_ptr<float>(), M, N);
+  hipLaunchKernelGGL(bias_grad_kernel, dim3(ceil_div(N, 256), chunks),
+                     dim3(256), 0, stream, dy.data_ptr<float>(),
+                     db.data_ptr<float>(), M, N);
   return db;
 }

@@ -44,6 +44,51 @@ __global__ void softmax_xent_fwd_kernel(const float* __restrict__ logits,
   }
 }
 
+// Class-index label variant: skips the host-side one-hot scatter entirely
+// (labels arrive as int64 [M], the engine's natural label form).
+__global__ void softmax_xent_idx_fwd_kernel(const float* __restrict__ logits,
+                                            const long* __restrict__ labels,
+                                            float* __restrict__ probs,
+                                            float* __restrict__ loss_sum,
+                                            int M, int C) {
+  const int row = blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (row >= M) return;
+  const float* xr = logits + (long)row * C;
+
+  float mx = -INFINITY;
+  for (int c = lane; c < C; c += WAVE) mx = fmaxf(mx, xr[c]);
+  mx = wave_max(mx);
+
+  float se = 0.0f;
+  for (int c = lane; c < C; c += WAVE) se += __expf(xr[c] - mx);
+  se = wave_sum(se);
+
+  const float inv = 1.0f / se;
+  for (int c = lane; c < C; c += WAVE) {
+    probs[(long)row * C + c] = __expf(xr[c] - mx) * inv;
+  }
+  if (lane == 0) {
+    const float lse = mx + __logf(se);
+    atomicAdd(loss_sum, lse - xr[labels[row]]);
+  }
+}
+
+__global__ void softmax_xent_idx_bwd_kernel(const float* __restrict__ probs,
+                                            const long* __restrict__ labels,
+                                            float* __restrict__ dlogits,
+                                            float gscale, int M, int C) {
+  const long n = (long)M * C;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    const int c = i % C;
+    const int row = i / C;
+    const float onehot = (c == (int)labels[row]) ? 1.0f : 0.0f;
+    dlogits[i] = (probs[i] - onehot) * gscale;
+  }
+}
+
 __global__ void softmax_xent_bwd_kernel(const float* __restrict__ probs,
                                         const float* __restrict__ onehot,
                                         float* __restrict__ dlogits,
@@ -107,6 +152,38 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
                      probs.data_ptr<float>(), loss.data_ptr<float>(), M, C);
   loss.div_((double)M);  // mean over rows
   return {loss, probs};
+}
+
+std::vector<torch::Tensor> softmax_xent_idx_fwd(torch::Tensor logits,
+                                                torch::Tensor labels) {
+  check_f32l(logits, "logits");
+  TORCH_CHECK(labels.is_cuda() && labels.scalar_type() == torch::kInt64 &&
+                  labels.is_contiguous(),
+              "labels must be contiguous int64 on GPU");
+  const int M = logits.size(0), C = logits.size(1);
+  auto probs = torch::empty_like(logits);
+  auto loss = torch::zeros({}, logits.options());
+  const int wpb = 4;
+  hipLaunchKernelGGL(softmax_xent_idx_fwd_kernel, dim3(ceil_div(M, wpb)),
+                     dim3(wpb * WAVE), 0, at::cuda::getCurrentCUDAStream(),
+                     logits.data_ptr<float>(), labels.data_ptr<long>(),
+                     probs.data_ptr<float>(), loss.data_ptr<float>(), M, C);
+  loss.div_((double)M);
+  return {loss, probs};
+}
+
+torch::Tensor softmax_xent_idx_bwd(torch::Tensor probs, torch::Tensor labels,
+                                   double gscale) {
+  check_f32l(probs, "probs");
+  const int M = probs.size(0), C = probs.size(1);
+  auto d = torch::empty_like(probs);
+  const long n = probs.numel();
+  const int blocks = (int)std::min<long>(4096, (n + 255) / 256);
+  hipLaunchKernelGGL(softmax_xent_idx_bwd_kernel, dim3(blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     probs.data_ptr<float>(), labels.data_ptr<long>(),
+                     d.data_ptr<float>(), (float)gscale, M, C);
+  return d;
 }
 
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,

@@ -52,6 +52,12 @@ class LocalTransport(WorkerTransport):
         self.device = torch.device(device)
 
     def pull(self) -> torch.Tensor:
+        if not self.ps.acquire_lock and self.ps.device == self.device:
+            # Hogwild same-device fast path: hand out the live HBM buffer;
+            # the worker copies it into its own flat buffer (the racy
+            # read IS the reference's lock-free /parameters semantics).
+            self.ps.pull_count += 1
+            return self.ps.weights
         return self.ps.read_weights().to(self.device, non_blocking=True)
 
     def push(self, flat_grads: torch.Tensor) -> None:

@@ -28,8 +28,8 @@ from sparkflow_amd.parallel.ps import TrainConfig
 from sparkflow_amd.parallel.transport import WorkerTransport
 from sparkflow_amd.utils.serialization import (
     module_from_json,
-    pack_grads,
-    unpack_params,
+    pack_params,
+    param_list,
 )
 
 
@@ -44,8 +44,22 @@ class Worker:
         self.loss_callback = loss_callback
         self.partition_id = uuid.uuid4().hex
         self.rng = np.random.default_rng(config.seed)
-        self._grad_buf = None
         self.steps_done = 0  # pushes performed (for benchmarking)
+
+        # Params and grads live as VIEWS into two flat fp32 buffers: a
+        # weight pull is ONE device copy into flat_params and gradient
+        # packing is free (autograd accumulates straight into flat_grads)
+        # — vs one copy per tensor per step in the reference's
+        # set-weights/collect-grads loops (ml_util.py:16-28).
+        self.flat_params = pack_params(self.module)
+        self.flat_grads = torch.zeros_like(self.flat_params)
+        off = 0
+        with torch.no_grad():
+            for p in param_list(self.module):
+                n = p.numel()
+                p.data = self.flat_params[off:off + n].view(p.shape)
+                p.grad = self.flat_grads[off:off + n].view(p.shape)
+                off += n
 
     # -- data staging --------------------------------------------------------
 
@@ -67,16 +81,16 @@ class Worker:
 
     def _pull_and_load(self):
         flat = self.transport.pull()
-        unpack_params(self.module, flat)
+        if flat.data_ptr() != self.flat_params.data_ptr():
+            self.flat_params.copy_(flat)
 
     def _grad_step(self, xb: torch.Tensor, yb: Optional[torch.Tensor]):
-        self.module.zero_grad(set_to_none=True)
+        self.flat_grads.zero_()
         out = self.module(xb)
         target = yb if yb is not None else xb
         loss = self.module.compute_loss(out, target)
         loss.backward()
-        self._grad_buf = pack_grads(self.module, self._grad_buf)
-        return self._grad_buf
+        return self.flat_grads
 
     def _push(self, grads: torch.Tensor):
         try:
