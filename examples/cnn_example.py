@@ -1,0 +1,38 @@
+"""MNIST CNN on Spark (port of reference examples/cnn_example.py) using
+the async-with-lock mode (BASELINE.json config 3)."""
+from pyspark.ml.feature import VectorAssembler
+from pyspark.sql import SparkSession
+
+from sparkflow_amd.graph_utils import build_graph
+from sparkflow_amd.models import mnist_cnn
+from sparkflow_amd.spark import SparkAsyncDL
+
+
+def main():
+    spark = SparkSession.builder.appName("sparkflow-amd cnn").getOrCreate()
+    df = (spark.read.option("inferSchema", "true")
+          .csv("examples/mnist_train.csv"))
+    va = VectorAssembler(inputCols=df.columns[1:785], outputCol="features")
+
+    spark_model = SparkAsyncDL(
+        inputCol="features",
+        tensorflowGraph=build_graph(mnist_cnn),
+        tfInput="x:0",
+        tfLabel="_c0",
+        tfOutput="out:0",
+        tfLearningRate=.001,
+        iters=20,
+        predictionCol="predicted",
+        labelCol="_c0",
+        partitions=8,
+        miniBatchSize=128,
+        miniStochasticIters=1,
+        acquireLock=True,  # RWLock-gated async mode
+        verbose=1,
+    )
+    model = spark_model.fit(va.transform(df))
+    model.transform(va.transform(df)).select("predicted", "_c0").show(10)
+
+
+if __name__ == "__main__":
+    main()

@@ -87,6 +87,12 @@ def module_from_json(graph_json: str) -> torch.nn.Module:
     return module
 
 
+def capture_module(module: torch.nn.Module) -> str:
+    """Serialize an already-built module instance (checkpoint import path —
+    no separate builder function available)."""
+    return module_to_json(lambda: module, module)
+
+
 # ---------------------------------------------------------------------------
 # Flat weight/grad packing.
 #

@@ -1,0 +1,78 @@
+"""Pipeline byte codec + checkpoint import (no JVM needed for these)."""
+import numpy as np
+import pytest
+import torch
+
+from sparkflow_amd.pipeline_util import (
+    PY_OBJ_GUID,
+    decode_py_obj,
+    encode_py_obj,
+    load_byte_array,
+)
+
+
+class Dummy:
+    def __init__(self, a, b):
+        self.a = a
+        self.b = b
+
+
+def test_byte_codec_roundtrip():
+    obj = Dummy([1, 2, 3], {"k": "v"})
+    words = encode_py_obj(obj)
+    assert len(words) == 2
+    assert words[1] == PY_OBJ_GUID
+    assert words[0].endswith(",")
+    # payload is strictly comma-separated byte ints (the reference format)
+    ints = words[0].split(",")[:-1]
+    assert all(0 <= int(i) <= 255 for i in ints)
+    back = decode_py_obj(words)
+    assert back.a == obj.a and back.b == obj.b
+    # reference-name alias with sentinel stripped (pipeline_util.py:34-45)
+    back2 = load_byte_array(words[:-1])
+    assert back2.a == obj.a
+
+
+def test_byte_codec_matches_reference_construction():
+    """Re-derive the encoding by hand exactly as the reference writes it."""
+    import zlib
+
+    import dill
+
+    obj = {"x": 42}
+    words = encode_py_obj(obj)
+    dmp = zlib.compress(dill.dumps(obj))
+    ref_payload = "".join(str(i) + "," for i in bytearray(dmp))
+    assert words[0] == ref_payload
+
+
+def test_load_torch_model_from_module_checkpoint(tmp_path):
+    from sparkflow_amd.models import mnist_mlp
+    from sparkflow_amd.utils.serialization import capture_module, module_from_json
+    from sparkflow_amd.ml_util import (
+        convert_json_to_weights,
+        torch_get_weights,
+        torch_set_weights,
+    )
+
+    m = mnist_mlp()
+    path = tmp_path / "model.pt"
+    torch.save(m, str(path))
+
+    # the loader path without pyspark: exercise the capture + codec pieces
+    graph_json = capture_module(m)
+    m2 = module_from_json(graph_json)
+    x = torch.randn(3, 784)
+    torch.testing.assert_close(m(x), m2(x))
+
+
+def test_load_torch_model_api(tmp_path):
+    pytest.importorskip("pyspark")
+    from sparkflow_amd.torch_model_loader import load_torch_model
+    from sparkflow_amd.models import mnist_mlp
+
+    m = mnist_mlp()
+    path = tmp_path / "model.pt"
+    torch.save(m, str(path))
+    model = load_torch_model(str(path))
+    assert model.getOrDefault(model.modelWeights)
