@@ -151,11 +151,27 @@ class Worker:
                     self.loss_callback(loss, it, self.partition_id)
 
 
-    def run_steps(self, k: int, mini_batch: int) -> int:
+    def run_steps(self, k: int, mini_batch: int,
+                  use_graph: Optional[bool] = None) -> int:
         """Benchmark-shaped loop: k steps of {pull -> one stochastic
-        minibatch grad -> push}.  Returns samples processed."""
+        minibatch grad -> push}.  Returns samples processed.
+
+        On GPU the forward/backward compute is captured once as a hipGraph
+        and replayed per step (launch-bound inner loops belong in graphs on
+        MI355X); pulls, minibatch selection and pushes stay eager so the
+        async-PS semantics (fresh weights per step, racy applies) are
+        unchanged.  Falls back to the eager loop if capture fails.
+        """
         n = self.x.shape[0]
         mbs = min(mini_batch, n)
+        if use_graph is None:
+            use_graph = self.device.type == "cuda"
+        if use_graph and self.device.type == "cuda":
+            try:
+                return self._run_steps_graphed(k, mbs)
+            except Exception as e:
+                print(f"[sparkflow_amd] hipGraph capture failed ({e}); "
+                      "running eager steps")
         for _ in range(k):
             self._pull_and_load()
             if mbs < n:
@@ -165,6 +181,46 @@ class Worker:
             else:
                 xb, yb = self.x, self.y
             self._push(self._grad_step(xb, yb))
+        return k * mbs
+
+    def _run_steps_graphed(self, k: int, mbs: int) -> int:
+        n = self.x.shape[0]
+        sub = mbs < n
+        xb = torch.empty((mbs,) + tuple(self.x.shape[1:]),
+                         dtype=self.x.dtype, device=self.device) \
+            if sub else self.x
+        yb = None
+        if self.y is not None:
+            yb = torch.empty((mbs,) + tuple(self.y.shape[1:]),
+                             dtype=self.y.dtype, device=self.device) \
+                if sub else self.y
+
+        def fill_batch():
+            if not sub:
+                return
+            pick = self._minibatch_indices(n, mbs)
+            torch.index_select(self.x, 0, pick, out=xb)
+            if yb is not None:
+                torch.index_select(self.y, 0, pick, out=yb)
+
+        fill_batch()
+        # warmup on a side stream (allocator state), then capture
+        s = torch.cuda.Stream(device=self.device)
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                self._grad_step(xb, yb)
+        torch.cuda.current_stream().wait_stream(s)
+
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            self._grad_step(xb, yb)
+
+        for _ in range(k):
+            self._pull_and_load()
+            fill_batch()
+            graph.replay()
+            self._push(self.flat_grads)
         return k * mbs
 
 
