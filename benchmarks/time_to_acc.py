@@ -1,0 +1,136 @@
+#!/usr/bin/env python3
+"""Time-to-97%-accuracy benchmark (the second half of BASELINE.json's
+headline metric) on the MNIST-shaped MLP.
+
+There is no network access for real MNIST, so the dataset is a synthetic
+10-class mixture in 784-dim (class template + gaussian noise, `--noise`
+controls difficulty) — random-init weights, fp32, async-PS Hogwild.
+
+Run like bench.py:
+  python benchmarks/time_to_acc.py --gpus 1
+  torchrun --nproc-per-node N benchmarks/time_to_acc.py --gpus N
+"""
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+import numpy as np
+import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+
+def make_synthetic_mnist(rng, n, n_classes=10, dim=784, noise=2.0):
+    templates = rng.normal(0, 1, (n_classes, dim)).astype(np.float32)
+    labels = rng.integers(0, n_classes, n)
+    x = templates[labels] + rng.normal(0, noise, (n, dim)).astype(np.float32)
+    return x.astype(np.float32), labels.astype(np.int64)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--target", type=float, default=0.97)
+    ap.add_argument("--mbs", type=int, default=300)
+    ap.add_argument("--noise", type=float, default=2.0)
+    ap.add_argument("--round-steps", type=int, default=20)
+    ap.add_argument("--max-rounds", type=int, default=200)
+    ap.add_argument("--train-rows", type=int, default=6000)
+    args = ap.parse_args()
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import mnist_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+    from sparkflow_amd.utils.serialization import unpack_params
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+
+    graph_json = build_graph(mnist_mlp)
+    cfg = TrainConfig(iters=args.max_rounds * args.round_steps,
+                      mini_batch=args.mbs, mini_stochastic_iters=1,
+                      shuffle=False)
+    engine = build_engine(graph_json, OptimizerSpec("adam", 0.001), cfg)
+    device = engine.device
+
+    rng = np.random.default_rng(1000 + rank)
+    x, y = make_synthetic_mnist(rng, args.train_rows, noise=args.noise)
+    engine.worker.stage(x, y.astype(np.float32).reshape(-1, 1))
+
+    # rank-0 evaluation set (same distribution, templates differ per rank
+    # seed, so eval uses rank 0's data distribution)
+    if rank == 0:
+        ev_rng = np.random.default_rng(1000)
+        _tx, _ty = make_synthetic_mnist(ev_rng, args.train_rows,
+                                        noise=args.noise)
+        ex, ey = make_synthetic_mnist(ev_rng, 2000, noise=args.noise)
+        ex_t = torch.as_tensor(ex, device=device)
+        ey_t = torch.as_tensor(ey, device=device)
+        eval_module = engine.worker.module  # weights loaded from PS below
+
+    # a couple of warmup steps to build communicators / graphs
+    engine.worker.run_steps(3, args.mbs)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    engine.barrier()
+
+    flag = torch.zeros(1, dtype=torch.int64)
+    t0 = time.perf_counter()
+    elapsed = float("nan")
+    rounds = 0
+    acc = 0.0
+    for rounds in range(1, args.max_rounds + 1):
+        engine.worker.run_steps(args.round_steps, args.mbs)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        engine.barrier()
+        if rank == 0:
+            unpack_params(eval_module, engine.ps.weights)
+            with torch.no_grad():
+                pred = engine.worker.module(ex_t).argmax(dim=1)
+            acc = float((pred == ey_t).float().mean())
+            flag.fill_(1 if acc >= args.target else 0)
+        if world > 1:
+            import torch.distributed as dist
+
+            dist.broadcast(flag, src=0)
+        if int(flag.item()) == 1:
+            elapsed = time.perf_counter() - t0
+            break
+
+    if rank == 0:
+        total_steps = rounds * args.round_steps
+        print(json.dumps({
+            "metric": "time_to_97pct_acc",
+            "value": elapsed,
+            "unit": "s",
+            "n_gpus": world,
+            "steps": total_steps,
+            "warmup": 3,
+            "ms_per_step": (elapsed / total_steps * 1000.0
+                            if elapsed == elapsed else None),
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "final_accuracy": acc,
+            "config": {"model": "mnist_mlp_784_256_256_10",
+                       "global_batch": args.mbs * world,
+                       "seq_len": None,
+                       "parallelism": f"async-ps-hogwild dp{world}"},
+        }))
+
+    engine.close()
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

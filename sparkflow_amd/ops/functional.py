@@ -61,8 +61,13 @@ class LinearAct(torch.autograd.Function):
         if ops.use_hip(x):
             dpre = ops.hip.act_bwd(y, dy, act) if act != ACT_NONE else dy
             dx = ops.hip.gemm_dgrad(dpre, w)
-            dw = ops.hip.gemm_wgrad(x, dpre)
-            db = ops.hip.bias_grad(dpre) if ctx.has_bias else None
+            if ctx.has_bias:
+                # bias grad rides the wgrad GEMM as a virtual ones-row
+                dwb = ops.hip.gemm_wgrad_bias(x, dpre)
+                k = x.shape[1]
+                dw, db = dwb[:k], dwb[k]
+            else:
+                dw, db = ops.hip.gemm_wgrad(x, dpre), None
         else:
             dpre = _act_bwd_from_out(y, dy, act)
             dx = dpre @ w.t()

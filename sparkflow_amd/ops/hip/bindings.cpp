@@ -8,6 +8,7 @@ torch::Tensor gemm_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                        int64_t act);
 torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w);
 torch::Tensor gemm_wgrad(torch::Tensor x, torch::Tensor dy);
+torch::Tensor gemm_wgrad_bias(torch::Tensor x, torch::Tensor dy);
 torch::Tensor bias_grad(torch::Tensor dy);
 
 // elementwise.hip
@@ -33,6 +34,9 @@ void fused_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor acc,
 void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr_t, double b1, double b2,
                 double eps);
+void fused_adam_dev(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                    torch::Tensor v, torch::Tensor state, double lr,
+                    double b1, double b2, double eps);
 void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
                    torch::Tensor mom, double lr, double decay,
                    double momentum, double eps);
@@ -55,6 +59,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_act", &gemm_act, "fused fp32 MFMA GEMM + bias + activation");
   m.def("gemm_dgrad", &gemm_dgrad, "dx = dy @ W^T");
   m.def("gemm_wgrad", &gemm_wgrad, "dW = x^T @ dy");
+  m.def("gemm_wgrad_bias", &gemm_wgrad_bias,
+        "dWb = [x;ones]^T @ dy (bias grad fused as last row)");
   m.def("bias_grad", &bias_grad, "column-sum bias gradient");
   m.def("act_bwd", &act_bwd, "activation backward from output");
   m.def("argmax_rows", &argmax_rows, "rowwise argmax head");
@@ -69,6 +75,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "single-launch SGD apply");
   m.def("fused_momentum", &fused_momentum, "single-launch momentum apply");
   m.def("fused_adam", &fused_adam, "single-launch Adam apply");
+  m.def("fused_adam_dev", &fused_adam_dev,
+        "Adam apply with device-resident step state (hipGraph-safe)");
   m.def("fused_rmsprop", &fused_rmsprop, "single-launch RMSProp apply");
   m.def("conv2d_act", &conv2d_act, "conv2d + bias + activation fwd");
   m.def("conv2d_dgrad", &conv2d_dgrad, "conv2d input grad");

@@ -30,7 +30,7 @@
 // to a [splitk, M, N] workspace with plain stores (no zero-fill pass, no
 // atomics, deterministic); a small reduce epilogue sums the slices and
 // applies bias+activation.
-template <bool TA, bool TB, bool HAS_BIAS, bool SPLITK>
+template <bool TA, bool TB, bool HAS_BIAS, bool SPLITK, bool BIASROW = false>
 __global__ __launch_bounds__(256) void gemm_f32_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     const float* __restrict__ bias, float* __restrict__ C, int M, int N,
@@ -77,7 +77,11 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
     } else {
       const int kk = t >> 4, m4 = (t & 15) * 4;  // 16 k-rows x 16 float4
       const int gk = k0 + kk;
-      if (gk < K && m0 + BM <= M) {
+      // BIASROW: logical row M-1 is a virtual all-ones row (it computes
+      // the bias gradient inside the wgrad GEMM); the fast path must not
+      // read it from memory.
+      const int m_phys = BIASROW ? M - 1 : M;
+      if (gk < K && m0 + BM <= m_phys) {
         const float4 v = *reinterpret_cast<const float4*>(
             &A[(long)gk * lda + m0 + m4]);
         As[m4 + 0][kk] = v.x;
@@ -88,8 +92,12 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
           const int gm = m0 + m4 + i;
-          As[m4 + i][kk] =
-              (gk < K && gm < M) ? A[(long)gk * lda + gm] : 0.0f;
+          float val = 0.0f;
+          if (gk < K && gm < M) {
+            val = (BIASROW && gm == M - 1) ? 1.0f
+                                           : A[(long)gk * lda + gm];
+          }
+          As[m4 + i][kk] = val;
         }
       }
     }
@@ -312,6 +320,41 @@ torch::Tensor gemm_wgrad(torch::Tensor x, torch::Tensor dy) {
   launch_gemm<true, false>(x, dy, empty, dw, Kin, Nout, Mb, Kin, Nout,
                            ACT_NONE);
   return dw;
+}
+
+torch::Tensor gemm_wgrad_bias(torch::Tensor x, torch::Tensor dy) {
+  // dwb[Kin+1, Nout] = [x; ones]^T @ dy : rows 0..Kin-1 are dW, row Kin
+  // is the bias gradient (column sums) — one GEMM instead of a GEMM plus
+  // a separate reduction kernel.
+  check_f32(x, "x");
+  check_f32(dy, "dy");
+  const int Mb = x.size(0), Kin = x.size(1), Nout = dy.size(1);
+  TORCH_CHECK(dy.size(0) == Mb, "shape mismatch wgrad_bias");
+  const int M = Kin + 1;
+  auto dwb = torch::empty({M, Nout}, x.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int splitk = pick_splitk(M, Nout, Mb);
+  if (splitk > 1) {
+    const int k_chunk = ceil_div(Mb, splitk * BK) * BK;
+    auto ws = torch::empty({splitk, M, Nout}, x.options());
+    dim3 grid(ceil_div(Nout, BN), ceil_div(M, BM), splitk);
+    hipLaunchKernelGGL((gemm_f32_kernel<true, false, false, true, true>),
+                       grid, dim3(256), 0, stream, x.data_ptr<float>(),
+                       dy.data_ptr<float>(), nullptr, ws.data_ptr<float>(),
+                       M, Nout, Mb, Kin, Nout, ACT_NONE, k_chunk);
+    const long n = (long)M * Nout;
+    const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
+    hipLaunchKernelGGL((splitk_reduce_kernel<false>), dim3(blocks),
+                       dim3(256), 0, stream, ws.data_ptr<float>(), nullptr,
+                       dwb.data_ptr<float>(), n, Nout, splitk, ACT_NONE);
+  } else {
+    dim3 grid(ceil_div(Nout, BN), ceil_div(M, BM));
+    hipLaunchKernelGGL((gemm_f32_kernel<true, false, false, false, true>),
+                       grid, dim3(256), 0, stream, x.data_ptr<float>(),
+                       dy.data_ptr<float>(), nullptr, dwb.data_ptr<float>(),
+                       M, Nout, Mb, Kin, Nout, ACT_NONE, Mb);
+  }
+  return dwb;
 }
 
 torch::Tensor bias_grad(torch::Tensor dy) {

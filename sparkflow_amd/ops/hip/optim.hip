@@ -49,6 +49,34 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
   }
 }
 
+// Device-state Adam: the step count and bias-corrected lr live in a small
+// device buffer (state[0]=t, state[1]=lr_t) advanced by a 1-thread prep
+// kernel — so a PS apply can be captured in a hipGraph with the bias
+// correction still advancing per replay (a host-computed lr_t would be
+// frozen into the graph).
+__global__ void adam_prep_kernel(float* __restrict__ state, float lr,
+                                 float b1, float b2) {
+  const float t = state[0] + 1.0f;
+  state[0] = t;
+  state[1] = lr * __fsqrt_rn(1.0f - __powf(b2, t)) / (1.0f - __powf(b1, t));
+}
+
+__global__ void adam_dev_kernel(float* __restrict__ p,
+                                const float* __restrict__ g,
+                                float* __restrict__ m, float* __restrict__ v,
+                                const float* __restrict__ state, float b1,
+                                float b2, float eps, long n) {
+  const float lr_t = state[1];
+  GS_LOOP(i, n) {
+    const float gi = g[i];
+    const float mi = b1 * m[i] + (1.0f - b1) * gi;
+    const float vi = b2 * v[i] + (1.0f - b2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    p[i] -= lr_t * mi / (__fsqrt_rn(vi) + eps);
+  }
+}
+
 __global__ void rmsprop_kernel(float* __restrict__ p,
                                const float* __restrict__ g,
                                float* __restrict__ ms,
@@ -109,6 +137,26 @@ void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      at::cuda::getCurrentCUDAStream(), p.data_ptr<float>(),
                      g.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), (float)lr_t, (float)b1, (float)b2,
+                     (float)eps, n);
+}
+
+void fused_adam_dev(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                    torch::Tensor v, torch::Tensor state, double lr,
+                    double b1, double b2, double eps) {
+  const long n = p.numel();
+  check_flat(p, "p", n);
+  check_flat(g, "g", n);
+  check_flat(m, "m", n);
+  check_flat(v, "v", n);
+  TORCH_CHECK(state.numel() >= 2 && state.is_cuda(), "bad adam state");
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(adam_prep_kernel, dim3(1), dim3(1), 0, stream,
+                     state.data_ptr<float>(), (float)lr, (float)b1,
+                     (float)b2);
+  hipLaunchKernelGGL(adam_dev_kernel, dim3(opt_blocks(n)), dim3(256), 0,
+                     stream, p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     state.data_ptr<float>(), (float)b1, (float)b2,
                      (float)eps, n);
 }
 

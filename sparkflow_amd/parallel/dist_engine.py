@@ -51,6 +51,26 @@ def init_from_env(backend: Optional[str] = None, device: Optional[str] = None):
     return rank, world, torch.device(device)
 
 
+def _warmup_pairs(pairs: DistPairs, device) -> None:
+    """Initialize every pair communicator SEQUENTIALLY from the main
+    thread before the PS service threads start.  Lazy init from N-1
+    concurrent service threads is where pairwise-NCCL designs deadlock;
+    one deterministic ping per pair avoids it."""
+    import torch.distributed as dist
+
+    t = torch.ones(1, device=device)
+    c = torch.ones(1)
+    if pairs.rank == 0:
+        for w in range(1, pairs.world):
+            dist.send(t, dst=w, group=pairs.payload[w])
+            dist.send(c, dst=w, group=pairs.ctrl[w])
+    else:
+        dist.recv(t, src=0, group=pairs.payload[pairs.rank])
+        dist.recv(c, src=0, group=pairs.ctrl[pairs.rank])
+    if device.type == "cuda":
+        torch.cuda.synchronize(device)
+
+
 @dataclass
 class EngineHandle:
     rank: int
@@ -82,6 +102,7 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
     ps = service = None
     if world > 1:
         pairs = DistPairs()
+        _warmup_pairs(pairs, device)
     if rank == 0:
         ps = PSCore(graph_json, optimizer, device=device,
                     acquire_lock=config.acquire_lock,

@@ -57,6 +57,12 @@ class FlatOptimizer:
         """Optimizer state for checkpoint/resume."""
         return {}
 
+    def step_count(self) -> int:
+        return self.t
+
+    def set_step_count(self, t: int) -> None:
+        self.t = int(t)
+
 
 class GradientDescent(FlatOptimizer):
     name = "gradient_descent"
@@ -110,24 +116,40 @@ class Adam(FlatOptimizer):
         self.beta1, self.beta2, self.eps = float(beta1), float(beta2), float(epsilon)
         self.m = self._zeros()
         self.v = self._zeros()
+        # GPU path: the step count + bias-corrected lr live on-device
+        # (state[0]=t, state[1]=lr_t) so an apply is hipGraph-capturable.
+        self.dev_state = (torch.zeros(2, dtype=torch.float32,
+                                      device=self.device)
+                          if self.device.type == "cuda" else None)
 
     def apply(self, param, grad):
         from sparkflow_amd import ops
 
+        if ops.use_hip(param):
+            ops.hip.fused_adam_dev(param, grad, self.m, self.v,
+                                   self.dev_state, self.lr, self.beta1,
+                                   self.beta2, self.eps)
+            return
         self.t += 1
         # TF1 Adam: lr_t = lr * sqrt(1-b2^t) / (1-b1^t); p -= lr_t * m/(sqrt(v)+eps)
         lr_t = self.lr * math.sqrt(1.0 - self.beta2 ** self.t) / (
             1.0 - self.beta1 ** self.t)
-        if ops.use_hip(param):
-            ops.hip.fused_adam(param, grad, self.m, self.v, lr_t,
-                               self.beta1, self.beta2, self.eps)
-            return
         self.m.mul_(self.beta1).add_(grad, alpha=1.0 - self.beta1)
         self.v.mul_(self.beta2).addcmul_(grad, grad, value=1.0 - self.beta2)
         param.addcdiv_(self.m, self.v.sqrt().add_(self.eps), value=-lr_t)
 
     def state_tensors(self):
         return {"m": self.m, "v": self.v}
+
+    def step_count(self) -> int:
+        if self.dev_state is not None:
+            return int(self.dev_state[0].item())
+        return self.t
+
+    def set_step_count(self, t: int) -> None:
+        self.t = int(t)
+        if self.dev_state is not None:
+            self.dev_state[0] = float(t)
 
 
 class RMSProp(FlatOptimizer):

@@ -145,7 +145,7 @@ class PSCore:
             "weights": self.weights.detach().cpu(),
             "optimizer": {k: v.detach().cpu()
                           for k, v in self.optimizer.state_tensors().items()},
-            "t": self.optimizer.t,
+            "t": self.optimizer.step_count(),
             "update_count": self.update_count,
         }
 
@@ -154,5 +154,5 @@ class PSCore:
             self.weights.copy_(state["weights"].to(self.device))
             for k, v in state.get("optimizer", {}).items():
                 self.optimizer.state_tensors()[k].copy_(v.to(self.device))
-        self.optimizer.t = state.get("t", 0)
+        self.optimizer.set_step_count(state.get("t", 0))
         self.update_count = state.get("update_count", 0)

@@ -183,9 +183,63 @@ class Worker:
             self._push(self._grad_step(xb, yb))
         return k * mbs
 
+    _IDX_POOL_CAP = 2048
+
     def _run_steps_graphed(self, k: int, mbs: int) -> int:
+        """One hipGraph replay per step.
+
+        For a same-GPU Hogwild LocalTransport (rank 0 / single-GPU jobs)
+        the ENTIRE step is captured — weight-pull copy, minibatch gather
+        (indices come from a device-side pool walked by a captured step
+        counter, preserving sample-without-replacement semantics), forward
+        /backward, and the PS's fused optimizer apply (device-state Adam)
+        — so a training step is a single graph launch.  For remote
+        (RCCL/TCP) transports only the compute is captured; pull/push stay
+        eager.
+        """
+        from sparkflow_amd.parallel.transport import LocalTransport
+
         n = self.x.shape[0]
         sub = mbs < n
+        full_step = (isinstance(self.transport, LocalTransport)
+                     and not self.transport.ps.acquire_lock
+                     and self.transport.ps.device == self.device)
+
+        key = (mbs, sub, full_step, n, self.x.data_ptr())
+        if getattr(self, "_graph_key", None) != key:
+            self._capture_graph(key, mbs, sub, full_step)
+
+        g = self._graph_state
+        # refill the index pool with this call's picks
+        if sub:
+            rows = min(k, self._IDX_POOL_CAP)
+            picks = np.stack([self.rng.choice(n, mbs, replace=False)
+                              for _ in range(rows)])
+            g["idx_pool"][:rows].copy_(
+                torch.as_tensor(picks, device=self.device))
+            g["ctr"].zero_()
+
+        ps = self.transport.ps if full_step else None
+        for _ in range(k):
+            if full_step:
+                g["graph"].replay()
+            else:
+                self._pull_and_load()
+                if sub:
+                    pick = self._minibatch_indices(n, mbs)
+                    torch.index_select(self.x, 0, pick, out=g["xb"])
+                    if g["yb"] is not None:
+                        torch.index_select(self.y, 0, pick, out=g["yb"])
+                g["graph"].replay()
+                self._push(self.flat_grads)
+        if full_step:
+            ps.update_count += k
+            ps.pull_count += k
+            self.steps_done += k
+        return k * mbs
+
+    def _capture_graph(self, key, mbs: int, sub: bool, full_step: bool):
+        n = self.x.shape[0]
         xb = torch.empty((mbs,) + tuple(self.x.shape[1:]),
                          dtype=self.x.dtype, device=self.device) \
             if sub else self.x
@@ -194,34 +248,47 @@ class Worker:
             yb = torch.empty((mbs,) + tuple(self.y.shape[1:]),
                              dtype=self.y.dtype, device=self.device) \
                 if sub else self.y
-
-        def fill_batch():
-            if not sub:
-                return
-            pick = self._minibatch_indices(n, mbs)
-            torch.index_select(self.x, 0, pick, out=xb)
+        idx_pool = None
+        ctr = None
+        if sub:
+            idx_pool = torch.zeros(self._IDX_POOL_CAP, mbs,
+                                   dtype=torch.int64, device=self.device)
+            ctr = torch.zeros(1, dtype=torch.int64, device=self.device)
+            pick0 = self._minibatch_indices(n, mbs)
+            torch.index_select(self.x, 0, pick0, out=xb)
             if yb is not None:
-                torch.index_select(self.y, 0, pick, out=yb)
+                torch.index_select(self.y, 0, pick0, out=yb)
 
-        fill_batch()
-        # warmup on a side stream (allocator state), then capture
+        ps = self.transport.ps if full_step else None
+
+        def body():
+            if full_step:
+                self.flat_params.copy_(ps.weights)  # the weight pull
+            if full_step and sub:
+                row = idx_pool.index_select(0, ctr).view(-1)
+                torch.index_select(self.x, 0, row, out=xb)
+                if yb is not None:
+                    torch.index_select(self.y, 0, row, out=yb)
+                ctr.add_(1).remainder_(self._IDX_POOL_CAP)
+            self._grad_step(xb, yb)
+            if full_step:
+                ps.optimizer.apply(ps.weights, self.flat_grads)
+
+        # allocator warmup on a side stream, then capture
         s = torch.cuda.Stream(device=self.device)
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(3):
-                self._grad_step(xb, yb)
+                body()
         torch.cuda.current_stream().wait_stream(s)
 
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
-            self._grad_step(xb, yb)
+            body()
 
-        for _ in range(k):
-            self._pull_and_load()
-            fill_batch()
-            graph.replay()
-            self._push(self.flat_grads)
-        return k * mbs
+        self._graph_state = {"graph": graph, "xb": xb, "yb": yb,
+                             "idx_pool": idx_pool, "ctr": ctr}
+        self._graph_key = key
 
 
 def run_partition(data_iter, graph_json: str, transport: WorkerTransport,
