@@ -23,8 +23,15 @@ import torch
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 
-def make_synthetic_mnist(rng, n, n_classes=10, dim=784, noise=2.0):
-    templates = rng.normal(0, 1, (n_classes, dim)).astype(np.float32)
+def make_templates(n_classes=10, dim=784, seed=77):
+    """The class templates ARE the data distribution — fixed seed so every
+    rank and the eval set share the same 10 classes."""
+    return np.random.default_rng(seed).normal(0, 1, (n_classes, dim)) \
+        .astype(np.float32)
+
+
+def make_synthetic_mnist(rng, n, templates, noise=2.0):
+    n_classes, dim = templates.shape
     labels = rng.integers(0, n_classes, n)
     x = templates[labels] + rng.normal(0, noise, (n, dim)).astype(np.float32)
     return x.astype(np.float32), labels.astype(np.int64)
@@ -35,7 +42,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--target", type=float, default=0.97)
     ap.add_argument("--mbs", type=int, default=300)
-    ap.add_argument("--noise", type=float, default=2.0)
+    ap.add_argument("--noise", type=float, default=6.0)
     ap.add_argument("--round-steps", type=int, default=20)
     ap.add_argument("--max-rounds", type=int, default=200)
     ap.add_argument("--train-rows", type=int, default=6000)
@@ -57,17 +64,16 @@ def main():
     engine = build_engine(graph_json, OptimizerSpec("adam", 0.001), cfg)
     device = engine.device
 
+    templates = make_templates()
     rng = np.random.default_rng(1000 + rank)
-    x, y = make_synthetic_mnist(rng, args.train_rows, noise=args.noise)
+    x, y = make_synthetic_mnist(rng, args.train_rows, templates,
+                                noise=args.noise)
     engine.worker.stage(x, y.astype(np.float32).reshape(-1, 1))
 
-    # rank-0 evaluation set (same distribution, templates differ per rank
-    # seed, so eval uses rank 0's data distribution)
     if rank == 0:
-        ev_rng = np.random.default_rng(1000)
-        _tx, _ty = make_synthetic_mnist(ev_rng, args.train_rows,
-                                        noise=args.noise)
-        ex, ey = make_synthetic_mnist(ev_rng, 2000, noise=args.noise)
+        ev_rng = np.random.default_rng(999)
+        ex, ey = make_synthetic_mnist(ev_rng, 2000, templates,
+                                      noise=args.noise)
         ex_t = torch.as_tensor(ex, device=device)
         ey_t = torch.as_tensor(ey, device=device)
         eval_module = engine.worker.module  # weights loaded from PS below

@@ -77,7 +77,7 @@ class ResNet18Shape(nn.Module):
         self.stem_conv = snn.Conv2d(3, 64, 7, stride=2, padding=3,
                                     activation="none", bias=False)
         self.stem_bn = nn.BatchNorm2d(64)
-        self.stem_pool = snn.MaxPool2d(3, 2)  # 112 -> 55 (valid pooling)
+        self.stem_pool = snn.MaxPool2d(3, 2, padding=1)  # 112 -> 56
         stages = []
         cin = 64
         for cout, stride in [(64, 1), (64, 1), (128, 2), (128, 1),

@@ -66,13 +66,14 @@ class Conv2d(nn.Module):
 
 
 class MaxPool2d(nn.Module):
-    def __init__(self, kernel_size: int, stride: int = None):
+    def __init__(self, kernel_size: int, stride: int = None, padding: int = 0):
         super().__init__()
         self.kernel_size = kernel_size
         self.stride = stride or kernel_size
+        self.padding = padding
 
     def forward(self, x):
-        return F.maxpool2d(x, self.kernel_size, self.stride)
+        return F.maxpool2d(x, self.kernel_size, self.stride, self.padding)
 
 
 class Flatten(nn.Module):

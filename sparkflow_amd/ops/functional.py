@@ -175,7 +175,7 @@ class Conv2dAct(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, stride: int, pad: int, act: int):
         if ops.use_hip(x):
-            y = ops.hip.conv2d_act(
+            y = ops.hip.conv_gemm_fwd(
                 x, w, b if b is not None else torch.empty(0, device=x.device),
                 stride, pad, act)
         else:
@@ -193,7 +193,8 @@ class Conv2dAct(torch.autograd.Function):
         if ops.use_hip(x):
             dpre = ops.hip.act_bwd(y, dy, act) if act != ACT_NONE else dy
             dx = ops.hip.conv2d_dgrad(dpre, w, x.shape[2], x.shape[3], stride, pad)
-            dw = ops.hip.conv2d_wgrad(x, dpre, w.shape[2], w.shape[3], stride, pad)
+            dw = ops.hip.conv_gemm_wgrad(x, dpre, w.shape[2], w.shape[3],
+                                         stride, pad)
             db = ops.hip.conv_bias_grad(dpre) if has_bias else None
         else:
             dpre = _act_bwd_from_out(y, dy, act)
@@ -212,31 +213,32 @@ def conv2d_act(x, w, b=None, stride=1, pad=0, act="none"):
 
 class MaxPool2d(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, k: int, stride: int):
+    def forward(ctx, x, k: int, stride: int, pad: int):
         if ops.use_hip(x):
-            y, idx = ops.hip.maxpool2d_fwd(x, k, stride)
+            y, idx = ops.hip.maxpool2d_fwd(x, k, stride, pad)
         else:
-            y, idx = torch.nn.functional.max_pool2d(x, k, stride,
+            y, idx = torch.nn.functional.max_pool2d(x, k, stride, padding=pad,
                                                     return_indices=True)
         ctx.save_for_backward(idx)
-        ctx.conf = (x.shape, k, stride)
+        ctx.conf = (x.shape, k, stride, pad)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         (idx,) = ctx.saved_tensors
-        shape, k, stride = ctx.conf
+        shape, k, stride, pad = ctx.conf
         dy = dy.contiguous()
         if ops.use_hip(dy):
             dx = ops.hip.maxpool2d_bwd(dy, idx, shape[2], shape[3])
         else:
             dx = torch.nn.functional.max_unpool2d(dy, idx, k, stride,
+                                                  padding=pad,
                                                   output_size=shape[2:])
-        return dx, None, None
+        return dx, None, None, None
 
 
-def maxpool2d(x, k, stride=None):
-    return MaxPool2d.apply(x, k, stride or k)
+def maxpool2d(x, k, stride=None, pad=0):
+    return MaxPool2d.apply(x, k, stride or k, pad)
 
 
 def argmax_rows(y: torch.Tensor) -> torch.Tensor:

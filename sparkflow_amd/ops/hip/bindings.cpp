@@ -44,13 +44,18 @@ void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
 // conv.hip
 torch::Tensor conv2d_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                          int64_t stride, int64_t pad, int64_t act);
+torch::Tensor conv_gemm_fwd(torch::Tensor x, torch::Tensor w,
+                            torch::Tensor bias, int64_t stride, int64_t pad,
+                            int64_t act);
+torch::Tensor conv_gemm_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
+                              int64_t Kw, int64_t stride, int64_t pad);
 torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t H,
                            int64_t W, int64_t stride, int64_t pad);
 torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
                            int64_t Kw, int64_t stride, int64_t pad);
 torch::Tensor conv_bias_grad(torch::Tensor dy);
 std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
-                                         int64_t stride);
+                                         int64_t stride, int64_t pad);
 torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
                             int64_t W);
 
@@ -78,7 +83,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam_dev", &fused_adam_dev,
         "Adam apply with device-resident step state (hipGraph-safe)");
   m.def("fused_rmsprop", &fused_rmsprop, "single-launch RMSProp apply");
-  m.def("conv2d_act", &conv2d_act, "conv2d + bias + activation fwd");
+  m.def("conv2d_act", &conv2d_act, "conv2d + bias + activation fwd (direct)");
+  m.def("conv_gemm_fwd", &conv_gemm_fwd,
+        "implicit-GEMM MFMA conv2d + bias + activation fwd");
+  m.def("conv_gemm_wgrad", &conv_gemm_wgrad,
+        "implicit-GEMM MFMA conv2d weight grad (split-K)");
   m.def("conv2d_dgrad", &conv2d_dgrad, "conv2d input grad");
   m.def("conv2d_wgrad", &conv2d_wgrad, "conv2d weight grad");
   m.def("conv_bias_grad", &conv_bias_grad, "conv bias grad");

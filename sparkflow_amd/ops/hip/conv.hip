@@ -145,8 +145,8 @@ __global__ void conv_bias_grad_kernel(const float* __restrict__ dy,
 __global__ void maxpool_fwd_kernel(const float* __restrict__ x,
                                    float* __restrict__ y,
                                    long* __restrict__ idx, int N, int C,
-                                   int H, int W, int K, int stride, int Ho,
-                                   int Wo) {
+                                   int H, int W, int K, int stride, int pad,
+                                   int Ho, int Wo) {
   const long total = (long)N * C * Ho * Wo;
   const long gstride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -159,10 +159,12 @@ __global__ void maxpool_fwd_kernel(const float* __restrict__ x,
     float best = -INFINITY;
     long best_hw = 0;
     for (int kh = 0; kh < K; ++kh) {
-      const int h = ho * stride + kh;
+      const int h = ho * stride - pad + kh;
+      if (h < 0) continue;
       if (h >= H) break;
       for (int kw = 0; kw < K; ++kw) {
-        const int ww = wo * stride + kw;
+        const int ww = wo * stride - pad + kw;
+        if (ww < 0) continue;
         if (ww >= W) break;
         const float v = xp[(long)h * W + ww];
         if (v > best) {
@@ -265,10 +267,11 @@ torch::Tensor conv_bias_grad(torch::Tensor dy) {
 }
 
 std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
-                                         int64_t stride) {
+                                         int64_t stride, int64_t pad) {
   check_f32c(x, "x");
   const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  const int Ho = (H - k) / stride + 1, Wo = (W - k) / stride + 1;
+  const int Ho = (H + 2 * pad - k) / stride + 1;
+  const int Wo = (W + 2 * pad - k) / stride + 1;
   auto y = torch::empty({N, C, Ho, Wo}, x.options());
   auto idx = torch::empty({N, C, Ho, Wo},
                           x.options().dtype(torch::kInt64));
@@ -277,7 +280,7 @@ std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
   hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(blocks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
                      y.data_ptr<float>(), idx.data_ptr<long>(), N, C, H, W,
-                     (int)k, (int)stride, Ho, Wo);
+                     (int)k, (int)stride, (int)pad, Ho, Wo);
   return {y, idx};
 }
 

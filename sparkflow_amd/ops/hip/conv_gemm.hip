@@ -1,0 +1,324 @@
+// Implicit-GEMM convolution on the fp32 MFMA pipeline (im2col gathered at
+// the LDS staging step, never materialized) — the MI355X-native conv for
+// the CNN / ResNet-18-shape configs (SURVEY.md §2.2 "implicit-GEMM conv
+// kernel (im2col-in-LDS)").
+//
+// GEMM views (all fp32, NCHW):
+//   fwd:   C[M=N*Ho*Wo, N'=Cout]       = im2col(x) @ W^T          (+bias+act)
+//   wgrad: C[M=Cout, N'=Cin*Kh*Kw]     = dyT @ im2col(x)
+//   dgrad: falls back to conv.hip's direct kernel (the gather predicate
+//          with stride divisibility does not tile as cleanly; direct dgrad
+//          has the same MAC count as fwd).
+//
+// The MFMA inner loop is the same 64x64-tile / 16-deep structure as
+// gemm.hip; only the A/B staging differs (indexed gathers).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+#define BM 64
+#define BN 64
+#define BK 16
+
+struct ConvShape {
+  int N, Cin, H, W, Cout, Kh, Kw, Ho, Wo, stride, pad;
+};
+
+// gather one im2col element: row m = output pixel (n,ho,wo),
+// col k = (ci,kh,kw)
+__device__ __forceinline__ float im2col_get(const float* __restrict__ x,
+                                            const ConvShape s, int m,
+                                            int k) {
+  const int wo = m % s.Wo;
+  const int ho = (m / s.Wo) % s.Ho;
+  const int n = m / (s.Wo * s.Ho);
+  const int kw = k % s.Kw;
+  const int kh = (k / s.Kw) % s.Kh;
+  const int ci = k / (s.Kw * s.Kh);
+  const int h = ho * s.stride - s.pad + kh;
+  const int w = wo * s.stride - s.pad + kw;
+  if (h < 0 || h >= s.H || w < 0 || w >= s.W) return 0.0f;
+  return x[(((long)n * s.Cin + ci) * s.H + h) * s.W + w];
+}
+
+// ---------------- forward ----------------
+// A (im2col x): M x K, gathered.  B = W [Cout, K] -> B'[k][n']=W[n'][k].
+template <bool HAS_BIAS>
+__global__ __launch_bounds__(256) void conv_gemm_fwd_kernel(
+    const float* __restrict__ x, const float* __restrict__ wgt,
+    const float* __restrict__ bias, float* __restrict__ y, ConvShape s,
+    int M, int Nc, int K, int act) {
+  __shared__ float As[BM][BK + 1];
+  __shared__ float Bs[BK][BN + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = (wv >> 1) * 32;
+  const int wn = (wv & 1) * 32;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+
+  f32x4 acc[2][2] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    {  // stage A: 64 rows x 16 cols, 4 gathers per thread
+      const int row = t >> 2, c4 = (t & 3) * 4;
+      const int gm = m0 + row;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int gk = k0 + c4 + i;
+        As[row][c4 + i] =
+            (gm < M && gk < K) ? im2col_get(x, s, gm, gk) : 0.0f;
+      }
+    }
+    {  // stage B: W[n'][k] -> Bs[k][n']; W rows are contiguous in k
+      const int row = t >> 2, c4 = (t & 3) * 4;
+      const int gn = n0 + row;
+      if (gn < Nc && k0 + BK <= K) {
+        const float4 v = *reinterpret_cast<const float4*>(
+            &wgt[(long)gn * K + k0 + c4]);
+        Bs[c4 + 0][row] = v.x;
+        Bs[c4 + 1][row] = v.y;
+        Bs[c4 + 2][row] = v.z;
+        Bs[c4 + 3][row] = v.w;
+      } else {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int gk = k0 + c4 + i;
+          Bs[c4 + i][row] =
+              (gn < Nc && gk < K) ? wgt[(long)gn * K + gk] : 0.0f;
+        }
+      }
+    }
+    __syncthreads();
+
+    const int fr = lane & 15;
+    const int fk = lane >> 4;
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      const float a0 = As[wm + fr][kk * 4 + fk];
+      const float a1 = As[wm + 16 + fr][kk * 4 + fk];
+      const float b0 = Bs[kk * 4 + fk][wn + fr];
+      const float b1 = Bs[kk * 4 + fk][wn + 16 + fr];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: y is NCHW, so output pixel m of channel n' scatters to
+  // y[n][n'][ho][wo]
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+  const int HoWo = s.Ho * s.Wo;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int co = n0 + wn + ni * 16 + fcol;
+      if (co >= Nc) continue;
+      const float bv = HAS_BIAS ? bias[co] : 0.0f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int m = m0 + wm + mi * 16 + frow + i;
+        if (m < M) {
+          const int hw = m % HoWo;
+          const int n = m / HoWo;
+          y[((long)n * Nc + co) * HoWo + hw] =
+              act_apply(acc[mi][ni][i] + bv, act);
+        }
+      }
+    }
+  }
+}
+
+// ---------------- wgrad ----------------
+// C[M=Cout, N'=K'=Cin*Kh*Kw] = sum_p dy[p, co] * im2col(x)[p, k']
+// A'[m=co][k=p]: dy NCHW gather; B'[k=p][n'=k']: im2col gather.
+template <bool SPLITK>
+__global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    float* __restrict__ dw, ConvShape s, int M, int Nc, int K,
+    int k_chunk) {
+  __shared__ float As[BM][BK + 1];
+  __shared__ float Bs[BK][BN + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = (wv >> 1) * 32;
+  const int wn = (wv & 1) * 32;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+  const int HoWo = s.Ho * s.Wo;
+
+  f32x4 acc[2][2] = {};
+
+  int k_begin = 0, k_end = K;
+  if (SPLITK) {
+    k_begin = blockIdx.z * k_chunk;
+    k_end = min(K, k_begin + k_chunk);
+  }
+
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+    {  // stage A: As[co][p] = dy[n][co][hw] with p=(n,hw)
+      const int row = t >> 2, c4 = (t & 3) * 4;  // row = co idx in tile
+      const int gm = m0 + row;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int gp = k0 + c4 + i;
+        float v = 0.0f;
+        if (gm < M && gp < K) {
+          const int hw = gp % HoWo;
+          const int n = gp / HoWo;
+          v = dy[((long)n * M + gm) * HoWo + hw];
+        }
+        As[row][c4 + i] = v;
+      }
+    }
+    {  // stage B: Bs[p][k'] = im2col(x)[p, k']
+      const int kk = t >> 4, n4 = (t & 15) * 4;
+      const int gp = k0 + kk;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int gk = n0 + n4 + i;
+        Bs[kk][n4 + i] =
+            (gp < K && gk < Nc) ? im2col_get(x, s, gp, gk) : 0.0f;
+      }
+    }
+    __syncthreads();
+
+    const int fr = lane & 15;
+    const int fk = lane >> 4;
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      const float a0 = As[wm + fr][kk * 4 + fk];
+      const float a1 = As[wm + 16 + fr][kk * 4 + fk];
+      const float b0 = Bs[kk * 4 + fk][wn + fr];
+      const float b1 = Bs[kk * 4 + fk][wn + 16 + fr];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int kcol = n0 + wn + ni * 16 + fcol;
+      if (kcol >= Nc) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int co = m0 + wm + mi * 16 + frow + i;
+        if (co < M) {
+          if (SPLITK) {
+            dw[(long)blockIdx.z * M * Nc + (long)co * Nc + kcol] =
+                acc[mi][ni][i];
+          } else {
+            dw[(long)co * Nc + kcol] = acc[mi][ni][i];
+          }
+        }
+      }
+    }
+  }
+}
+
+__global__ void conv_splitk_reduce_kernel(const float* __restrict__ W,
+                                          float* __restrict__ C, long n,
+                                          int splitk) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float sum = 0.0f;
+    for (int z = 0; z < splitk; ++z) sum += W[(long)z * n + i];
+    C[i] = sum;
+  }
+}
+
+// ---------------------------------------------------------------------------
+
+static inline void check_f32g(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                  t.is_contiguous(),
+              name, " must be contiguous fp32 on GPU");
+}
+
+torch::Tensor conv_gemm_fwd(torch::Tensor x, torch::Tensor w,
+                            torch::Tensor bias, int64_t stride, int64_t pad,
+                            int64_t act) {
+  check_f32g(x, "x");
+  check_f32g(w, "w");
+  ConvShape s;
+  s.N = x.size(0); s.Cin = x.size(1); s.H = x.size(2); s.W = x.size(3);
+  s.Cout = w.size(0); s.Kh = w.size(2); s.Kw = w.size(3);
+  s.stride = stride; s.pad = pad;
+  s.Ho = (s.H + 2 * s.pad - s.Kh) / s.stride + 1;
+  s.Wo = (s.W + 2 * s.pad - s.Kw) / s.stride + 1;
+  const int M = s.N * s.Ho * s.Wo;
+  const int Nc = s.Cout;
+  const int K = s.Cin * s.Kh * s.Kw;
+  auto y = torch::empty({s.N, s.Cout, s.Ho, s.Wo}, x.options());
+  dim3 grid(ceil_div(Nc, BN), ceil_div(M, BM));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (bias.numel() > 0) {
+    hipLaunchKernelGGL((conv_gemm_fwd_kernel<true>), grid, dim3(256), 0,
+                       stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                       bias.data_ptr<float>(), y.data_ptr<float>(), s, M, Nc,
+                       K, (int)act);
+  } else {
+    hipLaunchKernelGGL((conv_gemm_fwd_kernel<false>), grid, dim3(256), 0,
+                       stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                       nullptr, y.data_ptr<float>(), s, M, Nc, K, (int)act);
+  }
+  return y;
+}
+
+torch::Tensor conv_gemm_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
+                              int64_t Kw, int64_t stride, int64_t pad) {
+  check_f32g(x, "x");
+  check_f32g(dy, "dy");
+  ConvShape s;
+  s.N = x.size(0); s.Cin = x.size(1); s.H = x.size(2); s.W = x.size(3);
+  s.Cout = dy.size(1); s.Kh = Kh; s.Kw = Kw;
+  s.stride = stride; s.pad = pad;
+  s.Ho = dy.size(2); s.Wo = dy.size(3);
+  const int M = s.Cout;
+  const int Nc = s.Cin * s.Kh * s.Kw;
+  const int K = s.N * s.Ho * s.Wo;
+  auto dw = torch::empty({s.Cout, s.Cin, (long)Kh, (long)Kw}, x.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int base = ceil_div(Nc, BN) * ceil_div(M, BM);
+  int splitk = 1;
+  const int max_by_k = std::max(1, K / 256);
+  while (splitk < 64 && base * splitk * 2 <= 512 && splitk * 2 <= max_by_k) {
+    splitk *= 2;
+  }
+  if (splitk > 1) {
+    const int k_chunk = ceil_div(K, splitk * BK) * BK;
+    auto ws = torch::empty({splitk, M, Nc}, x.options());
+    dim3 grid(ceil_div(Nc, BN), ceil_div(M, BM), splitk);
+    hipLaunchKernelGGL((conv_gemm_wgrad_kernel<true>), grid, dim3(256), 0,
+                       stream, x.data_ptr<float>(), dy.data_ptr<float>(),
+                       ws.data_ptr<float>(), s, M, Nc, K, k_chunk);
+    const long n = (long)M * Nc;
+    const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
+    hipLaunchKernelGGL(conv_splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
+                       stream, ws.data_ptr<float>(), dw.data_ptr<float>(), n,
+                       splitk);
+  } else {
+    dim3 grid(ceil_div(Nc, BN), ceil_div(M, BM));
+    hipLaunchKernelGGL((conv_gemm_wgrad_kernel<false>), grid, dim3(256), 0,
+                       stream, x.data_ptr<float>(), dy.data_ptr<float>(),
+                       dw.data_ptr<float>(), s, M, Nc, K, K);
+  }
+  return dw;
+}

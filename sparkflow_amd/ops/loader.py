@@ -21,6 +21,7 @@ SOURCES = [
     HERE / "hip" / "losses.hip",
     HERE / "hip" / "optim.hip",
     HERE / "hip" / "conv.hip",
+    HERE / "hip" / "conv_gemm.hip",
 ]
 
 _ext = None

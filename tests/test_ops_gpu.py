@@ -149,9 +149,14 @@ def test_conv2d_gpu(conf):
                                                 padding=pad))
     _cmp(y, ref, atol=1e-3, rtol=1e-3)
 
+    # implicit-GEMM MFMA forward must match the direct kernel's reference
+    y2 = ops.hip.conv_gemm_fwd(x.to(DEV), w.to(DEV), b.to(DEV), stride, pad, 1)
+    _cmp(y2, ref, atol=1e-3, rtol=1e-3)
+
     dy = torch.randn_like(ref)
     dx = ops.hip.conv2d_dgrad(dy.to(DEV), w.to(DEV), H, W, stride, pad)
-    dw = ops.hip.conv2d_wgrad(x.to(DEV), dy.to(DEV), K, K, stride, pad)
+    dw = ops.hip.conv_gemm_wgrad(x.to(DEV), dy.to(DEV), K, K, stride, pad)
+    dw_direct = ops.hip.conv2d_wgrad(x.to(DEV), dy.to(DEV), K, K, stride, pad)
     db = ops.hip.conv_bias_grad(dy.to(DEV))
     ref_dx = torch.nn.grad.conv2d_input(x.shape, w, dy, stride=stride,
                                         padding=pad)
@@ -159,21 +164,23 @@ def test_conv2d_gpu(conf):
                                          padding=pad)
     _cmp(dx, ref_dx, atol=1e-3, rtol=1e-3)
     _cmp(dw, ref_dw, atol=2e-3, rtol=1e-3)
+    _cmp(dw_direct, ref_dw, atol=2e-3, rtol=1e-3)
     _cmp(db, dy.sum(dim=(0, 2, 3)), atol=2e-3, rtol=1e-3)
 
 
-@pytest.mark.parametrize("k,s,hw", [(2, 2, 28), (3, 2, 112), (2, 2, 9)])
-def test_maxpool_gpu(k, s, hw):
+@pytest.mark.parametrize("k,s,p,hw", [(2, 2, 0, 28), (3, 2, 0, 112),
+                                      (2, 2, 0, 9), (3, 2, 1, 112)])
+def test_maxpool_gpu(k, s, p, hw):
     torch.manual_seed(8)
     x = torch.randn(3, 4, hw, hw)
-    y, idx = ops.hip.maxpool2d_fwd(x.to(DEV), k, s)
-    ref, ref_idx = torch.nn.functional.max_pool2d(x, k, s,
+    y, idx = ops.hip.maxpool2d_fwd(x.to(DEV), k, s, p)
+    ref, ref_idx = torch.nn.functional.max_pool2d(x, k, s, padding=p,
                                                   return_indices=True)
     _cmp(y, ref)
     dy = torch.randn_like(ref)
     dx = ops.hip.maxpool2d_bwd(dy.to(DEV), idx, hw, hw)
     x2 = x.clone().requires_grad_()
-    (torch.nn.functional.max_pool2d(x2, k, s) * dy).sum().backward()
+    (torch.nn.functional.max_pool2d(x2, k, s, padding=p) * dy).sum().backward()
     _cmp(dx, x2.grad, atol=1e-5)
 
 
