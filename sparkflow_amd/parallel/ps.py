@@ -149,6 +149,15 @@ class PSCore:
             "update_count": self.update_count,
         }
 
+    def save(self, path: str) -> None:
+        """Mid-training PS checkpoint (weights + optimizer state + step) —
+        an addition over the reference, which persists only final weights."""
+        torch.save(self.state_dict(), path)
+
+    def restore(self, path: str) -> None:
+        self.load_state_dict(torch.load(path, map_location="cpu",
+                                        weights_only=False))
+
     def load_state_dict(self, state):
         with torch.no_grad():
             self.weights.copy_(state["weights"].to(self.device))

@@ -109,6 +109,11 @@ class Worker:
     # -- the loop ------------------------------------------------------------
 
     def run(self, features: np.ndarray, labels: Optional[np.ndarray]) -> None:
+        import time
+
+        from sparkflow_amd.utils.metrics import WorkerStats
+
+        self.stats = WorkerStats()
         self.stage(features, labels)
         cfg = self.config
         n = self.x.shape[0]
@@ -117,6 +122,7 @@ class Worker:
             mbs = n - 1  # reference clamp (ml_util.py:105-106)
 
         for it in range(cfg.iters):
+            _iter_t0 = time.perf_counter()
             self._pull_and_load()
 
             if cfg.shuffle:
@@ -139,6 +145,13 @@ class Worker:
             else:
                 self._push(self._grad_step(self.x, self.y))
 
+            samples_this_iter = (
+                cfg.mini_stochastic_iters * mbs
+                if (cfg.mini_stochastic_iters >= 1 and mbs >= 1)
+                else n)
+            self.stats.record_iter(time.perf_counter() - _iter_t0,
+                                   samples_this_iter)
+
             if cfg.verbose or self.loss_callback:
                 with torch.no_grad():
                     out = self.module(self.x)
@@ -147,6 +160,10 @@ class Worker:
                 if cfg.verbose:
                     print("Partition Id: %s, Iteration: %i, Loss: %f"
                           % (self.partition_id, it, loss))
+                if cfg.verbose >= 2:
+                    print("Partition Id: %s, samples/sec: %.1f"
+                          % (self.partition_id,
+                             self.stats.samples.rate()))
                 if self.loss_callback:
                     self.loss_callback(loss, it, self.partition_id)
 
