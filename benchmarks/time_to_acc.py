@@ -42,7 +42,8 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--target", type=float, default=0.97)
     ap.add_argument("--mbs", type=int, default=300)
-    ap.add_argument("--noise", type=float, default=6.0)
+    ap.add_argument("--noise", type=float, default=5.0)
+    ap.add_argument("--lr", type=float, default=0.003)
     ap.add_argument("--round-steps", type=int, default=20)
     ap.add_argument("--max-rounds", type=int, default=200)
     ap.add_argument("--train-rows", type=int, default=6000)
@@ -61,7 +62,7 @@ def main():
     cfg = TrainConfig(iters=args.max_rounds * args.round_steps,
                       mini_batch=args.mbs, mini_stochastic_iters=1,
                       shuffle=False)
-    engine = build_engine(graph_json, OptimizerSpec("adam", 0.001), cfg)
+    engine = build_engine(graph_json, OptimizerSpec("adam", args.lr), cfg)
     device = engine.device
 
     templates = make_templates()

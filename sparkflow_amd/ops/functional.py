@@ -57,10 +57,11 @@ class LinearAct(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, y = ctx.saved_tensors
         act = ctx.act
+        need_dx = ctx.needs_input_grad[0]
         dy = dy.contiguous()
         if ops.use_hip(x):
             dpre = ops.hip.act_bwd(y, dy, act) if act != ACT_NONE else dy
-            dx = ops.hip.gemm_dgrad(dpre, w)
+            dx = ops.hip.gemm_dgrad(dpre, w) if need_dx else None
             if ctx.has_bias:
                 # bias grad rides the wgrad GEMM as a virtual ones-row
                 dwb = ops.hip.gemm_wgrad_bias(x, dpre)
@@ -70,7 +71,7 @@ class LinearAct(torch.autograd.Function):
                 dw, db = ops.hip.gemm_wgrad(x, dpre), None
         else:
             dpre = _act_bwd_from_out(y, dy, act)
-            dx = dpre @ w.t()
+            dx = dpre @ w.t() if need_dx else None
             dw = x.t() @ dpre
             db = dpre.sum(dim=0) if ctx.has_bias else None
         return dx, dw, db, None
@@ -189,17 +190,29 @@ class Conv2dAct(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, y = ctx.saved_tensors
         stride, pad, act, has_bias = ctx.conf
+        need_dx = ctx.needs_input_grad[0]
         dy = dy.contiguous()
         if ops.use_hip(x):
             dpre = ops.hip.act_bwd(y, dy, act) if act != ACT_NONE else dy
-            dx = ops.hip.conv2d_dgrad(dpre, w, x.shape[2], x.shape[3], stride, pad)
+            if not need_dx:
+                dx = None
+            elif stride == 1:
+                # stride-1 dgrad IS a conv: dx = corr(dy, W swapped+rotated)
+                # with padding K-1-p — run it on the implicit-GEMM MFMA path
+                wt = w.transpose(0, 1).flip(2, 3).contiguous()
+                dx = ops.hip.conv_gemm_fwd(
+                    dpre, wt, torch.empty(0, device=x.device), 1,
+                    w.shape[2] - 1 - pad, ACT_NONE)
+            else:
+                dx = ops.hip.conv2d_dgrad(dpre, w, x.shape[2], x.shape[3],
+                                          stride, pad)
             dw = ops.hip.conv_gemm_wgrad(x, dpre, w.shape[2], w.shape[3],
                                          stride, pad)
             db = ops.hip.conv_bias_grad(dpre) if has_bias else None
         else:
             dpre = _act_bwd_from_out(y, dy, act)
             dx = torch.nn.grad.conv2d_input(x.shape, w, dpre, stride=stride,
-                                            padding=pad)
+                                            padding=pad) if need_dx else None
             dw = torch.nn.grad.conv2d_weight(x, w.shape, dpre, stride=stride,
                                              padding=pad)
             db = dpre.sum(dim=(0, 2, 3)) if has_bias else None

@@ -46,19 +46,18 @@ class Worker:
         self.rng = np.random.default_rng(config.seed)
         self.steps_done = 0  # pushes performed (for benchmarking)
 
-        # Params and grads live as VIEWS into two flat fp32 buffers: a
-        # weight pull is ONE device copy into flat_params and gradient
-        # packing is free (autograd accumulates straight into flat_grads)
-        # — vs one copy per tensor per step in the reference's
-        # set-weights/collect-grads loops (ml_util.py:16-28).
+        # Params live as VIEWS into one flat fp32 buffer: a weight pull is
+        # ONE device copy (vs one copy per tensor per step in the
+        # reference's set-weights loop, ml_util.py:16-28).  Grads are
+        # packed by a single cat into a stable flat buffer after backward.
+        self._params = param_list(self.module)
         self.flat_params = pack_params(self.module)
         self.flat_grads = torch.zeros_like(self.flat_params)
         off = 0
         with torch.no_grad():
-            for p in param_list(self.module):
+            for p in self._params:
                 n = p.numel()
                 p.data = self.flat_params[off:off + n].view(p.shape)
-                p.grad = self.flat_grads[off:off + n].view(p.shape)
                 off += n
 
     # -- data staging --------------------------------------------------------
@@ -85,11 +84,13 @@ class Worker:
             self.flat_params.copy_(flat)
 
     def _grad_step(self, xb: torch.Tensor, yb: Optional[torch.Tensor]):
-        self.flat_grads.zero_()
+        self.module.zero_grad(set_to_none=True)
         out = self.module(xb)
         target = yb if yb is not None else xb
         loss = self.module.compute_loss(out, target)
         loss.backward()
+        torch.cat([p.grad.reshape(-1) for p in self._params],
+                  out=self.flat_grads)
         return self.flat_grads
 
     def _push(self, grads: torch.Tensor):
