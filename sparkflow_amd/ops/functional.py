@@ -19,6 +19,15 @@ ACT_CODES = {"none": ACT_NONE, "linear": ACT_NONE, "relu": ACT_RELU,
              "sigmoid": ACT_SIGMOID}
 
 
+def _dev_scale(dloss: torch.Tensor) -> torch.Tensor:
+    """Upstream loss grad as a device fp32 scalar for the HIP bwd kernels.
+
+    Kept on-device (the kernel dereferences it) so the backward pass never
+    d2h-syncs — a host `float(dloss)` would break hipGraph capture of the
+    full training step."""
+    return dloss.detach().to(dtype=torch.float32).contiguous()
+
+
 def _act_fwd(pre: torch.Tensor, act: int) -> torch.Tensor:
     if act == ACT_RELU:
         return torch.relu(pre)
@@ -101,7 +110,8 @@ class SoftmaxXent(torch.autograd.Function):
         probs, onehot = ctx.saved_tensors
         m = probs.shape[0]
         if ops.use_hip(probs):
-            dlogits = ops.hip.softmax_xent_bwd(probs, onehot, float(dloss) / m)
+            dlogits = ops.hip.softmax_xent_bwd(probs, onehot,
+                                               _dev_scale(dloss), 1.0 / m)
         else:
             dlogits = (probs - onehot) * (dloss / m)
         return dlogits, None
@@ -129,7 +139,7 @@ class SoftmaxXentIdx(torch.autograd.Function):
         m = probs.shape[0]
         if ops.use_hip(probs):
             dlogits = ops.hip.softmax_xent_idx_bwd(probs, labels,
-                                                   float(dloss) / m)
+                                                   _dev_scale(dloss), 1.0 / m)
         else:
             dlogits = probs.clone()
             dlogits[torch.arange(m, device=probs.device), labels] -= 1.0
@@ -160,7 +170,7 @@ class MseLoss(torch.autograd.Function):
         pred, target = ctx.saved_tensors
         n = pred.numel()
         if ops.use_hip(pred):
-            dpred = ops.hip.mse_bwd(pred, target, 2.0 * float(dloss) / n)
+            dpred = ops.hip.mse_bwd(pred, target, _dev_scale(dloss), 2.0 / n)
         else:
             dpred = (pred - target) * (2.0 * dloss / n)
         return dpred, None

@@ -19,13 +19,14 @@ torch::Tensor argmax_rows(torch::Tensor x);
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
                                             torch::Tensor onehot);
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,
-                               double gscale);
+                               torch::Tensor gscale, double mult);
 std::vector<torch::Tensor> softmax_xent_idx_fwd(torch::Tensor logits,
                                                 torch::Tensor labels);
 torch::Tensor softmax_xent_idx_bwd(torch::Tensor probs, torch::Tensor labels,
-                                   double gscale);
+                                   torch::Tensor gscale, double mult);
 torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target);
-torch::Tensor mse_bwd(torch::Tensor pred, torch::Tensor target, double gscale);
+torch::Tensor mse_bwd(torch::Tensor pred, torch::Tensor target,
+                      torch::Tensor gscale, double mult);
 
 // optim.hip
 void fused_sgd(torch::Tensor p, torch::Tensor g, double lr);

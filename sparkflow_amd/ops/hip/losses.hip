@@ -74,10 +74,15 @@ __global__ void softmax_xent_idx_fwd_kernel(const float* __restrict__ logits,
   }
 }
 
+// The upstream loss grad arrives as a DEVICE scalar (gscale) and is read
+// on-device: converting it to a host float would d2h-sync and break
+// hipGraph stream capture of the full training step.
 __global__ void softmax_xent_idx_bwd_kernel(const float* __restrict__ probs,
                                             const long* __restrict__ labels,
                                             float* __restrict__ dlogits,
-                                            float gscale, int M, int C) {
+                                            const float* __restrict__ gscale,
+                                            float mult, int M, int C) {
+  const float gs = gscale[0] * mult;
   const long n = (long)M * C;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -85,18 +90,20 @@ __global__ void softmax_xent_idx_bwd_kernel(const float* __restrict__ probs,
     const int c = i % C;
     const int row = i / C;
     const float onehot = (c == (int)labels[row]) ? 1.0f : 0.0f;
-    dlogits[i] = (probs[i] - onehot) * gscale;
+    dlogits[i] = (probs[i] - onehot) * gs;
   }
 }
 
 __global__ void softmax_xent_bwd_kernel(const float* __restrict__ probs,
                                         const float* __restrict__ onehot,
                                         float* __restrict__ dlogits,
-                                        float gscale, long n) {
+                                        const float* __restrict__ gscale,
+                                        float mult, long n) {
+  const float gs = gscale[0] * mult;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    dlogits[i] = (probs[i] - onehot[i]) * gscale;
+    dlogits[i] = (probs[i] - onehot[i]) * gs;
   }
 }
 
@@ -124,11 +131,14 @@ __global__ void mse_fwd_kernel(const float* __restrict__ p,
 
 __global__ void mse_bwd_kernel(const float* __restrict__ p,
                                const float* __restrict__ t,
-                               float* __restrict__ dp, float gscale, long n) {
+                               float* __restrict__ dp,
+                               const float* __restrict__ gscale, float mult,
+                               long n) {
+  const float gs = gscale[0] * mult;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    dp[i] = (p[i] - t[i]) * gscale;
+    dp[i] = (p[i] - t[i]) * gs;
   }
 }
 
@@ -173,8 +183,9 @@ std::vector<torch::Tensor> softmax_xent_idx_fwd(torch::Tensor logits,
 }
 
 torch::Tensor softmax_xent_idx_bwd(torch::Tensor probs, torch::Tensor labels,
-                                   double gscale) {
+                                   torch::Tensor gscale, double mult) {
   check_f32l(probs, "probs");
+  check_f32l(gscale, "gscale");
   const int M = probs.size(0), C = probs.size(1);
   auto d = torch::empty_like(probs);
   const long n = probs.numel();
@@ -182,21 +193,24 @@ torch::Tensor softmax_xent_idx_bwd(torch::Tensor probs, torch::Tensor labels,
   hipLaunchKernelGGL(softmax_xent_idx_bwd_kernel, dim3(blocks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(),
                      probs.data_ptr<float>(), labels.data_ptr<long>(),
-                     d.data_ptr<float>(), (float)gscale, M, C);
+                     d.data_ptr<float>(), gscale.data_ptr<float>(),
+                     (float)mult, M, C);
   return d;
 }
 
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,
-                               double gscale) {
+                               torch::Tensor gscale, double mult) {
   check_f32l(probs, "probs");
   check_f32l(onehot, "onehot");
+  check_f32l(gscale, "gscale");
   auto d = torch::empty_like(probs);
   const long n = probs.numel();
   const int blocks = (int)std::min<long>(4096, (n + 255) / 256);
   hipLaunchKernelGGL(softmax_xent_bwd_kernel, dim3(blocks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(),
                      probs.data_ptr<float>(), onehot.data_ptr<float>(),
-                     d.data_ptr<float>(), (float)gscale, n);
+                     d.data_ptr<float>(), gscale.data_ptr<float>(),
+                     (float)mult, n);
   return d;
 }
 
@@ -214,15 +228,16 @@ torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target) {
 }
 
 torch::Tensor mse_bwd(torch::Tensor pred, torch::Tensor target,
-                      double gscale) {
+                      torch::Tensor gscale, double mult) {
   check_f32l(pred, "pred");
   check_f32l(target, "target");
+  check_f32l(gscale, "gscale");
   auto d = torch::empty_like(pred);
   const long n = pred.numel();
   const int blocks = (int)std::min<long>(4096, (n + 255) / 256);
   hipLaunchKernelGGL(mse_bwd_kernel, dim3(blocks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(), pred.data_ptr<float>(),
                      target.data_ptr<float>(), d.data_ptr<float>(),
-                     (float)gscale, n);
+                     gscale.data_ptr<float>(), (float)mult, n);
   return d;
 }

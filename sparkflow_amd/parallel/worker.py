@@ -182,12 +182,15 @@ class Worker:
         """
         n = self.x.shape[0]
         mbs = min(mini_batch, n)
+        explicit = use_graph is True
         if use_graph is None:
             use_graph = self.device.type == "cuda"
         if use_graph and self.device.type == "cuda":
             try:
                 return self._run_steps_graphed(k, mbs)
             except Exception as e:
+                if explicit:  # caller demanded the graph path: fail loudly
+                    raise
                 print(f"[sparkflow_amd] hipGraph capture failed ({e}); "
                       "running eager steps")
         for _ in range(k):
