@@ -100,7 +100,8 @@ def test_softmax_xent_gpu():
     torch.testing.assert_close(loss.cpu(), ref, atol=1e-4, rtol=1e-4)
     torch.testing.assert_close(probs.cpu(), torch.softmax(logits, 1),
                                atol=1e-5, rtol=1e-4)
-    d = ops.hip.softmax_xent_bwd(probs, onehot.to(DEV), 1.0 / 300)
+    one = torch.ones((), device=DEV)
+    d = ops.hip.softmax_xent_bwd(probs, onehot.to(DEV), one, 1.0 / 300)
     ref_logits = logits.clone().requires_grad_()
     torch.nn.functional.cross_entropy(ref_logits, labels).backward()
     torch.testing.assert_close(d.cpu(), ref_logits.grad, atol=1e-5, rtol=1e-4)
@@ -114,7 +115,8 @@ def test_mse_gpu():
                                torch.nn.functional.mse_loss(p, t),
                                atol=1e-5, rtol=1e-4)
     n = p.numel()
-    d = ops.hip.mse_bwd(p.to(DEV), t.to(DEV), 2.0 / n)
+    d = ops.hip.mse_bwd(p.to(DEV), t.to(DEV), torch.ones((), device=DEV),
+                        2.0 / n)
     torch.testing.assert_close(d.cpu(), 2 * (p - t) / n, atol=1e-7, rtol=1e-5)
 
 
