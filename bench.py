@@ -39,7 +39,12 @@ def main():
     ap.add_argument("--lr", type=float, default=0.001)
     ap.add_argument("--lock", action="store_true",
                     help="async-with-lock mode instead of Hogwild")
+    ap.add_argument("--workers", type=int, default=4,
+                    help="concurrent Hogwild workers per GPU (reference "
+                    "canonical config: 4 partitions)")
     args = ap.parse_args()
+    if args.lock:
+        args.workers = 1
 
     from sparkflow_amd import models
     from sparkflow_amd.graph_utils import build_graph
@@ -56,7 +61,7 @@ def main():
                       mini_stochastic_iters=1, shuffle=False,
                       acquire_lock=args.lock)
     engine = build_engine(graph_json, OptimizerSpec(args.optimizer, args.lr),
-                          cfg)
+                          cfg, workers_per_gpu=args.workers)
     device = engine.device
 
     # synthetic data of the benchmark config's shape
@@ -72,10 +77,10 @@ def main():
     else:  # autoencoder: unsupervised
         x = rng.random((n_rows, 784), dtype=np.float32)
         y = None
-    engine.worker.stage(x, y)
+    engine.stage_all(x, y)
 
     # warmup (also initializes the RCCL pair communicators)
-    engine.worker.run_steps(args.warmup, args.mbs)
+    engine.run_steps(args.warmup, args.mbs)
     if device.type == "cuda":
         torch.cuda.synchronize()
     engine.barrier()
@@ -83,7 +88,7 @@ def main():
         torch.cuda.synchronize()
 
     t0 = time.perf_counter()
-    engine.worker.run_steps(args.steps, args.mbs)
+    timed_samples = engine.run_steps(args.steps, args.mbs)
     if device.type == "cuda":
         torch.cuda.synchronize()
     engine.barrier()
@@ -101,7 +106,7 @@ def main():
         elapsed = float(t.item())
 
     mbs_eff = min(args.mbs, x.shape[0])
-    total_samples = args.steps * mbs_eff * world
+    total_samples = timed_samples * world  # every rank does the same work
     value = total_samples / elapsed
 
     if rank == 0:
@@ -123,10 +128,12 @@ def main():
                           "mnist_cnn": "mnist_cnn_5x5x32_3x3x64",
                           "tabular_autoencoder": "autoencoder_784_32_784",
                           "resnet18_shape": "resnet18_shape_3x224x224"}[args.model],
-                "global_batch": mbs_eff * world,
+                "global_batch": mbs_eff * world * args.workers,
                 "seq_len": None,
-                "parallelism": ("async-ps-%s dp%d"
-                                % ("lock" if args.lock else "hogwild", world)),
+                "workers_per_gpu": args.workers,
+                "parallelism": ("async-ps-%s dp%d x %dw"
+                                % ("lock" if args.lock else "hogwild", world,
+                                   args.workers)),
             },
         }
         print(json.dumps(result))

@@ -79,12 +79,60 @@ class EngineHandle:
     worker: Worker
     ps: Optional[PSCore]
     service: Optional[DistPSService]
+    workers: Optional[list] = None
+
+    def __post_init__(self):
+        if self.workers is None:
+            self.workers = [self.worker]
 
     def barrier(self):
         import torch.distributed as dist
 
         if self.world > 1:
             dist.barrier()
+
+    def run_steps(self, k: int, mini_batch: int) -> int:
+        """k async-PS steps on EVERY worker of this rank.
+
+        With several workers per GPU each worker replays its own captured
+        hipGraph on its own HIP stream — W concurrent Hogwild workers
+        racing pulls and fused-optimizer applies on the shared PS buffer,
+        the GPU-resident equivalent of the reference's N partitions
+        hitting the threaded Flask PS (HogwildSparkModel.py:244).  One
+        host thread drives all streams (replays are async enqueues)."""
+        if len(self.workers) == 1:
+            return self.worker.run_steps(k, mini_batch)
+        if self.device.type == "cuda":
+            for w in self.workers:
+                w.ensure_graph(mini_batch)
+            streams = getattr(self, "_streams", None)
+            if streams is None or len(streams) != len(self.workers):
+                streams = [torch.cuda.Stream(device=self.device)
+                           for _ in self.workers]
+                self._streams = streams
+            cur = torch.cuda.current_stream(self.device)
+            for s in streams:
+                s.wait_stream(cur)
+            for _ in range(k):
+                for w, s in zip(self.workers, streams):
+                    with torch.cuda.stream(s):
+                        w.replay_once()
+            for s in streams:
+                cur.wait_stream(s)
+            total = 0
+            for w in self.workers:
+                w.finish_steps(k)
+                total += k * min(mini_batch, w.x.shape[0])
+            return total
+        total = 0
+        for _ in range(k):  # CPU: interleave eager single steps
+            for w in self.workers:
+                total += w.run_steps(1, mini_batch, use_graph=False)
+        return total
+
+    def stage_all(self, x, y) -> None:
+        for w in self.workers:
+            w.stage(x, y)
 
     def close(self):
         if self.rank != 0:
@@ -95,8 +143,14 @@ class EngineHandle:
 
 def build_engine(graph_json: str, optimizer: OptimizerSpec,
                  config: TrainConfig, device=None,
-                 loss_callback=None) -> EngineHandle:
-    """Construct PS + worker for this rank (call on every rank)."""
+                 loss_callback=None, workers_per_gpu: int = 1) -> EngineHandle:
+    """Construct PS + worker(s) for this rank (call on every rank).
+
+    ``workers_per_gpu`` > 1 runs that many concurrent Hogwild workers on
+    this rank's GPU (the reference's multiple-partitions-per-host model,
+    e.g. the canonical README config's 4 partitions on local[4]); on the
+    PS rank each gets its own LocalTransport, on worker ranks they share
+    the rank's RCCL pair link (driven sequentially by one host thread)."""
     rank, world, device = init_from_env(device=device)
 
     ps = service = None
@@ -109,12 +163,22 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
                     max_errors=config.iters)
         if world > 1:
             service = DistPSService(ps, pairs).start()
-        transport = LocalTransport(ps, device)
+        transports = [LocalTransport(ps, device)
+                      for _ in range(max(1, workers_per_gpu))]
     else:
         numel = flat_size(module_from_json(graph_json))
-        transport = DistTransport(pairs, numel, device)
+        shared = DistTransport(pairs, numel, device)
+        transports = [shared] * max(1, workers_per_gpu)
 
-    worker = Worker(graph_json, transport, config, device=device,
-                    loss_callback=loss_callback)
-    return EngineHandle(rank=rank, world=world, device=device, worker=worker,
-                        ps=ps, service=service)
+    workers = []
+    for i, tr in enumerate(transports):
+        cfg_i = config
+        if config.seed is not None and i > 0:
+            from dataclasses import replace
+
+            cfg_i = replace(config, seed=config.seed + i)
+        workers.append(Worker(graph_json, tr, cfg_i, device=device,
+                              loss_callback=loss_callback))
+    return EngineHandle(rank=rank, world=world, device=device,
+                        worker=workers[0], ps=ps, service=service,
+                        workers=workers)

@@ -206,6 +206,60 @@ class Worker:
 
     _IDX_POOL_CAP = 2048
 
+    def ensure_graph(self, mini_batch: int) -> None:
+        """Capture (once) the hipGraph for steps of ``mini_batch`` and fill
+        the device-side minibatch index pool.  Idempotent per config."""
+        from sparkflow_amd.parallel.transport import LocalTransport
+
+        n = self.x.shape[0]
+        mbs = min(mini_batch, n)
+        sub = mbs < n
+        full_step = (isinstance(self.transport, LocalTransport)
+                     and not self.transport.ps.acquire_lock
+                     and self.transport.ps.device == self.device)
+        key = (mbs, sub, full_step, n, self.x.data_ptr())
+        if getattr(self, "_graph_key", None) != key:
+            self._capture_graph(key, mbs, sub, full_step)
+            if sub:
+                # Fill the pool ONCE with pre-drawn sample-without-
+                # replacement minibatches; the captured ctr walks it
+                # cyclically.  Keeping the fill out of run_steps keeps
+                # host-side RNG out of the steady-state loop.
+                g = self._graph_state
+                picks = np.stack([self.rng.choice(n, mbs, replace=False)
+                                  for _ in range(self._IDX_POOL_CAP)])
+                g["idx_pool"].copy_(torch.as_tensor(picks,
+                                                    device=self.device))
+                g["ctr"].zero_()
+
+    def replay_once(self) -> None:
+        """Enqueue ONE training step on the current HIP stream (graph must
+        be captured via ensure_graph).  Full-step mode: the replay IS the
+        whole pull->gather->fwd/bwd->apply step.  Remote mode: eager pull,
+        compute replay, eager push."""
+        g = self._graph_state
+        if g["full_step"]:
+            g["graph"].replay()
+            return
+        n = self.x.shape[0]
+        self._pull_and_load()
+        if g["xb"] is not self.x:
+            pick = self._minibatch_indices(n, g["xb"].shape[0])
+            torch.index_select(self.x, 0, pick, out=g["xb"])
+            if g["yb"] is not None:
+                torch.index_select(self.y, 0, pick, out=g["yb"])
+        g["graph"].replay()
+        self._push(self.flat_grads)
+
+    def finish_steps(self, k: int) -> None:
+        """Account k replayed full steps (host-side counters)."""
+        g = self._graph_state
+        if g["full_step"]:
+            ps = self.transport.ps
+            ps.update_count += k
+            ps.pull_count += k
+            self.steps_done += k
+
     def _run_steps_graphed(self, k: int, mbs: int) -> int:
         """One hipGraph replay per step.
 
@@ -218,45 +272,10 @@ class Worker:
         (RCCL/TCP) transports only the compute is captured; pull/push stay
         eager.
         """
-        from sparkflow_amd.parallel.transport import LocalTransport
-
-        n = self.x.shape[0]
-        sub = mbs < n
-        full_step = (isinstance(self.transport, LocalTransport)
-                     and not self.transport.ps.acquire_lock
-                     and self.transport.ps.device == self.device)
-
-        key = (mbs, sub, full_step, n, self.x.data_ptr())
-        if getattr(self, "_graph_key", None) != key:
-            self._capture_graph(key, mbs, sub, full_step)
-
-        g = self._graph_state
-        # refill the index pool with this call's picks
-        if sub:
-            rows = min(k, self._IDX_POOL_CAP)
-            picks = np.stack([self.rng.choice(n, mbs, replace=False)
-                              for _ in range(rows)])
-            g["idx_pool"][:rows].copy_(
-                torch.as_tensor(picks, device=self.device))
-            g["ctr"].zero_()
-
-        ps = self.transport.ps if full_step else None
+        self.ensure_graph(mbs)
         for _ in range(k):
-            if full_step:
-                g["graph"].replay()
-            else:
-                self._pull_and_load()
-                if sub:
-                    pick = self._minibatch_indices(n, mbs)
-                    torch.index_select(self.x, 0, pick, out=g["xb"])
-                    if g["yb"] is not None:
-                        torch.index_select(self.y, 0, pick, out=g["yb"])
-                g["graph"].replay()
-                self._push(self.flat_grads)
-        if full_step:
-            ps.update_count += k
-            ps.pull_count += k
-            self.steps_done += k
+            self.replay_once()
+        self.finish_steps(k)
         return k * mbs
 
     def _capture_graph(self, key, mbs: int, sub: bool, full_step: bool):
@@ -308,7 +327,8 @@ class Worker:
             body()
 
         self._graph_state = {"graph": graph, "xb": xb, "yb": yb,
-                             "idx_pool": idx_pool, "ctr": ctr}
+                             "idx_pool": idx_pool, "ctr": ctr,
+                             "full_step": full_step}
         self._graph_key = key
 
 

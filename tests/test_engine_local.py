@@ -154,3 +154,32 @@ def test_final_weights_json_roundtrip(tcp_port):
     weights = _train(rows2, graph_json, tcp_port, iters=5)
     js = convert_weights_to_json(weights)
     assert isinstance(js, str) and js.startswith("[")
+
+
+def test_multi_worker_engine_cpu():
+    """workers_per_gpu>1: W concurrent Hogwild workers share one PS
+    (reference: several partitions hitting the threaded Flask PS)."""
+    import numpy as np
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import mnist_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    cfg = TrainConfig(iters=1, mini_batch=32, mini_stochastic_iters=1,
+                      seed=11)
+    eng = build_engine(build_graph(mnist_mlp),
+                       OptimizerSpec("adam", 0.001), cfg, device="cpu",
+                       workers_per_gpu=3)
+    assert len(eng.workers) == 3
+    rng = np.random.default_rng(0)
+    x = rng.random((256, 784), dtype=np.float32)
+    y = rng.integers(0, 10, (256, 1)).astype(np.float32)
+    eng.stage_all(x, y)
+    samples = eng.run_steps(4, 32)
+    assert samples == 4 * 32 * 3
+    assert eng.ps.update_count == 12
+    # per-worker seeds differ -> different minibatch streams
+    assert eng.workers[0].rng.integers(0, 1 << 30) != \
+        eng.workers[1].rng.integers(0, 1 << 30)
+    eng.close()
