@@ -10,11 +10,35 @@ GEMM+bias+activation, conv2d, maxpool, softmax-cross-entropy, MSE, argmax.
 """
 from __future__ import annotations
 
+import contextlib
+import threading
+
 import torch
 
 from sparkflow_amd import ops
 
 ACT_NONE, ACT_RELU, ACT_SIGMOID = 0, 1, 2
+
+_TLS = threading.local()
+
+
+def _want_loss_value() -> bool:
+    return not getattr(_TLS, "grad_only", False)
+
+
+@contextlib.contextmanager
+def grad_only():
+    """Inside this context the loss ops may skip computing the scalar loss
+    VALUE (it is undefined) while producing exact gradients — used by the
+    training hot loop, where the value is never read.  Saves the
+    zero-fill / atomic-reduce / mean kernels per step (GPU path only; the
+    CPU path always computes the value)."""
+    prev = getattr(_TLS, "grad_only", False)
+    _TLS.grad_only = True
+    try:
+        yield
+    finally:
+        _TLS.grad_only = prev
 ACT_CODES = {"none": ACT_NONE, "linear": ACT_NONE, "relu": ACT_RELU,
              "sigmoid": ACT_SIGMOID}
 
@@ -96,7 +120,8 @@ class SoftmaxXent(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, onehot):
         if ops.use_hip(logits):
-            loss, probs = ops.hip.softmax_xent_fwd(logits, onehot)
+            loss, probs = ops.hip.softmax_xent_fwd(logits, onehot,
+                                                   _want_loss_value())
         else:
             lse = torch.logsumexp(logits, dim=1, keepdim=True)
             logp = logits - lse
@@ -124,7 +149,8 @@ class SoftmaxXentIdx(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, labels):
         if ops.use_hip(logits):
-            loss, probs = ops.hip.softmax_xent_idx_fwd(logits, labels)
+            loss, probs = ops.hip.softmax_xent_idx_fwd(logits, labels,
+                                                       _want_loss_value())
         else:
             lse = torch.logsumexp(logits, dim=1, keepdim=True)
             logp = logits - lse
@@ -162,7 +188,7 @@ class MseLoss(torch.autograd.Function):
     def forward(ctx, pred, target):
         ctx.save_for_backward(pred, target)
         if ops.use_hip(pred):
-            return ops.hip.mse_fwd(pred, target)
+            return ops.hip.mse_fwd(pred, target, _want_loss_value())
         return ((pred - target) ** 2).mean()
 
     @staticmethod

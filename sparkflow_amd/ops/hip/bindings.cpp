@@ -4,6 +4,8 @@
 #include <vector>
 
 // gemm.hip
+int sf_parallel_hint();
+void sf_set_parallel_hint(int n);
 torch::Tensor gemm_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                        int64_t act);
 torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w);
@@ -14,17 +16,24 @@ torch::Tensor bias_grad(torch::Tensor dy);
 // elementwise.hip
 torch::Tensor act_bwd(torch::Tensor y, torch::Tensor dy, int64_t act);
 torch::Tensor argmax_rows(torch::Tensor x);
+void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
+                  torch::Tensor ctr, torch::Tensor done,
+                  c10::optional<torch::Tensor> y,
+                  c10::optional<torch::Tensor> yb);
 
 // losses.hip
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
-                                            torch::Tensor onehot);
+                                            torch::Tensor onehot,
+                                            bool want_value);
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,
                                torch::Tensor gscale, double mult);
 std::vector<torch::Tensor> softmax_xent_idx_fwd(torch::Tensor logits,
-                                                torch::Tensor labels);
+                                                torch::Tensor labels,
+                                                bool want_value);
 torch::Tensor softmax_xent_idx_bwd(torch::Tensor probs, torch::Tensor labels,
                                    torch::Tensor gscale, double mult);
-torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target);
+torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target,
+                      bool want_value);
 torch::Tensor mse_bwd(torch::Tensor pred, torch::Tensor target,
                       torch::Tensor gscale, double mult);
 
@@ -70,14 +79,26 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_grad", &bias_grad, "column-sum bias gradient");
   m.def("act_bwd", &act_bwd, "activation backward from output");
   m.def("argmax_rows", &argmax_rows, "rowwise argmax head");
-  m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax+xent fwd");
+  m.def("gather_batch", &gather_batch,
+        "fused minibatch gather + pool-cursor advance (hipGraph step "
+        "prologue)", py::arg("x"), py::arg("xb"), py::arg("idx_pool"),
+        py::arg("ctr"), py::arg("done"), py::arg("y") = py::none(),
+        py::arg("yb") = py::none());
+  m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax+xent fwd",
+        py::arg("logits"), py::arg("onehot"), py::arg("want_value") = true);
   m.def("softmax_xent_bwd", &softmax_xent_bwd, "softmax+xent bwd");
   m.def("softmax_xent_idx_fwd", &softmax_xent_idx_fwd,
-        "fused softmax+xent fwd, class-index labels");
+        "fused softmax+xent fwd, class-index labels", py::arg("logits"),
+        py::arg("labels"), py::arg("want_value") = true);
   m.def("softmax_xent_idx_bwd", &softmax_xent_idx_bwd,
         "softmax+xent bwd, class-index labels");
-  m.def("mse_fwd", &mse_fwd, "fused MSE loss");
+  m.def("mse_fwd", &mse_fwd, "fused MSE loss", py::arg("pred"),
+        py::arg("target"), py::arg("want_value") = true);
   m.def("mse_bwd", &mse_bwd, "MSE grad");
+  m.def("set_parallel_hint", &sf_set_parallel_hint,
+        "number of concurrent worker streams sharing the GPU (shrinks "
+        "split-K)");
+  m.def("parallel_hint", &sf_parallel_hint, "current concurrency hint");
   m.def("fused_sgd", &fused_sgd, "single-launch SGD apply");
   m.def("fused_momentum", &fused_momentum, "single-launch momentum apply");
   m.def("fused_adam", &fused_adam, "single-launch Adam apply");

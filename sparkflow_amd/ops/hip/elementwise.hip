@@ -44,6 +44,79 @@ __global__ void argmax_rows_kernel(const float* __restrict__ x,
   if (lane == 0) out[row] = (float)best_idx;
 }
 
+// Fused minibatch staging for the captured training step: ONE launch
+// gathers the x rows (and labels) named by the index pool's current row
+// and advances the device-side pool cursor — replacing the
+// index_select / gather / add_ / remainder_ kernel train the eager
+// sequence needs (~5 launches).  The cursor bump uses a done-ticket:
+// every block reads the cursor at start, increments `done` at end, and
+// the last-finishing block (which can only fire after every block has
+// read) bumps the cursor and self-resets the ticket — so the kernel is
+// replay-deterministic inside a hipGraph.
+__global__ void gather_batch_kernel(const float* __restrict__ x,
+                                    float* __restrict__ xb,
+                                    const float* __restrict__ y,
+                                    float* __restrict__ yb,
+                                    const long* __restrict__ idx_pool,
+                                    long* __restrict__ ctr,
+                                    int* __restrict__ done, int mbs, long dx,
+                                    long dy_row, int cap) {
+  const long c = *ctr;
+  const long* row_idx = idx_pool + c * mbs;
+  // block b gathers x row b (grid.x == mbs)
+  const int r = blockIdx.x;
+  const long src = row_idx[r];
+  const float* xs = x + src * dx;
+  float* xd = xb + (long)r * dx;
+  for (long i = threadIdx.x; i < dx; i += blockDim.x) xd[i] = xs[i];
+  if (y != nullptr && threadIdx.x == 0) {
+    for (long j = 0; j < dy_row; ++j) yb[(long)r * dy_row + j] =
+        y[src * dy_row + j];
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    __threadfence();
+    if (atomicAdd(done, 1) == gridDim.x - 1) {
+      *ctr = (c + 1) % cap;
+      *done = 0;
+      __threadfence();
+    }
+  }
+}
+
+void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
+                  torch::Tensor ctr, torch::Tensor done,
+                  c10::optional<torch::Tensor> y,
+                  c10::optional<torch::Tensor> yb) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && xb.is_contiguous(),
+              "gather_batch: x/xb must be contiguous on GPU");
+  TORCH_CHECK(idx_pool.scalar_type() == torch::kInt64 &&
+                  ctr.scalar_type() == torch::kInt64 &&
+                  done.scalar_type() == torch::kInt32,
+              "gather_batch: idx_pool/ctr int64, done int32");
+  const int mbs = xb.size(0);
+  const long dx = xb.numel() / mbs;
+  const int cap = idx_pool.size(0);
+  TORCH_CHECK(idx_pool.size(1) == mbs, "pool width != mbs");
+  const float* yp = nullptr;
+  float* ybp = nullptr;
+  long dy_row = 0;
+  if (y.has_value() && yb.has_value()) {
+    TORCH_CHECK(y->scalar_type() == torch::kFloat32 &&
+                    yb->scalar_type() == torch::kFloat32,
+                "gather_batch: float labels only (int labels are cast by "
+                "the worker)");
+    yp = y->data_ptr<float>();
+    ybp = yb->data_ptr<float>();
+    dy_row = yb->numel() / mbs;
+  }
+  hipLaunchKernelGGL(gather_batch_kernel, dim3(mbs), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
+                     xb.data_ptr<float>(), yp, ybp,
+                     idx_pool.data_ptr<long>(), ctr.data_ptr<long>(),
+                     done.data_ptr<int>(), mbs, dx, dy_row, cap);
+}
+
 static inline void check_f32e(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
                   t.is_contiguous(),

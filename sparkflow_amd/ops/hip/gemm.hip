@@ -230,13 +230,33 @@ static inline void check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
 }
 
+static int g_parallel_hint = 1;
+
+int sf_parallel_hint() { return g_parallel_hint; }
+void sf_set_parallel_hint(int n) { g_parallel_hint = n < 1 ? 1 : n; }
+
+int sf_splitk_env_max() {
+  static int cached = -1;
+  if (cached < 0) {
+    const char* e = getenv("SPARKFLOW_SPLITK_MAX");
+    cached = e ? atoi(e) : 8;
+    if (cached < 1) cached = 1;
+  }
+  return cached;
+}
+
 // Pick a K split so the grid approaches the 256-CU mark (a launch needs
 // many more workgroups than the M/N tiling of a 300x256 layer provides).
+// With several concurrent worker streams the CUs are already shared, so
+// the target grid (and the split) shrinks proportionally.
 static int pick_splitk(int M, int N, int K) {
   const int base = ceil_div(M, BM) * ceil_div(N, BN);
   const int max_by_k = std::max(1, K / 64);  // keep >=64-deep slices
+  const int target = std::max(32, 512 / g_parallel_hint);
+  const int cap = sf_splitk_env_max();
   int splitk = 1;
-  while (splitk < 8 && base * splitk * 2 <= 512 && splitk * 2 <= max_by_k) {
+  while (splitk < 8 && splitk < cap && base * splitk * 2 <= target &&
+         splitk * 2 <= max_by_k) {
     splitk *= 2;
   }
   return splitk;

@@ -13,8 +13,8 @@
 __global__ void softmax_xent_fwd_kernel(const float* __restrict__ logits,
                                         const float* __restrict__ onehot,
                                         float* __restrict__ probs,
-                                        float* __restrict__ loss_sum, int M,
-                                        int C) {
+                                        float* __restrict__ loss_sum,
+                                        float inv_m, int M, int C) {
   const int row = blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   const int lane = threadIdx.x & (WAVE - 1);
   if (row >= M) return;
@@ -38,9 +38,9 @@ __global__ void softmax_xent_fwd_kernel(const float* __restrict__ logits,
   for (int c = lane; c < C; c += WAVE) {
     probs[(long)row * C + c] = __expf(xr[c] - mx) * inv;
   }
-  if (lane == 0) {
+  if (lane == 0 && loss_sum) {
     const float lse = mx + __logf(se);
-    atomicAdd(loss_sum, lse - dot);
+    atomicAdd(loss_sum, (lse - dot) * inv_m);
   }
 }
 
@@ -50,7 +50,7 @@ __global__ void softmax_xent_idx_fwd_kernel(const float* __restrict__ logits,
                                             const long* __restrict__ labels,
                                             float* __restrict__ probs,
                                             float* __restrict__ loss_sum,
-                                            int M, int C) {
+                                            float inv_m, int M, int C) {
   const int row = blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   const int lane = threadIdx.x & (WAVE - 1);
   if (row >= M) return;
@@ -68,9 +68,9 @@ __global__ void softmax_xent_idx_fwd_kernel(const float* __restrict__ logits,
   for (int c = lane; c < C; c += WAVE) {
     probs[(long)row * C + c] = __expf(xr[c] - mx) * inv;
   }
-  if (lane == 0) {
+  if (lane == 0 && loss_sum) {
     const float lse = mx + __logf(se);
-    atomicAdd(loss_sum, lse - xr[labels[row]]);
+    atomicAdd(loss_sum, (lse - xr[labels[row]]) * inv_m);
   }
 }
 
@@ -109,7 +109,7 @@ __global__ void softmax_xent_bwd_kernel(const float* __restrict__ probs,
 
 __global__ void mse_fwd_kernel(const float* __restrict__ p,
                                const float* __restrict__ t,
-                               float* __restrict__ out, long n) {
+                               float* __restrict__ out, float inv_n, long n) {
   const long stride = (long)gridDim.x * blockDim.x;
   float s = 0.0f;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -125,7 +125,7 @@ __global__ void mse_fwd_kernel(const float* __restrict__ p,
   if (threadIdx.x == 0) {
     float b = 0.0f;
     for (int i = 0; i < (int)(blockDim.x / WAVE); ++i) b += partials[i];
-    atomicAdd(out, b);
+    atomicAdd(out, b * inv_n);
   }
 }
 
@@ -149,36 +149,44 @@ static inline void check_f32l(const torch::Tensor& t, const char* name) {
 }
 
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
-                                            torch::Tensor onehot) {
+                                            torch::Tensor onehot,
+                                            bool want_value) {
   check_f32l(logits, "logits");
   check_f32l(onehot, "onehot");
   const int M = logits.size(0), C = logits.size(1);
   auto probs = torch::empty_like(logits);
-  auto loss = torch::zeros({}, logits.options());
+  // want_value=false (the captured hot path): the scalar is never read —
+  // skip the zero-fill/atomic/mean kernels and return an undefined scalar.
+  auto loss = want_value ? torch::zeros({}, logits.options())
+                         : torch::empty({}, logits.options());
   const int wpb = 4;
   hipLaunchKernelGGL(softmax_xent_fwd_kernel, dim3(ceil_div(M, wpb)),
                      dim3(wpb * WAVE), 0, at::cuda::getCurrentCUDAStream(),
                      logits.data_ptr<float>(), onehot.data_ptr<float>(),
-                     probs.data_ptr<float>(), loss.data_ptr<float>(), M, C);
-  loss.div_((double)M);  // mean over rows
+                     probs.data_ptr<float>(),
+                     want_value ? loss.data_ptr<float>() : nullptr,
+                     1.0f / (float)M, M, C);
   return {loss, probs};
 }
 
 std::vector<torch::Tensor> softmax_xent_idx_fwd(torch::Tensor logits,
-                                                torch::Tensor labels) {
+                                                torch::Tensor labels,
+                                                bool want_value) {
   check_f32l(logits, "logits");
   TORCH_CHECK(labels.is_cuda() && labels.scalar_type() == torch::kInt64 &&
                   labels.is_contiguous(),
               "labels must be contiguous int64 on GPU");
   const int M = logits.size(0), C = logits.size(1);
   auto probs = torch::empty_like(logits);
-  auto loss = torch::zeros({}, logits.options());
+  auto loss = want_value ? torch::zeros({}, logits.options())
+                         : torch::empty({}, logits.options());
   const int wpb = 4;
   hipLaunchKernelGGL(softmax_xent_idx_fwd_kernel, dim3(ceil_div(M, wpb)),
                      dim3(wpb * WAVE), 0, at::cuda::getCurrentCUDAStream(),
                      logits.data_ptr<float>(), labels.data_ptr<long>(),
-                     probs.data_ptr<float>(), loss.data_ptr<float>(), M, C);
-  loss.div_((double)M);
+                     probs.data_ptr<float>(),
+                     want_value ? loss.data_ptr<float>() : nullptr,
+                     1.0f / (float)M, M, C);
   return {loss, probs};
 }
 
@@ -214,16 +222,23 @@ torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,
   return d;
 }
 
-torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target) {
+torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target,
+                      bool want_value) {
   check_f32l(pred, "pred");
   check_f32l(target, "target");
-  auto out = torch::zeros({}, pred.options());
   const long n = pred.numel();
+  if (!want_value) {
+    // The MSE forward's ONLY product is the scalar; when it is not
+    // wanted (captured hot path) there is nothing to launch at all —
+    // the backward recomputes from pred/target.
+    return torch::empty({}, pred.options());
+  }
+  auto out = torch::zeros({}, pred.options());
   const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
   hipLaunchKernelGGL(mse_fwd_kernel, dim3(blocks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(), pred.data_ptr<float>(),
-                     target.data_ptr<float>(), out.data_ptr<float>(), n);
-  out.div_((double)n);
+                     target.data_ptr<float>(), out.data_ptr<float>(),
+                     1.0f / (float)n, n);
   return out;
 }
 

@@ -84,11 +84,20 @@ class Worker:
             self.flat_params.copy_(flat)
 
     def _grad_step(self, xb: torch.Tensor, yb: Optional[torch.Tensor]):
+        from sparkflow_amd.ops.functional import grad_only
+
         self.module.zero_grad(set_to_none=True)
         out = self.module(xb)
         target = yb if yb is not None else xb
-        loss = self.module.compute_loss(out, target)
-        loss.backward()
+        # The loss VALUE is never read here (monitoring computes its own
+        # loss) — grad_only() lets the loss ops skip the scalar kernels,
+        # and a preallocated ones-root skips autograd's fill per step.
+        with grad_only():
+            loss = self.module.compute_loss(out, target)
+        if getattr(self, "_ones_root", None) is None \
+                or self._ones_root.shape != loss.shape:
+            self._ones_root = torch.ones_like(loss)
+        loss.backward(gradient=self._ones_root)
         torch.cat([p.grad.reshape(-1) for p in self._params],
                   out=self.flat_grads)
         return self.flat_grads
@@ -290,26 +299,38 @@ class Worker:
                 if sub else self.y
         idx_pool = None
         ctr = None
+        done = None
         if sub:
             idx_pool = torch.zeros(self._IDX_POOL_CAP, mbs,
                                    dtype=torch.int64, device=self.device)
             ctr = torch.zeros(1, dtype=torch.int64, device=self.device)
+            done = torch.zeros(1, dtype=torch.int32, device=self.device)
             pick0 = self._minibatch_indices(n, mbs)
             torch.index_select(self.x, 0, pick0, out=xb)
             if yb is not None:
                 torch.index_select(self.y, 0, pick0, out=yb)
 
         ps = self.transport.ps if full_step else None
+        from sparkflow_amd import ops as _ops
+
+        fused_gather = (sub and _ops.available()
+                        and self.device.type == "cuda"
+                        and (yb is None or yb.dtype == torch.float32))
 
         def body():
             if full_step:
                 self.flat_params.copy_(ps.weights)  # the weight pull
             if full_step and sub:
-                row = idx_pool.index_select(0, ctr).view(-1)
-                torch.index_select(self.x, 0, row, out=xb)
-                if yb is not None:
-                    torch.index_select(self.y, 0, row, out=yb)
-                ctr.add_(1).remainder_(self._IDX_POOL_CAP)
+                if fused_gather:
+                    _ops.hip.gather_batch(self.x, xb, idx_pool, ctr, done,
+                                          self.y if yb is not None else None,
+                                          yb)
+                else:
+                    row = idx_pool.index_select(0, ctr).view(-1)
+                    torch.index_select(self.x, 0, row, out=xb)
+                    if yb is not None:
+                        torch.index_select(self.y, 0, row, out=yb)
+                    ctr.add_(1).remainder_(self._IDX_POOL_CAP)
             self._grad_step(xb, yb)
             if full_step:
                 ps.optimizer.apply(ps.weights, self.flat_grads)
