@@ -20,6 +20,7 @@ void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
                   torch::Tensor ctr, torch::Tensor done,
                   c10::optional<torch::Tensor> y,
                   c10::optional<torch::Tensor> yb);
+void pack_segments(std::vector<torch::Tensor> srcs, torch::Tensor flat);
 
 // losses.hip
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
@@ -84,6 +85,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "prologue)", py::arg("x"), py::arg("xb"), py::arg("idx_pool"),
         py::arg("ctr"), py::arg("done"), py::arg("y") = py::none(),
         py::arg("yb") = py::none());
+  m.def("pack_segments", &pack_segments,
+        "pack <=16 grad tensors into the flat push buffer in one launch");
   m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax+xent fwd",
         py::arg("logits"), py::arg("onehot"), py::arg("want_value") = true);
   m.def("softmax_xent_bwd", &softmax_xent_bwd, "softmax+xent bwd");

@@ -299,9 +299,8 @@ torch::Tensor conv_gemm_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
   const int base = ceil_div(Nc, BN) * ceil_div(M, BM);
   int splitk = 1;
   const int max_by_k = std::max(1, K / 256);
-  const int target = std::max(32, 512 / sf_parallel_hint());
   const int cap = sf_splitk_env_max();
-  while (splitk < 64 && splitk < cap && base * splitk * 2 <= target &&
+  while (splitk < 64 && splitk < cap && base * splitk * 2 <= 512 &&
          splitk * 2 <= max_by_k) {
     splitk *= 2;
   }

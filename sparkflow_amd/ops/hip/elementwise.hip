@@ -61,25 +61,36 @@ __global__ void gather_batch_kernel(const float* __restrict__ x,
                                     long* __restrict__ ctr,
                                     int* __restrict__ done, int mbs, long dx,
                                     long dy_row, int cap) {
-  const long c = *ctr;
+  const long c = *ctr;  // every thread's gather depends on this read
   const long* row_idx = idx_pool + c * mbs;
-  // block b gathers x row b (grid.x == mbs)
-  const int r = blockIdx.x;
-  const long src = row_idx[r];
-  const float* xs = x + src * dx;
-  float* xd = xb + (long)r * dx;
-  for (long i = threadIdx.x; i < dx; i += blockDim.x) xd[i] = xs[i];
-  if (y != nullptr && threadIdx.x == 0) {
-    for (long j = 0; j < dy_row; ++j) yb[(long)r * dy_row + j] =
-        y[src * dy_row + j];
+  const bool vec4 = (dx & 3) == 0;
+  for (int r = blockIdx.x; r < mbs; r += gridDim.x) {
+    const long src = row_idx[r];
+    const float* xs = x + src * dx;
+    float* xd = xb + (long)r * dx;
+    if (vec4) {
+      const float4* s4 = reinterpret_cast<const float4*>(xs);
+      float4* d4 = reinterpret_cast<float4*>(xd);
+      for (long i = threadIdx.x; i < (dx >> 2); i += blockDim.x)
+        d4[i] = s4[i];
+    } else {
+      for (long i = threadIdx.x; i < dx; i += blockDim.x) xd[i] = xs[i];
+    }
+    if (y != nullptr && threadIdx.x == 0) {
+      for (long j = 0; j < dy_row; ++j)
+        yb[(long)r * dy_row + j] = y[src * dy_row + j];
+    }
   }
-  __syncthreads();
+  __syncthreads();  // all of this block's threads have consumed `c`
   if (threadIdx.x == 0) {
-    __threadfence();
+    // done-ticket: the last-finishing block advances the cursor (it can
+    // only fire after every block has read `c`) and self-resets the
+    // ticket, so the kernel replays identically inside a hipGraph.
+    // Plain stores suffice: nothing reads ctr/done again until the next
+    // launch, and the kernel boundary orders that.
     if (atomicAdd(done, 1) == gridDim.x - 1) {
       *ctr = (c + 1) % cap;
       *done = 0;
-      __threadfence();
     }
   }
 }
@@ -110,11 +121,71 @@ void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
     ybp = yb->data_ptr<float>();
     dy_row = yb->numel() / mbs;
   }
-  hipLaunchKernelGGL(gather_batch_kernel, dim3(mbs), dim3(256), 0,
+  const int blocks = std::min(mbs, 64);  // few fat blocks: cheap ticket
+  hipLaunchKernelGGL(gather_batch_kernel, dim3(blocks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
                      xb.data_ptr<float>(), yp, ybp,
                      idx_pool.data_ptr<long>(), ctr.data_ptr<long>(),
                      done.data_ptr<int>(), mbs, dx, dy_row, cap);
+}
+
+// Pack up to 16 parameter-grad tensors into the flat push buffer in ONE
+// launch.  The segment pointers/offsets travel BY VALUE as kernel
+// arguments (baked at hipGraph capture; grad pointers are stable across
+// replays), so no descriptor memory or h2d copy is needed — this replaces
+// the per-step torch.cat kernel.
+#define PACK_MAX_SEGS 16
+struct PackArgs {
+  const float* src[PACK_MAX_SEGS];
+  long off[PACK_MAX_SEGS];   // dst offset of segment s
+  long n[PACK_MAX_SEGS];     // numel of segment s
+  int first_block[PACK_MAX_SEGS + 1];  // block range owning segment s
+};
+
+__global__ void pack_segments_kernel(PackArgs a, int nseg,
+                                     float* __restrict__ dst) {
+  int s = 0;
+  while (s + 1 < nseg && (int)blockIdx.x >= a.first_block[s + 1]) ++s;
+  const int b0 = a.first_block[s], b1 = a.first_block[s + 1];
+  const float* src = a.src[s];
+  float* d = dst + a.off[s];
+  const long n = a.n[s];
+  const long tid = ((long)(blockIdx.x - b0)) * blockDim.x + threadIdx.x;
+  const long stride = (long)(b1 - b0) * blockDim.x;
+  const bool vec4 =
+      (n & 3) == 0 && (((size_t)src | (size_t)d) & 15) == 0;
+  if (vec4) {
+    const float4* s4 = reinterpret_cast<const float4*>(src);
+    float4* d4 = reinterpret_cast<float4*>(d);
+    for (long i = tid; i < (n >> 2); i += stride) d4[i] = s4[i];
+  } else {
+    for (long i = tid; i < n; i += stride) d[i] = src[i];
+  }
+}
+
+void pack_segments(std::vector<torch::Tensor> srcs, torch::Tensor flat) {
+  const int nseg = (int)srcs.size();
+  TORCH_CHECK(nseg >= 1 && nseg <= PACK_MAX_SEGS,
+              "pack_segments: 1..16 segments");
+  PackArgs a;
+  long off = 0;
+  int blocks = 0;
+  for (int s = 0; s < nseg; ++s) {
+    TORCH_CHECK(srcs[s].is_cuda() && srcs[s].is_contiguous() &&
+                    srcs[s].scalar_type() == torch::kFloat32,
+                "pack_segments: contiguous fp32 on GPU");
+    a.src[s] = srcs[s].data_ptr<float>();
+    a.off[s] = off;
+    a.n[s] = srcs[s].numel();
+    a.first_block[s] = blocks;
+    blocks += (int)std::min<long>(128, (a.n[s] + 1023) / 1024);
+    off += a.n[s];
+  }
+  a.first_block[nseg] = blocks;
+  TORCH_CHECK(off == flat.numel(), "pack_segments: size mismatch");
+  hipLaunchKernelGGL(pack_segments_kernel, dim3(blocks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), a, nseg,
+                     flat.data_ptr<float>());
 }
 
 static inline void check_f32e(const torch::Tensor& t, const char* name) {

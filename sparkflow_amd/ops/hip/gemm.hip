@@ -247,15 +247,16 @@ int sf_splitk_env_max() {
 
 // Pick a K split so the grid approaches the 256-CU mark (a launch needs
 // many more workgroups than the M/N tiling of a 300x256 layer provides).
-// With several concurrent worker streams the CUs are already shared, so
-// the target grid (and the split) shrinks proportionally.
+// Measured (profiles/w4_kernels.md): shrinking the split under multi-
+// stream contention makes each GEMM slower without removing the reduce
+// launch, so the target stays at 512 regardless of the hint; the env cap
+// remains for experiments.
 static int pick_splitk(int M, int N, int K) {
   const int base = ceil_div(M, BM) * ceil_div(N, BN);
   const int max_by_k = std::max(1, K / 64);  // keep >=64-deep slices
-  const int target = std::max(32, 512 / g_parallel_hint);
   const int cap = sf_splitk_env_max();
   int splitk = 1;
-  while (splitk < 8 && splitk < cap && base * splitk * 2 <= target &&
+  while (splitk < 8 && splitk < cap && base * splitk * 2 <= 512 &&
          splitk * 2 <= max_by_k) {
     splitk *= 2;
   }

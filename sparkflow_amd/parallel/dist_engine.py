@@ -170,12 +170,6 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
         shared = DistTransport(pairs, numel, device)
         transports = [shared] * max(1, workers_per_gpu)
 
-    if device.type == "cuda":
-        from sparkflow_amd import ops as _ops
-
-        if _ops.available():  # split-K shrinks as worker streams share CUs
-            _ops.hip.set_parallel_hint(max(1, workers_per_gpu))
-
     workers = []
     for i, tr in enumerate(transports):
         cfg_i = config

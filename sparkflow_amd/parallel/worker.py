@@ -98,8 +98,15 @@ class Worker:
                 or self._ones_root.shape != loss.shape:
             self._ones_root = torch.ones_like(loss)
         loss.backward(gradient=self._ones_root)
-        torch.cat([p.grad.reshape(-1) for p in self._params],
-                  out=self.flat_grads)
+        from sparkflow_amd import ops as _ops
+
+        if (self.device.type == "cuda" and _ops.available()
+                and len(self._params) <= 16):
+            _ops.hip.pack_segments([p.grad.reshape(-1)
+                                    for p in self._params], self.flat_grads)
+        else:
+            torch.cat([p.grad.reshape(-1) for p in self._params],
+                      out=self.flat_grads)
         return self.flat_grads
 
     def _push(self, grads: torch.Tensor):

@@ -256,3 +256,11 @@ def test_loss_grad_only_mode_gpu():
     ref = logits.detach().clone().requires_grad_()
     torch.nn.functional.cross_entropy(ref, labels).backward()
     torch.testing.assert_close(logits.grad, ref.grad, atol=1e-5, rtol=1e-4)
+
+
+def test_pack_segments_gpu():
+    torch.manual_seed(11)
+    segs = [torch.randn(s, device=DEV) for s in (200704, 256, 10, 3, 65536)]
+    flat = torch.empty(sum(s.numel() for s in segs), device=DEV)
+    ops.hip.pack_segments(segs, flat)
+    torch.testing.assert_close(flat, torch.cat(segs))
