@@ -143,8 +143,7 @@ template <bool SPLITK>
 __global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
     const float* __restrict__ x, const float* __restrict__ dy,
     float* __restrict__ dw, ConvShape s, int M, int Nc, int K,
-    int k_chunk, float* __restrict__ ws = nullptr,
-    int* __restrict__ tickets = nullptr) {
+    int k_chunk) {
   __shared__ float As[BM][BK + 1];
   __shared__ float Bs[BK][BN + 4];
 
@@ -222,7 +221,7 @@ __global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
         const int co = m0 + wm + mi * 16 + frow + i;
         if (co < M) {
           if (SPLITK) {
-            ws[(long)blockIdx.z * M * Nc + (long)co * Nc + kcol] =
+            dw[(long)blockIdx.z * M * Nc + (long)co * Nc + kcol] =
                 acc[mi][ni][i];
           } else {
             dw[(long)co * Nc + kcol] = acc[mi][ni][i];
@@ -231,31 +230,17 @@ __global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
       }
     }
   }
+}
 
-  if (SPLITK) {
-    // Fused split-K reduce (same last-tile-ticket scheme as gemm.hip).
-    __threadfence();
-    __shared__ int is_last;
-    if (threadIdx.x == 0) {
-      const int tile = blockIdx.y * gridDim.x + blockIdx.x;
-      is_last = (atomicAdd(&tickets[tile], 1) == (int)gridDim.z - 1);
-    }
-    __syncthreads();
-    if (is_last) {
-      __threadfence();
-      const int splitk = gridDim.z;
-      for (int e = threadIdx.x; e < BM * BN; e += blockDim.x) {
-        const int co = m0 + e / BN;
-        const int kcol = n0 + (e % BN);
-        if (co < M && kcol < Nc) {
-          const long base = (long)co * Nc + kcol;
-          float sum = 0.0f;
-          for (int z = 0; z < splitk; ++z)
-            sum += ws[(long)z * M * Nc + base];
-          dw[base] = sum;
-        }
-      }
-    }
+__global__ void conv_splitk_reduce_kernel(const float* __restrict__ W,
+                                          float* __restrict__ C, long n,
+                                          int splitk) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float sum = 0.0f;
+    for (int z = 0; z < splitk; ++z) sum += W[(long)z * n + i];
+    C[i] = sum;
   }
 }
 
@@ -321,17 +306,16 @@ torch::Tensor conv_gemm_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
   }
   if (splitk > 1) {
     const int k_chunk = ceil_div(K, splitk * BK) * BK;
+    auto ws = torch::empty({splitk, M, Nc}, x.options());
     dim3 grid(ceil_div(Nc, BN), ceil_div(M, BM), splitk);
-    const int tiles = grid.x * grid.y;
-    const long planes = (long)splitk * M * Nc;
-    auto ws_t = torch::empty({planes + tiles}, x.options());
-    float* ws = ws_t.data_ptr<float>();
-    int* tickets = reinterpret_cast<int*>(ws + planes);
-    HIP_OK(hipMemsetAsync(tickets, 0, sizeof(int) * tiles, stream));
     hipLaunchKernelGGL((conv_gemm_wgrad_kernel<true>), grid, dim3(256), 0,
                        stream, x.data_ptr<float>(), dy.data_ptr<float>(),
-                       dw.data_ptr<float>(), s, M, Nc, K, k_chunk, ws,
-                       tickets);
+                       ws.data_ptr<float>(), s, M, Nc, K, k_chunk);
+    const long n = (long)M * Nc;
+    const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
+    hipLaunchKernelGGL(conv_splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
+                       stream, ws.data_ptr<float>(), dw.data_ptr<float>(), n,
+                       splitk);
   } else {
     dim3 grid(ceil_div(Nc, BN), ceil_div(M, BM));
     hipLaunchKernelGGL((conv_gemm_wgrad_kernel<false>), grid, dim3(256), 0,
