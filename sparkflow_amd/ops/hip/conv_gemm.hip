@@ -62,40 +62,56 @@ __global__ __launch_bounds__(256) void conv_gemm_fwd_kernel(
 
   f32x4 acc[2][2] = {};
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    {  // stage A: 64 rows x 16 cols, 4 gathers per thread
-      const int row = t >> 2, c4 = (t & 3) * 4;
-      const int gm = m0 + row;
+  // Register-prefetched staging (next chunk's gathers overlap this
+  // chunk's MFMAs — same pattern as gemm.hip).
+  const int st_row = t >> 2, st_c4 = (t & 3) * 4;
+
+  auto load_a = [&](int k0) -> float4 {
+    float4 r;
+    float* rr = reinterpret_cast<float*>(&r);
+    const int gm = m0 + st_row;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int gk = k0 + st_c4 + i;
+      rr[i] = (gm < M && gk < K) ? im2col_get(x, s, gm, gk) : 0.0f;
+    }
+    return r;
+  };
+  auto load_b = [&](int k0) -> float4 {
+    float4 r;
+    const int gn = n0 + st_row;
+    if (gn < Nc && k0 + BK <= K) {
+      r = *reinterpret_cast<const float4*>(&wgt[(long)gn * K + k0 + st_c4]);
+    } else {
+      float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        const int gk = k0 + c4 + i;
-        As[row][c4 + i] =
-            (gm < M && gk < K) ? im2col_get(x, s, gm, gk) : 0.0f;
+        const int gk = k0 + st_c4 + i;
+        rr[i] = (gn < Nc && gk < K) ? wgt[(long)gn * K + gk] : 0.0f;
       }
     }
-    {  // stage B: W[n'][k] -> Bs[k][n']; W rows are contiguous in k
-      const int row = t >> 2, c4 = (t & 3) * 4;
-      const int gn = n0 + row;
-      if (gn < Nc && k0 + BK <= K) {
-        const float4 v = *reinterpret_cast<const float4*>(
-            &wgt[(long)gn * K + k0 + c4]);
-        Bs[c4 + 0][row] = v.x;
-        Bs[c4 + 1][row] = v.y;
-        Bs[c4 + 2][row] = v.z;
-        Bs[c4 + 3][row] = v.w;
-      } else {
+    return r;
+  };
+
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  float4 ra = load_a(0);
+  float4 rb = load_b(0);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    {
+      const float* pa = reinterpret_cast<const float*>(&ra);
+      const float* pb = reinterpret_cast<const float*>(&rb);
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          const int gk = k0 + c4 + i;
-          Bs[c4 + i][row] =
-              (gn < Nc && gk < K) ? wgt[(long)gn * K + gk] : 0.0f;
-        }
-      }
+      for (int i = 0; i < 4; ++i) As[st_row][st_c4 + i] = pa[i];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) Bs[st_c4 + i][st_row] = pb[i];
     }
     __syncthreads();
-
-    const int fr = lane & 15;
-    const int fk = lane >> 4;
+    if (k0 + BK < K) {
+      ra = load_a(k0 + BK);
+      rb = load_b(k0 + BK);
+    }
 #pragma unroll
     for (int kk = 0; kk < BK / 4; ++kk) {
       const float a0 = As[wm + fr][kk * 4 + fk];
@@ -164,36 +180,57 @@ __global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
     k_end = min(K, k_begin + k_chunk);
   }
 
-  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
-    {  // stage A: As[co][p] = dy[n][co][hw] with p=(n,hw)
-      const int row = t >> 2, c4 = (t & 3) * 4;  // row = co idx in tile
-      const int gm = m0 + row;
+  const int a_row = t >> 2, a_c4 = (t & 3) * 4;  // A: row = co idx in tile
+  const int b_kk = t >> 4, b_n4 = (t & 15) * 4;  // B: im2col gather
+
+  auto load_a = [&](int k0) -> float4 {  // As[co][p] = dy[n][co][hw]
+    float4 r;
+    float* rr = reinterpret_cast<float*>(&r);
+    const int gm = m0 + a_row;
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int gp = k0 + c4 + i;
-        float v = 0.0f;
-        if (gm < M && gp < K) {
-          const int hw = gp % HoWo;
-          const int n = gp / HoWo;
-          v = dy[((long)n * M + gm) * HoWo + hw];
-        }
-        As[row][c4 + i] = v;
+    for (int i = 0; i < 4; ++i) {
+      const int gp = k0 + a_c4 + i;
+      float v = 0.0f;
+      if (gm < M && gp < K) {
+        const int hw = gp % HoWo;
+        const int n = gp / HoWo;
+        v = dy[((long)n * M + gm) * HoWo + hw];
       }
+      rr[i] = v;
     }
-    {  // stage B: Bs[p][k'] = im2col(x)[p, k']
-      const int kk = t >> 4, n4 = (t & 15) * 4;
-      const int gp = k0 + kk;
+    return r;
+  };
+  auto load_b = [&](int k0) -> float4 {  // Bs[p][k'] = im2col(x)[p, k']
+    float4 r;
+    float* rr = reinterpret_cast<float*>(&r);
+    const int gp = k0 + b_kk;
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int gk = n0 + n4 + i;
-        Bs[kk][n4 + i] =
-            (gp < K && gk < Nc) ? im2col_get(x, s, gp, gk) : 0.0f;
-      }
+    for (int i = 0; i < 4; ++i) {
+      const int gk = n0 + b_n4 + i;
+      rr[i] = (gp < K && gk < Nc) ? im2col_get(x, s, gp, gk) : 0.0f;
+    }
+    return r;
+  };
+
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  float4 ra = load_a(k_begin);
+  float4 rb = load_b(k_begin);
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+    {
+      const float* pa = reinterpret_cast<const float*>(&ra);
+      const float* pb = reinterpret_cast<const float*>(&rb);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) As[a_row][a_c4 + i] = pa[i];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) Bs[b_kk][b_n4 + i] = pb[i];
     }
     __syncthreads();
-
-    const int fr = lane & 15;
-    const int fk = lane >> 4;
+    if (k0 + BK < k_end) {
+      ra = load_a(k0 + BK);
+      rb = load_b(k0 + BK);
+    }
 #pragma unroll
     for (int kk = 0; kk < BK / 4; ++kk) {
       const float a0 = As[wm + fr][kk * 4 + fk];

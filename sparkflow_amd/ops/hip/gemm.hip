@@ -54,98 +54,115 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
     k_end = min(K, k_begin + k_chunk);
   }
 
-  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
-    // ---- stage A tile: As[m][k] = A'(m0+m, k0+k) ----
+  // Per-thread load/store split so the NEXT chunk's global loads issue
+  // during the CURRENT chunk's MFMAs (register prefetch — the K loop is
+  // load-latency-bound at these sizes, cdna_hip_programming.md "overlap
+  // global with MFMA").  Each thread owns one float4 of A and one of B
+  // per chunk.
+  const int a_row = t >> 2, a_c4 = (t & 3) * 4;       // TA=0 path
+  const int a_kk = t >> 4, a_m4 = (t & 15) * 4;       // TA=1 path
+  const int b_kk = t >> 4, b_n4 = (t & 15) * 4;       // TB=0 path
+  const int b_row = t >> 2, b_c4 = (t & 3) * 4;       // TB=1 path
+
+  auto load_a = [&](int k0) -> float4 {
+    float4 r;
     if (!TA) {
-      const int row = t >> 2, c4 = (t & 3) * 4;  // 64 rows x 4 float4
-      const int gm = m0 + row;
+      const int gm = m0 + a_row;
       if (gm < M && k0 + BK <= K) {
-        const float4 v = *reinterpret_cast<const float4*>(
-            &A[(long)gm * lda + k0 + c4]);
-        As[row][c4 + 0] = v.x;
-        As[row][c4 + 1] = v.y;
-        As[row][c4 + 2] = v.z;
-        As[row][c4 + 3] = v.w;
+        r = *reinterpret_cast<const float4*>(&A[(long)gm * lda + k0 + a_c4]);
       } else {
+        float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          const int gk = k0 + c4 + i;
-          As[row][c4 + i] =
-              (gm < M && gk < K) ? A[(long)gm * lda + gk] : 0.0f;
+          const int gk = k0 + a_c4 + i;
+          rr[i] = (gm < M && gk < K) ? A[(long)gm * lda + gk] : 0.0f;
         }
       }
     } else {
-      const int kk = t >> 4, m4 = (t & 15) * 4;  // 16 k-rows x 16 float4
-      const int gk = k0 + kk;
-      // BIASROW: logical row M-1 is a virtual all-ones row (it computes
-      // the bias gradient inside the wgrad GEMM); the fast path must not
-      // read it from memory.
+      const int gk = k0 + a_kk;
+      // BIASROW: logical row M-1 is a virtual all-ones row (bias grad
+      // rides the wgrad GEMM); the fast path must not read it.
       const int m_phys = BIASROW ? M - 1 : M;
       if (gk < K && m0 + BM <= m_phys) {
-        const float4 v = *reinterpret_cast<const float4*>(
-            &A[(long)gk * lda + m0 + m4]);
-        As[m4 + 0][kk] = v.x;
-        As[m4 + 1][kk] = v.y;
-        As[m4 + 2][kk] = v.z;
-        As[m4 + 3][kk] = v.w;
+        r = *reinterpret_cast<const float4*>(&A[(long)gk * lda + m0 + a_m4]);
       } else {
+        float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          const int gm = m0 + m4 + i;
+          const int gm = m0 + a_m4 + i;
           float val = 0.0f;
           if (gk < K && gm < M) {
-            val = (BIASROW && gm == M - 1) ? 1.0f
-                                           : A[(long)gk * lda + gm];
+            val = (BIASROW && gm == M - 1) ? 1.0f : A[(long)gk * lda + gm];
           }
-          As[m4 + i][kk] = val;
+          rr[i] = val;
         }
       }
     }
+    return r;
+  };
 
-    // ---- stage B tile: Bs[k][n] = B'(k0+k, n0+n) ----
+  auto load_b = [&](int k0) -> float4 {
+    float4 r;
     if (!TB) {
-      const int kk = t >> 4, n4 = (t & 15) * 4;  // 16 k-rows x 16 float4
-      const int gk = k0 + kk;
+      const int gk = k0 + b_kk;
       if (gk < K && n0 + BN <= N) {
-        const float4 v = *reinterpret_cast<const float4*>(
-            &B[(long)gk * ldb + n0 + n4]);
-        Bs[kk][n4 + 0] = v.x;
-        Bs[kk][n4 + 1] = v.y;
-        Bs[kk][n4 + 2] = v.z;
-        Bs[kk][n4 + 3] = v.w;
+        r = *reinterpret_cast<const float4*>(&B[(long)gk * ldb + n0 + b_n4]);
       } else {
+        float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          const int gn = n0 + n4 + i;
-          Bs[kk][n4 + i] =
-              (gk < K && gn < N) ? B[(long)gk * ldb + gn] : 0.0f;
+          const int gn = n0 + b_n4 + i;
+          rr[i] = (gk < K && gn < N) ? B[(long)gk * ldb + gn] : 0.0f;
         }
       }
     } else {
-      const int row = t >> 2, c4 = (t & 3) * 4;  // 64 n-rows x 4 float4
-      const int gn = n0 + row;
+      const int gn = n0 + b_row;
       if (gn < N && k0 + BK <= K) {
-        const float4 v = *reinterpret_cast<const float4*>(
-            &B[(long)gn * ldb + k0 + c4]);
-        Bs[c4 + 0][row] = v.x;
-        Bs[c4 + 1][row] = v.y;
-        Bs[c4 + 2][row] = v.z;
-        Bs[c4 + 3][row] = v.w;
+        r = *reinterpret_cast<const float4*>(&B[(long)gn * ldb + k0 + b_c4]);
       } else {
+        float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          const int gk = k0 + c4 + i;
-          Bs[c4 + i][row] =
-              (gn < N && gk < K) ? B[(long)gn * ldb + gk] : 0.0f;
+          const int gk = k0 + b_c4 + i;
+          rr[i] = (gn < N && gk < K) ? B[(long)gn * ldb + gk] : 0.0f;
         }
       }
     }
+    return r;
+  };
 
+  auto store_lds = [&](float4 ra, float4 rb) {
+    const float* pa = reinterpret_cast<const float*>(&ra);
+    const float* pb = reinterpret_cast<const float*>(&rb);
+    if (!TA) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) As[a_row][a_c4 + i] = pa[i];
+    } else {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) As[a_m4 + i][a_kk] = pa[i];
+    }
+    if (!TB) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) Bs[b_kk][b_n4 + i] = pb[i];
+    } else {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) Bs[b_c4 + i][b_row] = pb[i];
+    }
+  };
+
+  const int fr = lane & 15;   // fragment row/col position
+  const int fk = lane >> 4;   // k sub-index 0..3
+
+  float4 ra = load_a(k_begin);
+  float4 rb = load_b(k_begin);
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+    store_lds(ra, rb);
     __syncthreads();
-
+    if (k0 + BK < k_end) {  // prefetch next chunk over this chunk's MFMAs
+      ra = load_a(k0 + BK);
+      rb = load_b(k0 + BK);
+    }
     // ---- 16x16x4 MFMA over the 16-deep chunk ----
-    const int fr = lane & 15;   // fragment row/col position
-    const int fk = lane >> 4;   // k sub-index 0..3
 #pragma unroll
     for (int kk = 0; kk < BK / 4; ++kk) {
       const float a0 = As[wm + fr][kk * 4 + fk];
