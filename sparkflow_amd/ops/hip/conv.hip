@@ -120,15 +120,17 @@ __global__ void conv2d_wgrad_kernel(const float* __restrict__ x,
   }
 }
 
+// db[c] = sum over (n, hw) of dy[n][c][hw].  blockIdx.y chunks the batch
+// so small-C convs still fill the chip; contiguous hw sweeps per (n, c)
+// plane (no div/mod in the loop); partial sums combine via atomicAdd.
 __global__ void conv_bias_grad_kernel(const float* __restrict__ dy,
                                       float* __restrict__ db, int N, int C,
                                       int HW) {
   const int c = blockIdx.x;
   float s = 0.0f;
-  for (long i = threadIdx.x; i < (long)N * HW; i += blockDim.x) {
-    const int hw = i % HW;
-    const int n = i / HW;
-    s += dy[((long)n * C + c) * HW + hw];
+  for (int n = blockIdx.y; n < N; n += gridDim.y) {
+    const float* p = dy + ((long)n * C + c) * HW;
+    for (int i = threadIdx.x; i < HW; i += blockDim.x) s += p[i];
   }
   s = wave_sum(s);
   __shared__ float partials[4];
@@ -137,7 +139,11 @@ __global__ void conv_bias_grad_kernel(const float* __restrict__ dy,
   if (threadIdx.x == 0) {
     float b = 0.0f;
     for (int i = 0; i < (int)(blockDim.x / WAVE); ++i) b += partials[i];
-    db[c] = b;
+    if (gridDim.y == 1) {
+      db[c] = b;
+    } else {
+      atomicAdd(&db[c], b);
+    }
   }
 }
 
@@ -259,8 +265,10 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
 torch::Tensor conv_bias_grad(torch::Tensor dy) {
   check_f32c(dy, "dy");
   const int N = dy.size(0), C = dy.size(1), HW = dy.size(2) * dy.size(3);
-  auto db = torch::empty({C}, dy.options());
-  hipLaunchKernelGGL(conv_bias_grad_kernel, dim3(C), dim3(256), 0,
+  const int nchunks = std::min(N, std::max(1, 512 / C));
+  auto db = nchunks > 1 ? torch::zeros({C}, dy.options())
+                        : torch::empty({C}, dy.options());
+  hipLaunchKernelGGL(conv_bias_grad_kernel, dim3(C, nchunks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
                      db.data_ptr<float>(), N, C, HW);
   return db;

@@ -256,8 +256,8 @@ int sf_splitk_env_max() {
   static int cached = -1;
   if (cached < 0) {
     const char* e = getenv("SPARKFLOW_SPLITK_MAX");
-    cached = e ? atoi(e) : 8;
-    if (cached < 1) cached = 1;
+    cached = e ? atoi(e) : 1 << 20;  // unset: no env cap, structural
+    if (cached < 1) cached = 1;      // limits at the call sites govern
   }
   return cached;
 }
