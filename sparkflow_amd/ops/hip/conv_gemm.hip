@@ -335,15 +335,27 @@ __global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
   }
 }
 
+// One block per output element, threads parallel over the z planes (the
+// deep conv split — up to 256 slices — made a per-thread z loop a serial
+// latency chain).
 __global__ void conv_splitk_reduce_kernel(const float* __restrict__ W,
                                           float* __restrict__ C, long n,
                                           int splitk) {
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
+  for (long i = blockIdx.x; i < n; i += gridDim.x) {
     float sum = 0.0f;
-    for (int z = 0; z < splitk; ++z) sum += W[(long)z * n + i];
-    C[i] = sum;
+    for (int z = threadIdx.x; z < splitk; z += blockDim.x)
+      sum += W[(long)z * n + i];
+    sum = wave_sum(sum);
+    __shared__ float partials[4];
+    const int wid = threadIdx.x / WAVE;
+    if ((threadIdx.x & (WAVE - 1)) == 0) partials[wid] = sum;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float b = 0.0f;
+      for (int k = 0; k < (int)(blockDim.x / WAVE); ++k) b += partials[k];
+      C[i] = b;
+    }
+    __syncthreads();
   }
 }
 
@@ -418,7 +430,7 @@ torch::Tensor conv_gemm_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
                        stream, x.data_ptr<float>(), dy.data_ptr<float>(),
                        ws.data_ptr<float>(), s, M, Nc, K, k_chunk);
     const long n = (long)M * Nc;
-    const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
+    const int blocks = (int)std::min<long>(8192, n);
     hipLaunchKernelGGL(conv_splitk_reduce_kernel, dim3(blocks), dim3(256), 0,
                        stream, ws.data_ptr<float>(), dw.data_ptr<float>(), n,
                        splitk);
