@@ -62,11 +62,13 @@ def _warmup_pairs(pairs: DistPairs, device) -> None:
     c = torch.ones(1)
     if pairs.rank == 0:
         for w in range(1, pairs.world):
-            dist.send(t, dst=w, group=pairs.payload[w])
-            dist.send(c, dst=w, group=pairs.ctrl[w])
+            for i in range(pairs.workers_per_gpu):
+                dist.send(t, dst=w, group=pairs.payload[(w, i)])
+                dist.send(c, dst=w, group=pairs.ctrl[(w, i)])
     else:
-        dist.recv(t, src=0, group=pairs.payload[pairs.rank])
-        dist.recv(c, src=0, group=pairs.ctrl[pairs.rank])
+        for i in range(pairs.workers_per_gpu):
+            dist.recv(t, src=0, group=pairs.payload[(pairs.rank, i)])
+            dist.recv(c, src=0, group=pairs.ctrl[(pairs.rank, i)])
     if device.type == "cuda":
         torch.cuda.synchronize(device)
 
@@ -105,6 +107,9 @@ class EngineHandle:
         if self.device.type == "cuda":
             for w in self.workers:
                 w.ensure_graph(mini_batch)
+                if not w._graph_state["full_step"] \
+                        and hasattr(w.transport, "run_begin"):
+                    w.transport.run_begin(k)
             streams = getattr(self, "_streams", None)
             if streams is None or len(streams) != len(self.workers):
                 streams = [torch.cuda.Stream(device=self.device)
@@ -136,7 +141,8 @@ class EngineHandle:
 
     def close(self):
         if self.rank != 0:
-            self.worker.transport.close()
+            for w in self.workers:
+                w.transport.close()
         if self.service is not None:
             self.service.join()
 
@@ -155,7 +161,7 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
 
     ps = service = None
     if world > 1:
-        pairs = DistPairs()
+        pairs = DistPairs(workers_per_gpu=max(1, workers_per_gpu))
         _warmup_pairs(pairs, device)
     if rank == 0:
         ps = PSCore(graph_json, optimizer, device=device,
@@ -167,8 +173,8 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
                       for _ in range(max(1, workers_per_gpu))]
     else:
         numel = flat_size(module_from_json(graph_json))
-        shared = DistTransport(pairs, numel, device)
-        transports = [shared] * max(1, workers_per_gpu)
+        transports = [DistTransport(pairs, numel, device, widx=i)
+                      for i in range(max(1, workers_per_gpu))]
 
     workers = []
     for i, tr in enumerate(transports):

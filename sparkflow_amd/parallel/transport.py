@@ -30,7 +30,7 @@ import numpy as np
 import torch
 
 # control ops (DistTransport control channel + TCP op bytes)
-OP_PULL, OP_PUSH, OP_DONE, OP_PING = 1, 2, 3, 4
+OP_PULL, OP_PUSH, OP_DONE, OP_PING, OP_RUN = 1, 2, 3, 4, 5
 
 
 class WorkerTransport:
@@ -218,45 +218,79 @@ class DistPairs:
     Every rank must construct this collectively (``new_group`` is a
     collective call).  Payload groups use the default backend of the world
     (nccl on GPU = one RCCL communicator per worker<->PS xGMI link); control
-    groups are always gloo (small CPU int tensors carry PULL/PUSH/DONE)."""
+    groups are always gloo (small CPU int tensors carry PULL/PUSH/DONE).
 
-    def __init__(self, payload_backend: Optional[str] = None):
+    With ``workers_per_gpu`` > 1 every LOGICAL worker on a rank gets its
+    own pair communicator (keyed ``(rank, widx)``), so concurrent worker
+    streams never share an RCCL communicator."""
+
+    def __init__(self, payload_backend: Optional[str] = None,
+                 workers_per_gpu: int = 1):
         import torch.distributed as dist
 
         self.world = dist.get_world_size()
         self.rank = dist.get_rank()
+        self.workers_per_gpu = max(1, workers_per_gpu)
         self.payload = {}
         self.ctrl = {}
         for w in range(1, self.world):
-            self.payload[w] = dist.new_group([0, w], backend=payload_backend)
-            self.ctrl[w] = dist.new_group([0, w], backend="gloo")
+            for i in range(self.workers_per_gpu):
+                self.payload[(w, i)] = dist.new_group(
+                    [0, w], backend=payload_backend)
+                self.ctrl[(w, i)] = dist.new_group([0, w], backend="gloo")
 
 
 class DistTransport(WorkerTransport):
-    """Worker side of the RCCL star."""
+    """Worker side of the RCCL star.
 
-    def __init__(self, pairs: DistPairs, numel: int, device):
+    Two protocols share the pair link:
+
+    * per-op: every pull/push is announced on the gloo control channel
+      (the eager ``Worker.run`` path, where the op pattern varies);
+    * session (``run_begin(k)``): ONE control message announces k
+      steps, then the link runs k x {recv weights; send grads} with no
+      per-step control traffic or host syncs — RCCL enqueues are
+      stream-ordered, so the whole session pipelines on the worker's
+      HIP stream (this is what the driver's scaling bench exercises).
+    """
+
+    def __init__(self, pairs: DistPairs, numel: int, device, widx: int = 0):
         import torch.distributed as dist
 
         self.dist = dist
         self.rank = pairs.rank
-        self.pg = pairs.payload[self.rank]
-        self.cg = pairs.ctrl[self.rank]
+        self.pg = pairs.payload[(self.rank, widx)]
+        self.cg = pairs.ctrl[(self.rank, widx)]
         self.device = torch.device(device)
         self.buf = torch.empty(numel, dtype=torch.float32, device=self.device)
-        self._ctrl_buf = torch.zeros(1, dtype=torch.int64)
+        self._ctrl_buf = torch.zeros(2, dtype=torch.int64)
+        self._session = 0  # pulls+pushes prepaid by run_begin
 
-    def _ctrl(self, op: int):
-        self._ctrl_buf.fill_(op)
+    def _ctrl(self, op: int, arg: int = 0):
+        self._ctrl_buf[0] = op
+        self._ctrl_buf[1] = arg
         self.dist.send(self._ctrl_buf, dst=0, group=self.cg)
 
+    def run_begin(self, k: int) -> None:
+        """Announce a k-step session: the PS will k times {send weights,
+        recv grads, apply} on this pair with no further control."""
+        self._ctrl(OP_RUN, k)
+        self._session = k
+
     def pull(self) -> torch.Tensor:
-        self._ctrl(OP_PULL)
-        self.dist.recv(self.buf, src=0, group=self.pg)
+        self.pull_into(self.buf)
         return self.buf
 
+    def pull_into(self, out: torch.Tensor) -> None:
+        if self._session <= 0:
+            self._ctrl(OP_PULL)
+        self.dist.recv(out, src=0, group=self.pg)
+
     def push(self, flat_grads: torch.Tensor) -> None:
-        self._ctrl(OP_PUSH)
+        if self._session <= 0:
+            self._ctrl(OP_PUSH)
+        else:
+            self._session -= 1
         self.dist.send(flat_grads.contiguous(), dst=0, group=self.pg)
 
     def close(self) -> None:
@@ -280,38 +314,56 @@ class DistPSService:
 
     def start(self):
         for w in range(1, self.pairs.world):
-            t = threading.Thread(target=self._serve, args=(w,), daemon=True)
-            t.start()
-            self._threads.append(t)
+            for i in range(self.pairs.workers_per_gpu):
+                t = threading.Thread(target=self._serve, args=(w, i),
+                                     daemon=True)
+                t.start()
+                self._threads.append(t)
         return self
 
-    def _serve(self, w: int):
+    def _serve(self, w: int, widx: int = 0):
         dist = self.dist
-        pg = self.pairs.payload[w]
-        cg = self.pairs.ctrl[w]
+        pg = self.pairs.payload[(w, widx)]
+        cg = self.pairs.ctrl[(w, widx)]
         dev = self.ps.device
         use_cuda = dev.type == "cuda"
         stream = torch.cuda.Stream(device=dev) if use_cuda else None
         grad_buf = torch.empty(self.ps.numel, dtype=torch.float32, device=dev)
-        ctrl = torch.zeros(1, dtype=torch.int64)
+        ctrl = torch.zeros(2, dtype=torch.int64)
+
+        def handle(op: int):
+            if op == OP_PULL:
+                self.ps.send_weights(
+                    lambda buf: dist.send(buf, dst=w, group=pg))
+            elif op == OP_PUSH:
+                dist.recv(grad_buf, src=w, group=pg)
+                self.ps.apply_gradients(grad_buf)
+
         while True:
             dist.recv(ctrl, src=w, group=cg)
-            op = int(ctrl.item())
+            op = int(ctrl[0].item())
             if op == OP_DONE:
                 return
+            if op == OP_RUN:
+                # k-step session: {send weights, recv grads, apply} x k
+                # with no further control traffic — every op is a
+                # stream-ordered RCCL enqueue on this worker's stream.
+                k = int(ctrl[1].item())
+                if use_cuda:
+                    with torch.cuda.stream(stream):
+                        for _ in range(k):
+                            handle(OP_PULL)
+                            handle(OP_PUSH)
+                else:
+                    for _ in range(k):
+                        handle(OP_PULL)
+                        handle(OP_PUSH)
+                continue
             if use_cuda:
                 with torch.cuda.stream(stream):
-                    if op == OP_PULL:
-                        self.ps.send_weights(lambda buf: dist.send(buf, dst=w, group=pg))
-                    elif op == OP_PUSH:
-                        dist.recv(grad_buf, src=w, group=pg)
-                        self.ps.apply_gradients(grad_buf)
+                    handle(op)
             else:
-                if op == OP_PULL:
-                    self.ps.send_weights(lambda buf: dist.send(buf, dst=w, group=pg))
-                elif op == OP_PUSH:
-                    dist.recv(grad_buf, src=w, group=pg)
-                    self.ps.apply_gradients(grad_buf)
+                handle(op)
 
     def join(self):
         for t in self._threads:

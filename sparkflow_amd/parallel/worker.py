@@ -79,6 +79,10 @@ class Worker:
     # -- core step pieces ----------------------------------------------------
 
     def _pull_and_load(self):
+        if hasattr(self.transport, "pull_into"):
+            # remote transports recv straight into the param buffer
+            self.transport.pull_into(self.flat_params)
+            return
         flat = self.transport.pull()
         if flat.data_ptr() != self.flat_params.data_ptr():
             self.flat_params.copy_(flat)
@@ -259,7 +263,7 @@ class Worker:
             return
         n = self.x.shape[0]
         self._pull_and_load()
-        if g["xb"] is not self.x:
+        if g["xb"] is not self.x and not g["gather_inside"]:
             pick = self._minibatch_indices(n, g["xb"].shape[0])
             torch.index_select(self.x, 0, pick, out=g["xb"])
             if g["yb"] is not None:
@@ -289,6 +293,9 @@ class Worker:
         eager.
         """
         self.ensure_graph(mbs)
+        if not self._graph_state["full_step"] \
+                and hasattr(self.transport, "run_begin"):
+            self.transport.run_begin(k)  # zero per-step control traffic
         for _ in range(k):
             self.replay_once()
         self.finish_steps(k)
@@ -324,10 +331,16 @@ class Worker:
                         and self.device.type == "cuda"
                         and (yb is None or yb.dtype == torch.float32))
 
+        # the device-pool gather is captured in the graph whenever the
+        # fused kernel applies (full-step AND remote compute graphs);
+        # the non-fused torch-op sequence is only captured full-step
+        # (the remote fallback gathers eagerly in replay_once)
+        gather_inside = sub and (fused_gather or full_step)
+
         def body():
             if full_step:
                 self.flat_params.copy_(ps.weights)  # the weight pull
-            if full_step and sub:
+            if gather_inside:
                 if fused_gather:
                     _ops.hip.gather_batch(self.x, xb, idx_pool, ctr, done,
                                           self.y if yb is not None else None,
@@ -356,7 +369,8 @@ class Worker:
 
         self._graph_state = {"graph": graph, "xb": xb, "yb": yb,
                              "idx_pool": idx_pool, "ctr": ctr,
-                             "full_step": full_step}
+                             "full_step": full_step,
+                             "gather_inside": gather_inside}
         self._graph_key = key
 
 

@@ -91,3 +91,58 @@ def _dist_lock_worker(rank, world, port):
     if engine.rank == 0:
         assert engine.ps.update_count == world * 3
     torch.distributed.destroy_process_group()
+
+
+def _session_worker(rank, world, port, k):
+    """OP_RUN session protocol: k steps of {recv weights, send grads,
+    apply} with one control message (the multi-GPU bench hot path)."""
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["LOCAL_RANK"] = str(rank)
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import gaussian_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    graph_json = build_graph(gaussian_mlp)
+    cfg = TrainConfig(iters=1, mini_batch=16, mini_stochastic_iters=1,
+                      seed=4)
+    engine = build_engine(graph_json, OptimizerSpec("gradient_descent", 0.01),
+                          cfg, device="cpu", workers_per_gpu=2)
+    rng = np.random.default_rng(rank)
+    x = rng.normal(0, 1, (64, 10)).astype(np.float32)
+    y = (rng.random((64, 1)) > 0.5).astype(np.float32)
+    engine.stage_all(x, y)
+    engine.barrier()
+    if engine.rank == 0:
+        engine.run_steps(k, 16)
+        # remote ranks' sessions land on the PS service threads; wait for
+        # all updates: 2 local workers x k + (world-1) x 2 workers x k
+        import time
+
+        expected = 2 * k * world
+        deadline = time.time() + 30
+        while engine.ps.update_count < expected and time.time() < deadline:
+            time.sleep(0.05)
+        assert engine.ps.update_count == expected, engine.ps.update_count
+        assert torch.isfinite(engine.ps.weights).all()
+    else:
+        for w in engine.workers:
+            w.transport.run_begin(k)
+            for _ in range(k):
+                w._pull_and_load()
+                g = w._grad_step(w.x[:16], w.y[:16])
+                w.transport.push(g)
+    engine.barrier()
+    engine.close()
+    engine.barrier()
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+def test_dist_session_protocol_gloo():
+    port = free_port()
+    mp.spawn(_session_worker, args=(2, port, 3), nprocs=2, join=True)
