@@ -70,6 +70,20 @@ std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
 torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
                             int64_t W);
 
+// dist_session.hip — native RCCL session loops (GIL-free)
+py::bytes nccl_unique_id();
+int64_t nccl_comm_init(py::bytes id_bytes, int64_t nranks, int64_t rank);
+void nccl_comm_destroy(int64_t h);
+void ps_session(int64_t comm_h, int64_t peer, torch::Tensor weights,
+                torch::Tensor grad_buf, int64_t k, int64_t opt_id,
+                std::vector<torch::Tensor> state, double lr, double h1,
+                double h2, double h3);
+void worker_session(int64_t comm_h, int64_t peer, torch::Tensor w_flat,
+                    torch::Tensor g_flat, int64_t graph_exec, int64_t k);
+void raw_send(int64_t comm_h, int64_t peer, torch::Tensor t);
+void raw_recv(int64_t comm_h, int64_t peer, torch::Tensor t);
+void multi_replay(std::vector<int64_t> execs, int64_t k);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sparkflow_amd gfx950 (CDNA4) kernels";
   m.def("gemm_act", &gemm_act, "fused fp32 MFMA GEMM + bias + activation");
@@ -118,4 +132,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_bias_grad", &conv_bias_grad, "conv bias grad");
   m.def("maxpool2d_fwd", &maxpool2d_fwd, "maxpool fwd with argmax indices");
   m.def("maxpool2d_bwd", &maxpool2d_bwd, "maxpool bwd scatter");
+  m.def("nccl_unique_id", &nccl_unique_id, "RCCL unique id for a pair comm");
+  m.def("nccl_comm_init", &nccl_comm_init,
+        "init this rank's side of a raw RCCL communicator");
+  m.def("nccl_comm_destroy", &nccl_comm_destroy, "destroy a raw comm");
+  m.def("ps_session", &ps_session,
+        "GIL-free k-step PS loop: send weights / recv grads / fused apply");
+  m.def("worker_session", &worker_session,
+        "GIL-free k-step worker loop: recv weights / graph launch / send");
+  m.def("raw_send", &raw_send, "one fp32 send on a raw comm");
+  m.def("raw_recv", &raw_recv, "one fp32 recv on a raw comm");
+  m.def("multi_replay", &multi_replay,
+        "k round-robin launches of W captured step graphs on W streams");
 }

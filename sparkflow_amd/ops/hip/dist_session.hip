@@ -1,0 +1,201 @@
+// Native (C++/HIP) session loops for the async-PS RCCL star.
+//
+// The Python-thread-per-worker PS service is correct but GIL-bound: an
+// 8-GPU node with 4 workers/GPU makes ~2 torch.distributed calls per
+// worker-step on rank 0 — tens of thousands of GIL-serialized Python
+// calls per bench.  These entry points run a whole k-step session in
+// C++ with the GIL RELEASED:
+//
+//   ps_session:     k x { ncclSend(weights) ; ncclRecv(grads) ; fused
+//                   optimizer apply } on this service thread's stream —
+//                   concurrent sessions race their applies on the
+//                   weight buffer (Hogwild, reference
+//                   HogwildSparkModel.py:219-242 semantics).
+//   worker_session: k x { ncclRecv(weights) ; hipGraphLaunch(step) ;
+//                   ncclSend(grads) } — the captured compute graph
+//                   comes from torch.cuda.CUDAGraph.raw_cuda_graph_exec.
+//   multi_replay:   k round-robin launches of W captured full-step
+//                   graphs on W streams (the single-GPU multi-worker
+//                   Hogwild loop) in one call.
+//
+// Pair communicators are bootstrapped with ncclGetUniqueId /
+// ncclCommInitRank (the unique id travels over the existing gloo
+// control channel), independent of torch's process-group internals.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <c10/cuda/CUDAGuard.h>
+#include <rccl/rccl.h>
+
+#include <mutex>
+#include <vector>
+
+#include "common.h"
+
+// fused optimizer entry points (optim.hip)
+void fused_sgd(torch::Tensor p, torch::Tensor g, double lr);
+void fused_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor acc,
+                    double lr, double mom, bool nesterov);
+void fused_adam_dev(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                    torch::Tensor v, torch::Tensor state, double lr,
+                    double b1, double b2, double eps);
+void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
+                   torch::Tensor mom, double lr, double decay,
+                   double momentum, double eps);
+
+#define NCCL_OK(expr)                                                       \
+  do {                                                                      \
+    ncclResult_t _r = (expr);                                               \
+    TORCH_CHECK(_r == ncclSuccess, "RCCL error: ",                          \
+                ncclGetErrorString(_r), " at ", __FILE__, ":", __LINE__);   \
+  } while (0)
+
+static std::vector<ncclComm_t> g_comms;
+static std::mutex g_comms_mu;
+
+py::bytes nccl_unique_id() {
+  ncclUniqueId id;
+  NCCL_OK(ncclGetUniqueId(&id));
+  return py::bytes(reinterpret_cast<const char*>(&id), sizeof(id));
+}
+
+int64_t nccl_comm_init(py::bytes id_bytes, int64_t nranks, int64_t rank) {
+  std::string s = id_bytes;
+  TORCH_CHECK(s.size() == sizeof(ncclUniqueId), "bad unique id size");
+  ncclUniqueId id;
+  memcpy(&id, s.data(), sizeof(id));
+  ncclComm_t comm = nullptr;
+  {
+    // init rendezvouses over TCP with the peer — release the GIL
+    py::gil_scoped_release rel;
+    NCCL_OK(ncclCommInitRank(&comm, (int)nranks, id, (int)rank));
+  }
+  std::lock_guard<std::mutex> lk(g_comms_mu);
+  g_comms.push_back(comm);
+  return (int64_t)g_comms.size() - 1;
+}
+
+static ncclComm_t get_comm(int64_t h) {
+  std::lock_guard<std::mutex> lk(g_comms_mu);
+  TORCH_CHECK(h >= 0 && h < (int64_t)g_comms.size() && g_comms[h],
+              "bad comm handle");
+  return g_comms[h];
+}
+
+void nccl_comm_destroy(int64_t h) {
+  ncclComm_t c = get_comm(h);
+  {
+    py::gil_scoped_release rel;
+    ncclCommDestroy(c);
+  }
+  std::lock_guard<std::mutex> lk(g_comms_mu);
+  g_comms[h] = nullptr;
+}
+
+// opt ids shared with parallel/optimizers.py native dispatch
+enum { OPT_SGD = 0, OPT_ADAM_DEV = 1, OPT_MOMENTUM = 2, OPT_RMSPROP = 3 };
+
+void ps_session(int64_t comm_h, int64_t peer, torch::Tensor weights,
+                torch::Tensor grad_buf, int64_t k, int64_t opt_id,
+                std::vector<torch::Tensor> state, double lr, double h1,
+                double h2, double h3) {
+  ncclComm_t comm = get_comm(comm_h);
+  TORCH_CHECK(weights.is_cuda() && grad_buf.is_cuda(), "GPU buffers only");
+  const size_t n = weights.numel();
+  TORCH_CHECK((size_t)grad_buf.numel() == n, "size mismatch");
+  at::cuda::CUDAStream stream =
+      at::cuda::getStreamFromPool(false, weights.get_device());
+  at::cuda::CUDAStreamGuard guard(stream);
+  float* wp = weights.data_ptr<float>();
+  float* gp = grad_buf.data_ptr<float>();
+
+  py::gil_scoped_release rel;
+  for (int64_t i = 0; i < k; ++i) {
+    NCCL_OK(ncclSend(wp, n, ncclFloat32, (int)peer, comm, stream.stream()));
+    NCCL_OK(ncclRecv(gp, n, ncclFloat32, (int)peer, comm, stream.stream()));
+    switch (opt_id) {
+      case OPT_SGD:
+        fused_sgd(weights, grad_buf, lr);
+        break;
+      case OPT_ADAM_DEV:
+        fused_adam_dev(weights, grad_buf, state[0], state[1], state[2], lr,
+                       h1, h2, h3);
+        break;
+      case OPT_MOMENTUM:
+        fused_momentum(weights, grad_buf, state[0], lr, h1, h2 != 0.0);
+        break;
+      case OPT_RMSPROP:
+        fused_rmsprop(weights, grad_buf, state[0], state[1], lr, h1, h2,
+                      h3);
+        break;
+      default:
+        TORCH_CHECK(false, "unsupported native optimizer id");
+    }
+  }
+  // the service thread owns this stream; make session completion
+  // observable to the Python side that joins on update counters
+  HIP_OK(hipStreamSynchronize(stream.stream()));
+}
+
+void worker_session(int64_t comm_h, int64_t peer, torch::Tensor w_flat,
+                    torch::Tensor g_flat, int64_t graph_exec, int64_t k) {
+  ncclComm_t comm = get_comm(comm_h);
+  const size_t n = w_flat.numel();
+  hipGraphExec_t exec = reinterpret_cast<hipGraphExec_t>(graph_exec);
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  float* wp = w_flat.data_ptr<float>();
+  float* gp = g_flat.data_ptr<float>();
+
+  py::gil_scoped_release rel;
+  for (int64_t i = 0; i < k; ++i) {
+    NCCL_OK(ncclRecv(wp, n, ncclFloat32, (int)peer, comm, stream));
+    HIP_OK(hipGraphLaunch(exec, stream));
+    NCCL_OK(ncclSend(gp, (size_t)g_flat.numel(), ncclFloat32, (int)peer,
+                     comm, stream));
+  }
+}
+
+// Per-op raw-comm helpers: the Python PS loop uses these when a session
+// arrives on the raw channel but the optimizer has no native kernel.
+void raw_send(int64_t comm_h, int64_t peer, torch::Tensor t) {
+  ncclComm_t comm = get_comm(comm_h);
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  py::gil_scoped_release rel;
+  NCCL_OK(ncclSend(t.data_ptr<float>(), t.numel(), ncclFloat32, (int)peer,
+                   comm, stream));
+}
+
+void raw_recv(int64_t comm_h, int64_t peer, torch::Tensor t) {
+  ncclComm_t comm = get_comm(comm_h);
+  hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
+  py::gil_scoped_release rel;
+  NCCL_OK(ncclRecv(t.data_ptr<float>(), t.numel(), ncclFloat32, (int)peer,
+                   comm, stream));
+}
+
+void multi_replay(std::vector<int64_t> execs, int64_t k) {
+  // W full-step graphs round-robin onto W pool streams: the whole
+  // multi-worker Hogwild loop is enqueued from one GIL-free C++ call.
+  const int W = (int)execs.size();
+  std::vector<at::cuda::CUDAStream> streams;
+  streams.reserve(W);
+  for (int w = 0; w < W; ++w) streams.push_back(at::cuda::getStreamFromPool());
+  at::cuda::CUDAStream cur = at::cuda::getCurrentCUDAStream();
+  py::gil_scoped_release rel;
+  // order the pool streams after whatever is pending on the current one
+  hipEvent_t ev;
+  HIP_OK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  HIP_OK(hipEventRecord(ev, cur.stream()));
+  for (int w = 0; w < W; ++w)
+    HIP_OK(hipStreamWaitEvent(streams[w].stream(), ev, 0));
+  for (int64_t i = 0; i < k; ++i) {
+    for (int w = 0; w < W; ++w) {
+      HIP_OK(hipGraphLaunch(reinterpret_cast<hipGraphExec_t>(execs[w]),
+                            streams[w].stream()));
+    }
+  }
+  for (int w = 0; w < W; ++w) {
+    HIP_OK(hipEventRecord(ev, streams[w].stream()));
+    HIP_OK(hipStreamWaitEvent(cur.stream(), ev, 0));
+  }
+  HIP_OK(hipEventDestroy(ev));
+}

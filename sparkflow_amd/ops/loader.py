@@ -22,6 +22,7 @@ SOURCES = [
     HERE / "hip" / "optim.hip",
     HERE / "hip" / "conv.hip",
     HERE / "hip" / "conv_gemm.hip",
+    HERE / "hip" / "dist_session.hip",
 ]
 
 _ext = None
@@ -80,6 +81,7 @@ def build(verbose: bool = True):
         build_directory=str(BUILD_DIR),
         extra_cflags=["-O3"],
         extra_cuda_cflags=["-O3", "--offload-arch=gfx950"],
+        extra_ldflags=["-L/opt/rocm/lib", "-lrccl"],
         verbose=verbose,
         is_python_module=True,
     )

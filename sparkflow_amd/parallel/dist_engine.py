@@ -105,11 +105,28 @@ class EngineHandle:
         if len(self.workers) == 1:
             return self.worker.run_steps(k, mini_batch)
         if self.device.type == "cuda":
+            from sparkflow_amd import ops as _ops
+
             for w in self.workers:
                 w.ensure_graph(mini_batch)
-                if not w._graph_state["full_step"] \
-                        and hasattr(w.transport, "run_begin"):
-                    w.transport.run_begin(k)
+            # All-local full-step graphs + extension: the whole W x k
+            # Hogwild loop is one GIL-free C++ call (round-robin graph
+            # launches onto W pool streams).
+            if _ops.available() and all(
+                    w._graph_state["full_step"] for w in self.workers):
+                execs = [w._graph_state["graph"].raw_cuda_graph_exec()
+                         for w in self.workers]
+                _ops.hip.multi_replay(execs, k)
+                total = 0
+                for w in self.workers:
+                    w.finish_steps(k)
+                    total += k * min(mini_batch, w.x.shape[0])
+                return total
+            # Remote / lock-mode workers: each worker's whole k-step run
+            # is enqueued under its own stream (run_steps uses the
+            # GIL-free C++ session when the raw comm exists, else the
+            # per-step replay loop — either way enqueues are async, so
+            # the W workers overlap on the device).
             streams = getattr(self, "_streams", None)
             if streams is None or len(streams) != len(self.workers):
                 streams = [torch.cuda.Stream(device=self.device)
@@ -118,16 +135,12 @@ class EngineHandle:
             cur = torch.cuda.current_stream(self.device)
             for s in streams:
                 s.wait_stream(cur)
-            for _ in range(k):
-                for w, s in zip(self.workers, streams):
-                    with torch.cuda.stream(s):
-                        w.replay_once()
+            total = 0
+            for w, s in zip(self.workers, streams):
+                with torch.cuda.stream(s):
+                    total += w.run_steps(k, mini_batch)
             for s in streams:
                 cur.wait_stream(s)
-            total = 0
-            for w in self.workers:
-                w.finish_steps(k)
-                total += k * min(mini_batch, w.x.shape[0])
             return total
         total = 0
         for _ in range(k):  # CPU: interleave eager single steps

@@ -63,6 +63,12 @@ class FlatOptimizer:
     def set_step_count(self, t: int) -> None:
         self.t = int(t)
 
+    def native_spec(self):
+        """(opt_id, state tensors, lr, h1, h2, h3) for the C++ GIL-free
+        PS session loop (ops/hip/dist_session.hip), or None when this
+        optimizer has no fused native apply."""
+        return None
+
 
 class GradientDescent(FlatOptimizer):
     name = "gradient_descent"
@@ -77,6 +83,9 @@ class GradientDescent(FlatOptimizer):
             ops.hip.fused_sgd(param, grad, self.lr)
         else:
             param.add_(grad, alpha=-self.lr)
+
+    def native_spec(self):
+        return (0, [], self.lr, 0.0, 0.0, 0.0)
 
 
 class Momentum(FlatOptimizer):
@@ -105,6 +114,10 @@ class Momentum(FlatOptimizer):
 
     def state_tensors(self):
         return {"acc": self.acc}
+
+    def native_spec(self):
+        return (2, [self.acc], self.lr, self.momentum,
+                1.0 if self.nesterov else 0.0, 0.0)
 
 
 class Adam(FlatOptimizer):
@@ -140,6 +153,12 @@ class Adam(FlatOptimizer):
 
     def state_tensors(self):
         return {"m": self.m, "v": self.v}
+
+    def native_spec(self):
+        if self.dev_state is None:
+            return None
+        return (1, [self.m, self.v, self.dev_state], self.lr, self.beta1,
+                self.beta2, self.eps)
 
     def step_count(self) -> int:
         if self.dev_state is not None:
@@ -186,6 +205,12 @@ class RMSProp(FlatOptimizer):
         if self.mg is not None:
             s["mg"] = self.mg
         return s
+
+    def native_spec(self):
+        if self.centered:
+            return None
+        return (3, [self.ms, self.mom], self.lr, self.decay, self.momentum,
+                self.eps)
 
 
 class Adadelta(FlatOptimizer):

@@ -20,6 +20,7 @@ transports behind one interface:
 """
 from __future__ import annotations
 
+import os
 import socket
 import struct
 import threading
@@ -233,11 +234,58 @@ class DistPairs:
         self.workers_per_gpu = max(1, workers_per_gpu)
         self.payload = {}
         self.ctrl = {}
+        self.native = {}  # (w, i) -> raw RCCL comm handle (both sides ok)
         for w in range(1, self.world):
             for i in range(self.workers_per_gpu):
                 self.payload[(w, i)] = dist.new_group(
                     [0, w], backend=payload_backend)
                 self.ctrl[(w, i)] = dist.new_group([0, w], backend="gloo")
+
+    def init_native(self, device) -> None:
+        """Bootstrap one raw RCCL communicator per pair for the GIL-free
+        C++ session loops (ops/hip/dist_session.hip).  The ncclUniqueId
+        travels over the gloo control channel; a success flag is
+        exchanged afterwards so both sides agree whether the pair may use
+        the raw channel (a one-sided failure would otherwise deadlock
+        the wire).  No-op (empty ``native``) on CPU or without the
+        extension."""
+        import torch.distributed as dist
+
+        if torch.device(device).type != "cuda":
+            return
+        from sparkflow_amd import ops
+
+        have_ext = ops.available()
+        if os.environ.get("SPARKFLOW_NATIVE_SESSION", "1") == "0":
+            have_ext = False
+        my_can = torch.tensor([1 if have_ext else 0], dtype=torch.int64)
+        if self.rank == 0:
+            for w in range(1, self.world):
+                for i in range(self.workers_per_gpu):
+                    cg = self.ctrl[(w, i)]
+                    peer_can = torch.zeros(1, dtype=torch.int64)
+                    dist.send(my_can, dst=w, group=cg)
+                    dist.recv(peer_can, src=w, group=cg)
+                    if not (have_ext and int(peer_can.item())):
+                        continue
+                    raw = ops.hip.nccl_unique_id()
+                    idbuf = torch.frombuffer(bytearray(raw),
+                                             dtype=torch.uint8).clone()
+                    dist.send(idbuf, dst=w, group=cg)
+                    self.native[(w, i)] = ops.hip.nccl_comm_init(
+                        bytes(idbuf.numpy().tobytes()), 2, 0)
+        else:
+            for i in range(self.workers_per_gpu):
+                cg = self.ctrl[(self.rank, i)]
+                peer_can = torch.zeros(1, dtype=torch.int64)
+                dist.recv(peer_can, src=0, group=cg)
+                dist.send(my_can, dst=0, group=cg)
+                if not (have_ext and int(peer_can.item())):
+                    continue
+                idbuf = torch.zeros(128, dtype=torch.uint8)
+                dist.recv(idbuf, src=0, group=cg)
+                self.native[(self.rank, i)] = ops.hip.nccl_comm_init(
+                    bytes(idbuf.numpy().tobytes()), 2, 1)
 
 
 class DistTransport(WorkerTransport):
@@ -261,21 +309,30 @@ class DistTransport(WorkerTransport):
         self.rank = pairs.rank
         self.pg = pairs.payload[(self.rank, widx)]
         self.cg = pairs.ctrl[(self.rank, widx)]
+        self.native_comm = pairs.native.get((self.rank, widx))
         self.device = torch.device(device)
         self.buf = torch.empty(numel, dtype=torch.float32, device=self.device)
-        self._ctrl_buf = torch.zeros(2, dtype=torch.int64)
+        self._ctrl_buf = torch.zeros(3, dtype=torch.int64)
         self._session = 0  # pulls+pushes prepaid by run_begin
+        self._session_native = False
 
-    def _ctrl(self, op: int, arg: int = 0):
+    def _ctrl(self, op: int, arg: int = 0, chan: int = 0):
         self._ctrl_buf[0] = op
         self._ctrl_buf[1] = arg
+        self._ctrl_buf[2] = chan
         self.dist.send(self._ctrl_buf, dst=0, group=self.cg)
 
-    def run_begin(self, k: int) -> None:
+    def run_begin(self, k: int, native: Optional[bool] = None) -> None:
         """Announce a k-step session: the PS will k times {send weights,
-        recv grads, apply} on this pair with no further control."""
-        self._ctrl(OP_RUN, k)
+        recv grads, apply} on this pair with no further control.  With
+        ``native`` (default: whenever the raw comm exists) the payload
+        moves on the raw RCCL channel so BOTH sides can run their loop
+        in C++ with the GIL released."""
+        if native is None:
+            native = self.native_comm is not None
+        self._ctrl(OP_RUN, k, 1 if native else 0)
         self._session = k
+        self._session_native = bool(native)
 
     def pull(self) -> torch.Tensor:
         self.pull_into(self.buf)
@@ -284,14 +341,43 @@ class DistTransport(WorkerTransport):
     def pull_into(self, out: torch.Tensor) -> None:
         if self._session <= 0:
             self._ctrl(OP_PULL)
-        self.dist.recv(out, src=0, group=self.pg)
+            self.dist.recv(out, src=0, group=self.pg)
+            return
+        if self._session_native:
+            from sparkflow_amd import ops
+
+            ops.hip.raw_recv(self.native_comm, 0, out)
+        else:
+            self.dist.recv(out, src=0, group=self.pg)
 
     def push(self, flat_grads: torch.Tensor) -> None:
         if self._session <= 0:
             self._ctrl(OP_PUSH)
+            self.dist.send(flat_grads.contiguous(), dst=0, group=self.pg)
+            return
+        self._session -= 1
+        if self._session_native:
+            from sparkflow_amd import ops
+
+            ops.hip.raw_send(self.native_comm, 0, flat_grads.contiguous())
         else:
-            self._session -= 1
-        self.dist.send(flat_grads.contiguous(), dst=0, group=self.pg)
+            self.dist.send(flat_grads.contiguous(), dst=0, group=self.pg)
+
+    def run_session(self, k: int, flat_params: torch.Tensor,
+                    flat_grads: torch.Tensor, graph_exec: int) -> bool:
+        """Whole k-step session in ONE C++ call (GIL released): k x
+        {recv weights into flat_params, launch the captured step graph,
+        send flat_grads}.  Returns False when the raw channel is
+        unavailable (caller falls back to the per-step loop)."""
+        if self.native_comm is None:
+            return False
+        from sparkflow_amd import ops
+
+        self.run_begin(k, native=True)
+        ops.hip.worker_session(self.native_comm, 0, flat_params, flat_grads,
+                               graph_exec, k)
+        self._session = 0
+        return True
 
     def close(self) -> None:
         self._ctrl(OP_DONE)
@@ -325,13 +411,24 @@ class DistPSService:
         dist = self.dist
         pg = self.pairs.payload[(w, widx)]
         cg = self.pairs.ctrl[(w, widx)]
+        native_comm = self.pairs.native.get((w, widx))
         dev = self.ps.device
         use_cuda = dev.type == "cuda"
         stream = torch.cuda.Stream(device=dev) if use_cuda else None
         grad_buf = torch.empty(self.ps.numel, dtype=torch.float32, device=dev)
-        ctrl = torch.zeros(2, dtype=torch.int64)
+        ctrl = torch.zeros(3, dtype=torch.int64)
 
-        def handle(op: int):
+        def handle(op: int, native: bool = False):
+            if native:
+                from sparkflow_amd import ops
+
+                if op == OP_PULL:
+                    self.ps.send_weights(
+                        lambda buf: ops.hip.raw_send(native_comm, 1, buf))
+                elif op == OP_PUSH:
+                    ops.hip.raw_recv(native_comm, 1, grad_buf)
+                    self.ps.apply_gradients(grad_buf)
+                return
             if op == OP_PULL:
                 self.ps.send_weights(
                     lambda buf: dist.send(buf, dst=w, group=pg))
@@ -339,25 +436,40 @@ class DistPSService:
                 dist.recv(grad_buf, src=w, group=pg)
                 self.ps.apply_gradients(grad_buf)
 
+        def run_session(k: int, native: bool):
+            # k-step session: {send weights, recv grads, apply} x k with
+            # no further control traffic.  On the raw channel with a
+            # native-fused optimizer and no lock, the WHOLE loop runs in
+            # C++ with the GIL released (ops/hip/dist_session.hip);
+            # otherwise per-step ops from Python, still session-paced.
+            spec = self.ps.optimizer.native_spec() if use_cuda else None
+            if native and spec is not None and not self.ps.acquire_lock:
+                from sparkflow_amd import ops
+
+                opt_id, state, lr, h1, h2, h3 = spec
+                ops.hip.ps_session(native_comm, 1, self.ps.weights,
+                                   grad_buf, k, opt_id, state, lr, h1, h2,
+                                   h3)
+                self.ps.pull_count += k
+                self.ps.update_count += k
+                return
+            for _ in range(k):
+                handle(OP_PULL, native)
+                handle(OP_PUSH, native)
+
         while True:
             dist.recv(ctrl, src=w, group=cg)
             op = int(ctrl[0].item())
             if op == OP_DONE:
                 return
             if op == OP_RUN:
-                # k-step session: {send weights, recv grads, apply} x k
-                # with no further control traffic — every op is a
-                # stream-ordered RCCL enqueue on this worker's stream.
                 k = int(ctrl[1].item())
+                native = bool(int(ctrl[2].item())) and native_comm is not None
                 if use_cuda:
                     with torch.cuda.stream(stream):
-                        for _ in range(k):
-                            handle(OP_PULL)
-                            handle(OP_PUSH)
+                        run_session(k, native)
                 else:
-                    for _ in range(k):
-                        handle(OP_PULL)
-                        handle(OP_PUSH)
+                    run_session(k, native)
                 continue
             if use_cuda:
                 with torch.cuda.stream(stream):

@@ -293,8 +293,16 @@ class Worker:
         eager.
         """
         self.ensure_graph(mbs)
-        if not self._graph_state["full_step"] \
-                and hasattr(self.transport, "run_begin"):
+        g = self._graph_state
+        if not g["full_step"] and hasattr(self.transport, "run_session") \
+                and (g["gather_inside"] or g["xb"] is self.x):
+            # whole session in one GIL-free C++ call over the raw comm
+            if self.transport.run_session(
+                    k, self.flat_params, self.flat_grads,
+                    g["graph"].raw_cuda_graph_exec()):
+                self.steps_done += k
+                return k * mbs
+        if not g["full_step"] and hasattr(self.transport, "run_begin"):
             self.transport.run_begin(k)  # zero per-step control traffic
         for _ in range(k):
             self.replay_once()

@@ -179,3 +179,26 @@ def test_multi_worker_streams_gpu():
     assert eng.ps.update_count == 40
     assert torch.isfinite(eng.ps.weights).all()
     eng.close()
+
+
+def test_multi_replay_native_gpu():
+    """The C++ multi_replay loop (W graphs x k round-robin launches)
+    produces the same update counts as the Python stream loop."""
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import mnist_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    cfg = TrainConfig(iters=1, seed=6)
+    eng = build_engine(build_graph(mnist_mlp),
+                       OptimizerSpec("adam", 0.001), cfg, device=DEV,
+                       workers_per_gpu=2)
+    rng = np.random.default_rng(6)
+    eng.stage_all(rng.random((512, 784), dtype=np.float32),
+                  rng.integers(0, 10, (512, 1)).astype(np.float32))
+    samples = eng.run_steps(7, 128)  # engine picks the C++ multi_replay
+    torch.cuda.synchronize()
+    assert samples == 7 * 128 * 2
+    assert eng.ps.update_count == 14
+    assert torch.isfinite(eng.ps.weights).all()
+    eng.close()
