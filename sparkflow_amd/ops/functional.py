@@ -144,31 +144,44 @@ class SoftmaxXent(torch.autograd.Function):
 
 class SoftmaxXentIdx(torch.autograd.Function):
     """Class-index label path — no one-hot materialization (the fused
-    kernel reads int64 labels directly)."""
+    kernel reads int64 OR integral-float labels directly).
+
+    Under ``grad_only()`` the forward kernel writes the GRADIENT
+    ``(probs - onehot)/m`` instead of the probs and backward returns it
+    as-is — the whole loss fwd+bwd is ONE kernel.  That contract assumes
+    a unit upstream gradient (which ``grad_only`` callers guarantee by
+    backpropagating a ones root)."""
 
     @staticmethod
     def forward(ctx, logits, labels):
+        grad_fused = False
         if ops.use_hip(logits):
-            loss, probs = ops.hip.softmax_xent_idx_fwd(logits, labels,
-                                                       _want_loss_value())
+            grad_fused = not _want_loss_value()
+            loss, probs = ops.hip.softmax_xent_idx_fwd(
+                logits, labels, not grad_fused, grad_fused)
         else:
+            lab = labels.long()
             lse = torch.logsumexp(logits, dim=1, keepdim=True)
             logp = logits - lse
-            loss = -logp.gather(1, labels.view(-1, 1)).mean()
+            loss = -logp.gather(1, lab.view(-1, 1)).mean()
             probs = torch.exp(logp)
         ctx.save_for_backward(probs, labels)
+        ctx.grad_fused = grad_fused
         return loss
 
     @staticmethod
     def backward(ctx, dloss):
         probs, labels = ctx.saved_tensors
         m = probs.shape[0]
+        if ctx.grad_fused:
+            return probs, None  # probs holds (softmax - onehot)/m already
         if ops.use_hip(probs):
             dlogits = ops.hip.softmax_xent_idx_bwd(probs, labels,
                                                    _dev_scale(dloss), 1.0 / m)
         else:
             dlogits = probs.clone()
-            dlogits[torch.arange(m, device=probs.device), labels] -= 1.0
+            dlogits[torch.arange(m, device=probs.device),
+                    labels.long()] -= 1.0
             dlogits.mul_(dloss / m)
         return dlogits, None
 
@@ -179,6 +192,10 @@ def softmax_xent(logits, labels):
     if labels.dim() == 2 and labels.size(1) == 1 and logits.size(1) != 1:
         labels = labels.reshape(-1)
     if labels.dim() == 1:
+        if ops.use_hip(logits) and labels.dtype in (torch.float32,
+                                                    torch.int64):
+            # HIP kernels read float labels directly — no cast kernel
+            return SoftmaxXentIdx.apply(logits, labels.contiguous())
         return SoftmaxXentIdx.apply(logits, labels.long().contiguous())
     return SoftmaxXent.apply(logits, labels.to(logits.dtype).contiguous())
 

@@ -30,7 +30,8 @@ torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,
                                torch::Tensor gscale, double mult);
 std::vector<torch::Tensor> softmax_xent_idx_fwd(torch::Tensor logits,
                                                 torch::Tensor labels,
-                                                bool want_value);
+                                                bool want_value,
+                                                bool dlogits_mode);
 torch::Tensor softmax_xent_idx_bwd(torch::Tensor probs, torch::Tensor labels,
                                    torch::Tensor gscale, double mult);
 torch::Tensor mse_fwd(torch::Tensor pred, torch::Tensor target,
@@ -105,8 +106,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("logits"), py::arg("onehot"), py::arg("want_value") = true);
   m.def("softmax_xent_bwd", &softmax_xent_bwd, "softmax+xent bwd");
   m.def("softmax_xent_idx_fwd", &softmax_xent_idx_fwd,
-        "fused softmax+xent fwd, class-index labels", py::arg("logits"),
-        py::arg("labels"), py::arg("want_value") = true);
+        "fused softmax+xent fwd, class-index labels (int64 or fp32); "
+        "dlogits_mode fuses the backward into the forward",
+        py::arg("logits"), py::arg("labels"), py::arg("want_value") = true,
+        py::arg("dlogits_mode") = false);
   m.def("softmax_xent_idx_bwd", &softmax_xent_idx_bwd,
         "softmax+xent bwd, class-index labels");
   m.def("mse_fwd", &mse_fwd, "fused MSE loss", py::arg("pred"),

@@ -44,10 +44,16 @@ __global__ void softmax_xent_fwd_kernel(const float* __restrict__ logits,
   }
 }
 
-// Class-index label variant: skips the host-side one-hot scatter entirely
-// (labels arrive as int64 [M], the engine's natural label form).
+// Class-index label variant: skips the host-side one-hot scatter entirely.
+// LT templates the label dtype (int64 OR float holding integral values —
+// the engine's labels arrive as float [M,1], so reading them directly
+// saves a cast kernel per step).  DLOGITS mode (the grad_only hot path)
+// writes the GRADIENT (probs - onehot) * inv_m instead of the probs —
+// the whole loss fwd+bwd collapses into this one kernel (valid because
+// the training loop backpropagates a unit upstream grad).
+template <typename LT, bool DLOGITS>
 __global__ void softmax_xent_idx_fwd_kernel(const float* __restrict__ logits,
-                                            const long* __restrict__ labels,
+                                            const LT* __restrict__ labels,
                                             float* __restrict__ probs,
                                             float* __restrict__ loss_sum,
                                             float inv_m, int M, int C) {
@@ -55,6 +61,7 @@ __global__ void softmax_xent_idx_fwd_kernel(const float* __restrict__ logits,
   const int lane = threadIdx.x & (WAVE - 1);
   if (row >= M) return;
   const float* xr = logits + (long)row * C;
+  const int lab = (int)labels[row];
 
   float mx = -INFINITY;
   for (int c = lane; c < C; c += WAVE) mx = fmaxf(mx, xr[c]);
@@ -66,19 +73,22 @@ __global__ void softmax_xent_idx_fwd_kernel(const float* __restrict__ logits,
 
   const float inv = 1.0f / se;
   for (int c = lane; c < C; c += WAVE) {
-    probs[(long)row * C + c] = __expf(xr[c] - mx) * inv;
+    const float p = __expf(xr[c] - mx) * inv;
+    probs[(long)row * C + c] =
+        DLOGITS ? (p - (c == lab ? 1.0f : 0.0f)) * inv_m : p;
   }
   if (lane == 0 && loss_sum) {
     const float lse = mx + __logf(se);
-    atomicAdd(loss_sum, (lse - xr[labels[row]]) * inv_m);
+    atomicAdd(loss_sum, (lse - xr[lab]) * inv_m);
   }
 }
 
 // The upstream loss grad arrives as a DEVICE scalar (gscale) and is read
 // on-device: converting it to a host float would d2h-sync and break
 // hipGraph stream capture of the full training step.
+template <typename LT>
 __global__ void softmax_xent_idx_bwd_kernel(const float* __restrict__ probs,
-                                            const long* __restrict__ labels,
+                                            const LT* __restrict__ labels,
                                             float* __restrict__ dlogits,
                                             const float* __restrict__ gscale,
                                             float mult, int M, int C) {
@@ -169,24 +179,50 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
   return {loss, probs};
 }
 
+static void check_labels(const torch::Tensor& labels) {
+  TORCH_CHECK(labels.is_cuda() && labels.is_contiguous() &&
+                  (labels.scalar_type() == torch::kInt64 ||
+                   labels.scalar_type() == torch::kFloat32),
+              "labels must be contiguous int64 or fp32 on GPU");
+}
+
 std::vector<torch::Tensor> softmax_xent_idx_fwd(torch::Tensor logits,
                                                 torch::Tensor labels,
-                                                bool want_value) {
+                                                bool want_value,
+                                                bool dlogits_mode) {
   check_f32l(logits, "logits");
-  TORCH_CHECK(labels.is_cuda() && labels.scalar_type() == torch::kInt64 &&
-                  labels.is_contiguous(),
-              "labels must be contiguous int64 on GPU");
+  check_labels(labels);
   const int M = logits.size(0), C = logits.size(1);
   auto probs = torch::empty_like(logits);
   auto loss = want_value ? torch::zeros({}, logits.options())
                          : torch::empty({}, logits.options());
   const int wpb = 4;
-  hipLaunchKernelGGL(softmax_xent_idx_fwd_kernel, dim3(ceil_div(M, wpb)),
-                     dim3(wpb * WAVE), 0, at::cuda::getCurrentCUDAStream(),
-                     logits.data_ptr<float>(), labels.data_ptr<long>(),
-                     probs.data_ptr<float>(),
-                     want_value ? loss.data_ptr<float>() : nullptr,
-                     1.0f / (float)M, M, C);
+  const dim3 grid(ceil_div(M, wpb)), blk(wpb * WAVE);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  float* lp = want_value ? loss.data_ptr<float>() : nullptr;
+  const float inv_m = 1.0f / (float)M;
+  const bool is_long = labels.scalar_type() == torch::kInt64;
+  if (is_long && dlogits_mode) {
+    hipLaunchKernelGGL((softmax_xent_idx_fwd_kernel<long, true>), grid, blk,
+                       0, stream, logits.data_ptr<float>(),
+                       labels.data_ptr<long>(), probs.data_ptr<float>(), lp,
+                       inv_m, M, C);
+  } else if (is_long) {
+    hipLaunchKernelGGL((softmax_xent_idx_fwd_kernel<long, false>), grid, blk,
+                       0, stream, logits.data_ptr<float>(),
+                       labels.data_ptr<long>(), probs.data_ptr<float>(), lp,
+                       inv_m, M, C);
+  } else if (dlogits_mode) {
+    hipLaunchKernelGGL((softmax_xent_idx_fwd_kernel<float, true>), grid, blk,
+                       0, stream, logits.data_ptr<float>(),
+                       labels.data_ptr<float>(), probs.data_ptr<float>(), lp,
+                       inv_m, M, C);
+  } else {
+    hipLaunchKernelGGL((softmax_xent_idx_fwd_kernel<float, false>), grid,
+                       blk, 0, stream, logits.data_ptr<float>(),
+                       labels.data_ptr<float>(), probs.data_ptr<float>(), lp,
+                       inv_m, M, C);
+  }
   return {loss, probs};
 }
 
@@ -194,15 +230,24 @@ torch::Tensor softmax_xent_idx_bwd(torch::Tensor probs, torch::Tensor labels,
                                    torch::Tensor gscale, double mult) {
   check_f32l(probs, "probs");
   check_f32l(gscale, "gscale");
+  check_labels(labels);
   const int M = probs.size(0), C = probs.size(1);
   auto d = torch::empty_like(probs);
   const long n = probs.numel();
   const int blocks = (int)std::min<long>(4096, (n + 255) / 256);
-  hipLaunchKernelGGL(softmax_xent_idx_bwd_kernel, dim3(blocks), dim3(256), 0,
-                     at::cuda::getCurrentCUDAStream(),
-                     probs.data_ptr<float>(), labels.data_ptr<long>(),
-                     d.data_ptr<float>(), gscale.data_ptr<float>(),
-                     (float)mult, M, C);
+  if (labels.scalar_type() == torch::kInt64) {
+    hipLaunchKernelGGL((softmax_xent_idx_bwd_kernel<long>), dim3(blocks),
+                       dim3(256), 0, at::cuda::getCurrentCUDAStream(),
+                       probs.data_ptr<float>(), labels.data_ptr<long>(),
+                       d.data_ptr<float>(), gscale.data_ptr<float>(),
+                       (float)mult, M, C);
+  } else {
+    hipLaunchKernelGGL((softmax_xent_idx_bwd_kernel<float>), dim3(blocks),
+                       dim3(256), 0, at::cuda::getCurrentCUDAStream(),
+                       probs.data_ptr<float>(), labels.data_ptr<float>(),
+                       d.data_ptr<float>(), gscale.data_ptr<float>(),
+                       (float)mult, M, C);
+  }
   return d;
 }
 

@@ -49,23 +49,20 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
   }
 }
 
-// Device-state Adam: the step count and bias-corrected lr live in a small
-// device buffer (state[0]=t, state[1]=lr_t) advanced by a 1-thread prep
-// kernel — so a PS apply can be captured in a hipGraph with the bias
-// correction still advancing per replay (a host-computed lr_t would be
-// frozen into the graph).
-__global__ void adam_prep_kernel(float* __restrict__ state, float lr,
-                                 float b1, float b2) {
-  const float t = state[0] + 1.0f;
-  state[0] = t;
-  state[1] = lr * __fsqrt_rn(1.0f - __powf(b2, t)) / (1.0f - __powf(b1, t));
-}
-
+// Device-state Adam in ONE launch: the step count and the NEXT step's
+// bias-corrected lr live in a small device buffer
+// (state[0]=t, state[1]=lr_t for step t+1, state[2]=block ticket), so a
+// PS apply is hipGraph-capturable with the bias correction still
+// advancing per replay.  Every thread reads this step's lr_t before the
+// block's __syncthreads(); the LAST block to finish (ticket) advances t
+// and precomputes lr_t for the next step — look-ahead scheduling that
+// replaces the former 1-thread prep launch.  Concurrent Hogwild applies
+// racing the ticket is the same benign lr_t race the prep kernel had.
 __global__ void adam_dev_kernel(float* __restrict__ p,
                                 const float* __restrict__ g,
                                 float* __restrict__ m, float* __restrict__ v,
-                                const float* __restrict__ state, float b1,
-                                float b2, float eps, long n) {
+                                float* __restrict__ state, float lr,
+                                float b1, float b2, float eps, long n) {
   const float lr_t = state[1];
   GS_LOOP(i, n) {
     const float gi = g[i];
@@ -74,6 +71,18 @@ __global__ void adam_dev_kernel(float* __restrict__ p,
     m[i] = mi;
     v[i] = vi;
     p[i] -= lr_t * mi / (__fsqrt_rn(vi) + eps);
+  }
+  __syncthreads();  // every thread of this block has consumed lr_t
+  if (threadIdx.x == 0) {
+    int* tick = reinterpret_cast<int*>(state + 2);
+    if (atomicAdd(tick, 1) == (int)gridDim.x - 1) {
+      *tick = 0;
+      const float tdone = state[0] + 1.0f;  // this step's index
+      state[0] = tdone;
+      const float tn = tdone + 1.0f;
+      state[1] = lr * __fsqrt_rn(1.0f - __powf(b2, tn)) /
+                 (1.0f - __powf(b1, tn));
+    }
   }
 }
 
@@ -148,16 +157,13 @@ void fused_adam_dev(torch::Tensor p, torch::Tensor g, torch::Tensor m,
   check_flat(g, "g", n);
   check_flat(m, "m", n);
   check_flat(v, "v", n);
-  TORCH_CHECK(state.numel() >= 2 && state.is_cuda(), "bad adam state");
+  TORCH_CHECK(state.numel() >= 3 && state.is_cuda(), "bad adam state");
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(adam_prep_kernel, dim3(1), dim3(1), 0, stream,
-                     state.data_ptr<float>(), (float)lr, (float)b1,
-                     (float)b2);
   hipLaunchKernelGGL(adam_dev_kernel, dim3(opt_blocks(n)), dim3(256), 0,
                      stream, p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(),
-                     state.data_ptr<float>(), (float)b1, (float)b2,
-                     (float)eps, n);
+                     state.data_ptr<float>(), (float)lr, (float)b1,
+                     (float)b2, (float)eps, n);
 }
 
 void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor ms,

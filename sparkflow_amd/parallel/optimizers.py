@@ -129,11 +129,22 @@ class Adam(FlatOptimizer):
         self.beta1, self.beta2, self.eps = float(beta1), float(beta2), float(epsilon)
         self.m = self._zeros()
         self.v = self._zeros()
-        # GPU path: the step count + bias-corrected lr live on-device
-        # (state[0]=t, state[1]=lr_t) so an apply is hipGraph-capturable.
-        self.dev_state = (torch.zeros(2, dtype=torch.float32,
-                                      device=self.device)
-                          if self.device.type == "cuda" else None)
+        # GPU path: [t, lr_t for step t+1, block ticket, pad] on-device so
+        # an apply is ONE hipGraph-capturable launch (the apply kernel's
+        # last block advances t and precomputes the next lr_t).
+        self.dev_state = None
+        if self.device.type == "cuda":
+            self.dev_state = torch.zeros(4, dtype=torch.float32,
+                                         device=self.device)
+            self._seed_dev_lr_t()
+
+    def _seed_dev_lr_t(self):
+        t_next = float(self.t) + 1.0
+        lr_t = self.lr * math.sqrt(1.0 - self.beta2 ** t_next) / (
+            1.0 - self.beta1 ** t_next)
+        self.dev_state[0] = float(self.t)
+        self.dev_state[1] = lr_t
+        self.dev_state[2] = 0.0
 
     def apply(self, param, grad):
         from sparkflow_amd import ops
@@ -168,7 +179,7 @@ class Adam(FlatOptimizer):
     def set_step_count(self, t: int) -> None:
         self.t = int(t)
         if self.dev_state is not None:
-            self.dev_state[0] = float(t)
+            self._seed_dev_lr_t()
 
 
 class RMSProp(FlatOptimizer):
