@@ -93,6 +93,15 @@ class LinearAct(torch.autograd.Function):
         need_dx = ctx.needs_input_grad[0]
         dy = dy.contiguous()
         if ops.use_hip(x):
+            if act != ACT_NONE and ctx.has_bias:
+                # activation backward fused into BOTH backward GEMMs'
+                # operand loads — no standalone act_bwd pass or dpre buf
+                dx = ops.hip.gemm_dgrad_act(dy, y, w, act) \
+                    if need_dx else None
+                dwb = ops.hip.gemm_wgrad_bias_act(x, dy, y, act)
+                k = x.shape[1]
+                dw, db = dwb[:k], dwb[k]
+                return dx, dw, db, None
             dpre = ops.hip.act_bwd(y, dy, act) if act != ACT_NONE else dy
             dx = ops.hip.gemm_dgrad(dpre, w) if need_dx else None
             if ctx.has_bias:

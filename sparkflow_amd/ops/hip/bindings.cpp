@@ -9,8 +9,12 @@ void sf_set_parallel_hint(int n);
 torch::Tensor gemm_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                        int64_t act);
 torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w);
+torch::Tensor gemm_dgrad_act(torch::Tensor dy, torch::Tensor y,
+                             torch::Tensor w, int64_t act);
 torch::Tensor gemm_wgrad(torch::Tensor x, torch::Tensor dy);
 torch::Tensor gemm_wgrad_bias(torch::Tensor x, torch::Tensor dy);
+torch::Tensor gemm_wgrad_bias_act(torch::Tensor x, torch::Tensor dy,
+                                  torch::Tensor y, int64_t act);
 torch::Tensor bias_grad(torch::Tensor dy);
 
 // elementwise.hip
@@ -89,9 +93,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sparkflow_amd gfx950 (CDNA4) kernels";
   m.def("gemm_act", &gemm_act, "fused fp32 MFMA GEMM + bias + activation");
   m.def("gemm_dgrad", &gemm_dgrad, "dx = dy @ W^T");
+  m.def("gemm_dgrad_act", &gemm_dgrad_act,
+        "dx = act_grad(y, dy) @ W^T (act bwd fused into the A load)");
   m.def("gemm_wgrad", &gemm_wgrad, "dW = x^T @ dy");
   m.def("gemm_wgrad_bias", &gemm_wgrad_bias,
         "dWb = [x;ones]^T @ dy (bias grad fused as last row)");
+  m.def("gemm_wgrad_bias_act", &gemm_wgrad_bias_act,
+        "dWb = [x;ones]^T @ act_grad(y, dy) (act bwd fused into B load)");
   m.def("bias_grad", &bias_grad, "column-sum bias gradient");
   m.def("act_bwd", &act_bwd, "activation backward from output");
   m.def("argmax_rows", &argmax_rows, "rowwise argmax head");

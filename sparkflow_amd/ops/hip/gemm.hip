@@ -30,11 +30,17 @@
 // to a [splitk, M, N] workspace with plain stores (no zero-fill pass, no
 // atomics, deterministic); a small reduce epilogue sums the slices and
 // applies bias+activation.
-template <bool TA, bool TB, bool HAS_BIAS, bool SPLITK, bool BIASROW = false>
+// DPRE_A / DPRE_B: the A (TA=0) or B (TB=0) operand is the activation
+// backward dpre = act_grad(preY, A_phys) computed AT LOAD TIME — fusing
+// the standalone act_bwd kernel (and its dpre HBM round trip) into the
+// dgrad / wgrad GEMMs that consume it.
+template <bool TA, bool TB, bool HAS_BIAS, bool SPLITK,
+          bool BIASROW = false, bool DPRE_A = false, bool DPRE_B = false>
 __global__ __launch_bounds__(256) void gemm_f32_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     const float* __restrict__ bias, float* __restrict__ C, int M, int N,
-    int K, int lda, int ldb, int act, int k_chunk) {
+    int K, int lda, int ldb, int act, int k_chunk,
+    const float* __restrict__ preY = nullptr, int pre_act = 0) {
   __shared__ float As[BM][BK + 1];
   __shared__ float Bs[BK][BN + 4];
 
@@ -70,12 +76,25 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
       const int gm = m0 + a_row;
       if (gm < M && k0 + BK <= K) {
         r = *reinterpret_cast<const float4*>(&A[(long)gm * lda + k0 + a_c4]);
+        if (DPRE_A) {
+          const float4 y4 = *reinterpret_cast<const float4*>(
+              &preY[(long)gm * lda + k0 + a_c4]);
+          float* rr = reinterpret_cast<float*>(&r);
+          const float* yy = reinterpret_cast<const float*>(&y4);
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+            rr[i] = act_grad_from_out(yy[i], rr[i], pre_act);
+        }
       } else {
         float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
           const int gk = k0 + a_c4 + i;
-          rr[i] = (gm < M && gk < K) ? A[(long)gm * lda + gk] : 0.0f;
+          const bool ok = gm < M && gk < K;
+          float v = ok ? A[(long)gm * lda + gk] : 0.0f;
+          if (DPRE_A && ok)
+            v = act_grad_from_out(preY[(long)gm * lda + gk], v, pre_act);
+          rr[i] = v;
         }
       }
     } else {
@@ -107,12 +126,25 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
       const int gk = k0 + b_kk;
       if (gk < K && n0 + BN <= N) {
         r = *reinterpret_cast<const float4*>(&B[(long)gk * ldb + n0 + b_n4]);
+        if (DPRE_B) {
+          const float4 y4 = *reinterpret_cast<const float4*>(
+              &preY[(long)gk * ldb + n0 + b_n4]);
+          float* rr = reinterpret_cast<float*>(&r);
+          const float* yy = reinterpret_cast<const float*>(&y4);
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+            rr[i] = act_grad_from_out(yy[i], rr[i], pre_act);
+        }
       } else {
         float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
           const int gn = n0 + b_n4 + i;
-          rr[i] = (gk < K && gn < N) ? B[(long)gk * ldb + gn] : 0.0f;
+          const bool ok = gk < K && gn < N;
+          float v = ok ? B[(long)gk * ldb + gn] : 0.0f;
+          if (DPRE_B && ok)
+            v = act_grad_from_out(preY[(long)gk * ldb + gn], v, pre_act);
+          rr[i] = v;
         }
       }
     } else {
@@ -280,10 +312,11 @@ static int pick_splitk(int M, int N, int K) {
   return splitk;
 }
 
-template <bool TA, bool TB>
+template <bool TA, bool TB, bool DPRE_A = false, bool DPRE_B = false>
 static void launch_gemm(const torch::Tensor& A, const torch::Tensor& B,
                         const torch::Tensor& bias, torch::Tensor& C, int M,
-                        int N, int K, int lda, int ldb, int act) {
+                        int N, int K, int lda, int ldb, int act,
+                        const float* preY = nullptr, int pre_act = 0) {
   auto stream = at::cuda::getCurrentCUDAStream();
   const bool has_bias = bias.numel() > 0;
   const int splitk = pick_splitk(M, N, K);
@@ -291,10 +324,11 @@ static void launch_gemm(const torch::Tensor& A, const torch::Tensor& B,
     const int k_chunk = ceil_div(K, splitk * BK) * BK;
     auto ws = torch::empty({splitk, M, N}, C.options());
     dim3 grid(ceil_div(N, BN), ceil_div(M, BM), splitk);
-    hipLaunchKernelGGL((gemm_f32_kernel<TA, TB, false, true>), grid,
-                       dim3(256), 0, stream, A.data_ptr<float>(),
-                       B.data_ptr<float>(), nullptr, ws.data_ptr<float>(), M,
-                       N, K, lda, ldb, act, k_chunk);
+    hipLaunchKernelGGL(
+        (gemm_f32_kernel<TA, TB, false, true, false, DPRE_A, DPRE_B>), grid,
+        dim3(256), 0, stream, A.data_ptr<float>(), B.data_ptr<float>(),
+        nullptr, ws.data_ptr<float>(), M, N, K, lda, ldb, act, k_chunk,
+        preY, pre_act);
     const long n = (long)M * N;
     const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
     if (has_bias) {
@@ -311,15 +345,17 @@ static void launch_gemm(const torch::Tensor& A, const torch::Tensor& B,
   }
   dim3 grid(ceil_div(N, BN), ceil_div(M, BM));
   if (has_bias) {
-    hipLaunchKernelGGL((gemm_f32_kernel<TA, TB, true, false>), grid,
-                       dim3(256), 0, stream, A.data_ptr<float>(),
-                       B.data_ptr<float>(), bias.data_ptr<float>(),
-                       C.data_ptr<float>(), M, N, K, lda, ldb, act, K);
+    hipLaunchKernelGGL(
+        (gemm_f32_kernel<TA, TB, true, false, false, DPRE_A, DPRE_B>), grid,
+        dim3(256), 0, stream, A.data_ptr<float>(), B.data_ptr<float>(),
+        bias.data_ptr<float>(), C.data_ptr<float>(), M, N, K, lda, ldb, act,
+        K, preY, pre_act);
   } else {
-    hipLaunchKernelGGL((gemm_f32_kernel<TA, TB, false, false>), grid,
-                       dim3(256), 0, stream, A.data_ptr<float>(),
-                       B.data_ptr<float>(), nullptr, C.data_ptr<float>(), M,
-                       N, K, lda, ldb, act, K);
+    hipLaunchKernelGGL(
+        (gemm_f32_kernel<TA, TB, false, false, false, DPRE_A, DPRE_B>),
+        grid, dim3(256), 0, stream, A.data_ptr<float>(),
+        B.data_ptr<float>(), nullptr, C.data_ptr<float>(), M, N, K, lda,
+        ldb, act, K, preY, pre_act);
   }
 }
 
@@ -347,6 +383,23 @@ torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w) {
   return dx;
 }
 
+torch::Tensor gemm_dgrad_act(torch::Tensor dy, torch::Tensor y,
+                             torch::Tensor w, int64_t act) {
+  // dx = act_grad(y, dy) @ w^T with the activation backward fused into
+  // the A-operand load (no standalone act_bwd pass or dpre buffer).
+  check_f32(dy, "dy");
+  check_f32(y, "y");
+  check_f32(w, "w");
+  const int M = dy.size(0), Nout = dy.size(1), Kin = w.size(0);
+  TORCH_CHECK(w.size(1) == Nout, "shape mismatch dgrad");
+  auto dx = torch::empty({M, Kin}, dy.options());
+  auto empty = torch::empty({0}, dy.options());
+  launch_gemm<false, true, true, false>(dy, w, empty, dx, M, Kin, Nout,
+                                        Nout, Nout, ACT_NONE,
+                                        y.data_ptr<float>(), (int)act);
+  return dx;
+}
+
 torch::Tensor gemm_wgrad(torch::Tensor x, torch::Tensor dy) {
   check_f32(x, "x");
   check_f32(dy, "dy");
@@ -360,10 +413,13 @@ torch::Tensor gemm_wgrad(torch::Tensor x, torch::Tensor dy) {
   return dw;
 }
 
-torch::Tensor gemm_wgrad_bias(torch::Tensor x, torch::Tensor dy) {
-  // dwb[Kin+1, Nout] = [x; ones]^T @ dy : rows 0..Kin-1 are dW, row Kin
-  // is the bias gradient (column sums) — one GEMM instead of a GEMM plus
-  // a separate reduction kernel.
+template <bool DPRE_B>
+static torch::Tensor wgrad_bias_impl(torch::Tensor x, torch::Tensor dy,
+                                     const float* preY, int pre_act) {
+  // dwb[Kin+1, Nout] = [x; ones]^T @ dpre : rows 0..Kin-1 are dW, row
+  // Kin is the bias gradient (column sums) — one GEMM instead of a GEMM
+  // plus a separate reduction kernel; with DPRE_B the activation
+  // backward is fused into the B-operand load as well.
   check_f32(x, "x");
   check_f32(dy, "dy");
   const int Mb = x.size(0), Kin = x.size(1), Nout = dy.size(1);
@@ -376,10 +432,11 @@ torch::Tensor gemm_wgrad_bias(torch::Tensor x, torch::Tensor dy) {
     const int k_chunk = ceil_div(Mb, splitk * BK) * BK;
     auto ws = torch::empty({splitk, M, Nout}, x.options());
     dim3 grid(ceil_div(Nout, BN), ceil_div(M, BM), splitk);
-    hipLaunchKernelGGL((gemm_f32_kernel<true, false, false, true, true>),
-                       grid, dim3(256), 0, stream, x.data_ptr<float>(),
-                       dy.data_ptr<float>(), nullptr, ws.data_ptr<float>(),
-                       M, Nout, Mb, Kin, Nout, ACT_NONE, k_chunk);
+    hipLaunchKernelGGL(
+        (gemm_f32_kernel<true, false, false, true, true, false, DPRE_B>),
+        grid, dim3(256), 0, stream, x.data_ptr<float>(),
+        dy.data_ptr<float>(), nullptr, ws.data_ptr<float>(), M, Nout, Mb,
+        Kin, Nout, ACT_NONE, k_chunk, preY, pre_act);
     const long n = (long)M * Nout;
     const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
     hipLaunchKernelGGL((splitk_reduce_kernel<false>), dim3(blocks),
@@ -387,12 +444,23 @@ torch::Tensor gemm_wgrad_bias(torch::Tensor x, torch::Tensor dy) {
                        dwb.data_ptr<float>(), n, Nout, splitk, ACT_NONE);
   } else {
     dim3 grid(ceil_div(Nout, BN), ceil_div(M, BM));
-    hipLaunchKernelGGL((gemm_f32_kernel<true, false, false, false, true>),
-                       grid, dim3(256), 0, stream, x.data_ptr<float>(),
-                       dy.data_ptr<float>(), nullptr, dwb.data_ptr<float>(),
-                       M, Nout, Mb, Kin, Nout, ACT_NONE, Mb);
+    hipLaunchKernelGGL(
+        (gemm_f32_kernel<true, false, false, false, true, false, DPRE_B>),
+        grid, dim3(256), 0, stream, x.data_ptr<float>(),
+        dy.data_ptr<float>(), nullptr, dwb.data_ptr<float>(), M, Nout, Mb,
+        Kin, Nout, ACT_NONE, Mb, preY, pre_act);
   }
   return dwb;
+}
+
+torch::Tensor gemm_wgrad_bias(torch::Tensor x, torch::Tensor dy) {
+  return wgrad_bias_impl<false>(x, dy, nullptr, 0);
+}
+
+torch::Tensor gemm_wgrad_bias_act(torch::Tensor x, torch::Tensor dy,
+                                  torch::Tensor y, int64_t act) {
+  check_f32(y, "y");
+  return wgrad_bias_impl<true>(x, dy, y.data_ptr<float>(), (int)act);
 }
 
 torch::Tensor bias_grad(torch::Tensor dy) {

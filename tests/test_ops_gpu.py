@@ -264,3 +264,23 @@ def test_pack_segments_gpu():
     flat = torch.empty(sum(s.numel() for s in segs), device=DEV)
     ops.hip.pack_segments(segs, flat)
     torch.testing.assert_close(flat, torch.cat(segs))
+
+
+@pytest.mark.parametrize("act", [1, 2])
+def test_gemm_act_fused_bwd_gpu(act):
+    """dgrad/wgrad with the activation backward fused into the operand
+    load == the act_bwd + plain GEMM composition."""
+    torch.manual_seed(12)
+    M, K, N = 300, 784, 256
+    x = torch.randn(M, K, device=DEV)
+    w = torch.randn(K, N, device=DEV) * 0.05
+    dy = torch.randn(M, N, device=DEV)
+    pre = x @ w
+    y = torch.relu(pre) if act == 1 else torch.sigmoid(pre)
+    dpre = ops.hip.act_bwd(y, dy, act)
+    dx_ref = ops.hip.gemm_dgrad(dpre, w)
+    dwb_ref = ops.hip.gemm_wgrad_bias(x, dpre)
+    dx = ops.hip.gemm_dgrad_act(dy, y, w, act)
+    dwb = ops.hip.gemm_wgrad_bias_act(x, dy, y, act)
+    torch.testing.assert_close(dx, dx_ref, atol=1e-4, rtol=1e-4)
+    torch.testing.assert_close(dwb, dwb_ref, atol=1e-4, rtol=1e-4)
