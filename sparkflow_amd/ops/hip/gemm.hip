@@ -252,6 +252,33 @@ __global__ void splitk_reduce_kernel(const float* __restrict__ W,
   }
 }
 
+// float4 variant (n % 4 == 0, and N % 4 == 0 when biased, so a bias
+// float4 never straddles a row boundary).
+template <bool HAS_BIAS>
+__global__ void splitk_reduce4_kernel(const float4* __restrict__ W,
+                                      const float* __restrict__ bias,
+                                      float4* __restrict__ C, long n4, int N,
+                                      int splitk, int act) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 s = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (HAS_BIAS) {
+      const int col = (int)((i * 4) % N);
+      s = *reinterpret_cast<const float4*>(&bias[col]);
+    }
+    for (int z = 0; z < splitk; ++z) {
+      const float4 w = W[(long)z * n4 + i];
+      s.x += w.x; s.y += w.y; s.z += w.z; s.w += w.w;
+    }
+    s.x = act_apply(s.x, act);
+    s.y = act_apply(s.y, act);
+    s.z = act_apply(s.z, act);
+    s.w = act_apply(s.w, act);
+    C[i] = s;
+  }
+}
+
 // Column sums with 2D parallelism: blockIdx.y owns a row chunk so narrow
 // layers (N=10..256) still fill more than one CU; partial sums combine
 // with one atomicAdd per (col, chunk).  db must arrive zeroed.
@@ -330,6 +357,23 @@ static void launch_gemm(const torch::Tensor& A, const torch::Tensor& B,
         nullptr, ws.data_ptr<float>(), M, N, K, lda, ldb, act, k_chunk,
         preY, pre_act);
     const long n = (long)M * N;
+    const bool v4 = (n & 3) == 0 && (!has_bias || (N & 3) == 0);
+    if (v4) {
+      const long n4 = n >> 2;
+      const int blocks = (int)std::min<long>(2048, (n4 + 255) / 256);
+      auto* w4 = reinterpret_cast<const float4*>(ws.data_ptr<float>());
+      auto* c4 = reinterpret_cast<float4*>(C.data_ptr<float>());
+      if (has_bias) {
+        hipLaunchKernelGGL((splitk_reduce4_kernel<true>), dim3(blocks),
+                           dim3(256), 0, stream, w4,
+                           bias.data_ptr<float>(), c4, n4, N, splitk, act);
+      } else {
+        hipLaunchKernelGGL((splitk_reduce4_kernel<false>), dim3(blocks),
+                           dim3(256), 0, stream, w4, nullptr, c4, n4, N,
+                           splitk, act);
+      }
+      return;
+    }
     const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
     if (has_bias) {
       hipLaunchKernelGGL((splitk_reduce_kernel<true>), dim3(blocks),
@@ -438,10 +482,20 @@ static torch::Tensor wgrad_bias_impl(torch::Tensor x, torch::Tensor dy,
         dy.data_ptr<float>(), nullptr, ws.data_ptr<float>(), M, Nout, Mb,
         Kin, Nout, ACT_NONE, k_chunk, preY, pre_act);
     const long n = (long)M * Nout;
-    const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
-    hipLaunchKernelGGL((splitk_reduce_kernel<false>), dim3(blocks),
-                       dim3(256), 0, stream, ws.data_ptr<float>(), nullptr,
-                       dwb.data_ptr<float>(), n, Nout, splitk, ACT_NONE);
+    if ((n & 3) == 0) {
+      const long n4 = n >> 2;
+      const int blocks = (int)std::min<long>(2048, (n4 + 255) / 256);
+      hipLaunchKernelGGL(
+          (splitk_reduce4_kernel<false>), dim3(blocks), dim3(256), 0,
+          stream, reinterpret_cast<const float4*>(ws.data_ptr<float>()),
+          nullptr, reinterpret_cast<float4*>(dwb.data_ptr<float>()), n4,
+          Nout, splitk, ACT_NONE);
+    } else {
+      const int blocks = (int)std::min<long>(2048, (n + 255) / 256);
+      hipLaunchKernelGGL((splitk_reduce_kernel<false>), dim3(blocks),
+                         dim3(256), 0, stream, ws.data_ptr<float>(), nullptr,
+                         dwb.data_ptr<float>(), n, Nout, splitk, ACT_NONE);
+    }
   } else {
     dim3 grid(ceil_div(Nout, BN), ceil_div(M, BM));
     hipLaunchKernelGGL(

@@ -113,6 +113,13 @@ static inline int opt_blocks(long n) {
   return (int)std::min<long>(2048, (n + 255) / 256);
 }
 
+// The ticketed (look-ahead lr_t) adam kernel pays one same-address
+// atomic per block — cap its grid so the ticket line stays short; the
+// grid-stride loop amortizes the extra elements per block.
+static inline int adam_blocks(long n) {
+  return (int)std::min<long>(256, (n + 1023) / 1024);
+}
+
 void fused_sgd(torch::Tensor p, torch::Tensor g, double lr) {
   const long n = p.numel();
   check_flat(p, "p", n);
@@ -159,7 +166,7 @@ void fused_adam_dev(torch::Tensor p, torch::Tensor g, torch::Tensor m,
   check_flat(v, "v", n);
   TORCH_CHECK(state.numel() >= 3 && state.is_cuda(), "bad adam state");
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(adam_dev_kernel, dim3(opt_blocks(n)), dim3(256), 0,
+  hipLaunchKernelGGL(adam_dev_kernel, dim3(adam_blocks(n)), dim3(256), 0,
                      stream, p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(),
                      state.data_ptr<float>(), (float)lr, (float)b1,
