@@ -295,11 +295,15 @@ class Worker:
         self.ensure_graph(mbs)
         g = self._graph_state
         if not g["full_step"] and hasattr(self.transport, "run_session") \
-                and (g["gather_inside"] or g["xb"] is self.x):
+                and (g["gather_inside"] or g["xb"] is self.x) \
+                and getattr(self.transport, "native_comm", None) is not None:
             # whole session in one GIL-free C++ call over the raw comm
-            if self.transport.run_session(
-                    k, self.flat_params, self.flat_grads,
-                    g["graph"].raw_cuda_graph_exec()):
+            try:
+                exec_h = g["graph"].raw_cuda_graph_exec()
+            except Exception:
+                exec_h = None
+            if exec_h is not None and self.transport.run_session(
+                    k, self.flat_params, self.flat_grads, exec_h):
                 self.steps_done += k
                 return k * mbs
         if not g["full_step"] and hasattr(self.transport, "run_begin"):
