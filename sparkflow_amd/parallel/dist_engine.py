@@ -55,7 +55,9 @@ def _warmup_pairs(pairs: DistPairs, device) -> None:
     """Initialize every pair communicator SEQUENTIALLY from the main
     thread before the PS service threads start.  Lazy init from N-1
     concurrent service threads is where pairwise-NCCL designs deadlock;
-    one deterministic ping per pair avoids it."""
+    one deterministic ping per pair avoids it.  Pairs whose raw RCCL
+    channel exists skip the torch payload ping entirely — that
+    communicator would never be used, so it is never created."""
     import torch.distributed as dist
 
     t = torch.ones(1, device=device)
@@ -63,11 +65,13 @@ def _warmup_pairs(pairs: DistPairs, device) -> None:
     if pairs.rank == 0:
         for w in range(1, pairs.world):
             for i in range(pairs.workers_per_gpu):
-                dist.send(t, dst=w, group=pairs.payload[(w, i)])
+                if (w, i) not in pairs.native:
+                    dist.send(t, dst=w, group=pairs.payload[(w, i)])
                 dist.send(c, dst=w, group=pairs.ctrl[(w, i)])
     else:
         for i in range(pairs.workers_per_gpu):
-            dist.recv(t, src=0, group=pairs.payload[(pairs.rank, i)])
+            if (pairs.rank, i) not in pairs.native:
+                dist.recv(t, src=0, group=pairs.payload[(pairs.rank, i)])
             dist.recv(c, src=0, group=pairs.ctrl[(pairs.rank, i)])
     if device.type == "cuda":
         torch.cuda.synchronize(device)
@@ -175,6 +179,7 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
     ps = service = None
     if world > 1:
         pairs = DistPairs(workers_per_gpu=max(1, workers_per_gpu))
+        pairs.init_native(device)  # raw RCCL channel (GIL-free sessions)
         _warmup_pairs(pairs, device)
     if rank == 0:
         ps = PSCore(graph_json, optimizer, device=device,

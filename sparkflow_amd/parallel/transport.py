@@ -338,12 +338,17 @@ class DistTransport(WorkerTransport):
         self.pull_into(self.buf)
         return self.buf
 
+    def _use_raw(self) -> bool:
+        # Whenever the raw comm exists it carries ALL payload (sessions
+        # AND per-op) so the torch payload pair group is never touched —
+        # build_engine then skips creating those communicators entirely.
+        return self.native_comm is not None
+
     def pull_into(self, out: torch.Tensor) -> None:
+        raw = self._use_raw()
         if self._session <= 0:
-            self._ctrl(OP_PULL)
-            self.dist.recv(out, src=0, group=self.pg)
-            return
-        if self._session_native:
+            self._ctrl(OP_PULL, chan=1 if raw else 0)
+        if raw:
             from sparkflow_amd import ops
 
             ops.hip.raw_recv(self.native_comm, 0, out)
@@ -351,12 +356,12 @@ class DistTransport(WorkerTransport):
             self.dist.recv(out, src=0, group=self.pg)
 
     def push(self, flat_grads: torch.Tensor) -> None:
+        raw = self._use_raw()
         if self._session <= 0:
-            self._ctrl(OP_PUSH)
-            self.dist.send(flat_grads.contiguous(), dst=0, group=self.pg)
-            return
-        self._session -= 1
-        if self._session_native:
+            self._ctrl(OP_PUSH, chan=1 if raw else 0)
+        else:
+            self._session -= 1
+        if raw:
             from sparkflow_amd import ops
 
             ops.hip.raw_send(self.native_comm, 0, flat_grads.contiguous())
@@ -419,6 +424,7 @@ class DistPSService:
         ctrl = torch.zeros(3, dtype=torch.int64)
 
         def handle(op: int, native: bool = False):
+            native = native and native_comm is not None
             if native:
                 from sparkflow_amd import ops
 
@@ -471,11 +477,12 @@ class DistPSService:
                 else:
                     run_session(k, native)
                 continue
+            per_op_native = bool(int(ctrl[2].item()))
             if use_cuda:
                 with torch.cuda.stream(stream):
-                    handle(op)
+                    handle(op, per_op_native)
             else:
-                handle(op)
+                handle(op, per_op_native)
 
     def join(self):
         for t in self._threads:
