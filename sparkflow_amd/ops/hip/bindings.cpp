@@ -70,6 +70,8 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t H,
 torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
                            int64_t Kw, int64_t stride, int64_t pad);
 torch::Tensor conv_bias_grad(torch::Tensor dy);
+torch::Tensor conv_dilate(torch::Tensor dy, int64_t s, int64_t extra_h,
+                          int64_t extra_w);
 std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
                                          int64_t stride, int64_t pad);
 torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
@@ -141,6 +143,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_dgrad", &conv2d_dgrad, "conv2d input grad");
   m.def("conv2d_wgrad", &conv2d_wgrad, "conv2d weight grad");
   m.def("conv_bias_grad", &conv_bias_grad, "conv bias grad");
+  m.def("conv_dilate", &conv_dilate,
+        "zero-stuff dy by stride (strided dgrad as stride-1 conv)");
   m.def("maxpool2d_fwd", &maxpool2d_fwd, "maxpool fwd with argmax indices");
   m.def("maxpool2d_bwd", &maxpool2d_bwd, "maxpool bwd scatter");
   m.def("nccl_unique_id", &nccl_unique_id, "RCCL unique id for a pair comm");

@@ -148,39 +148,44 @@ __global__ void conv_bias_grad_kernel(const float* __restrict__ dy,
 }
 
 // ---- maxpool with argmax indices (bwd scatters through them) ----
+// blockIdx.x owns one (n, c) plane, blockIdx.y chunks ho, threads sweep
+// wo — every coordinate is loop state, no div/mod anywhere in the sweep
+// (the flat-index decode was ~20x off the bandwidth bound on the
+// ResNet-shape 112x112 pool).
 __global__ void maxpool_fwd_kernel(const float* __restrict__ x,
                                    float* __restrict__ y,
                                    long* __restrict__ idx, int N, int C,
                                    int H, int W, int K, int stride, int pad,
                                    int Ho, int Wo) {
-  const long total = (long)N * C * Ho * Wo;
-  const long gstride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += gstride) {
-    const int wo = i % Wo;
-    const int ho = (i / Wo) % Ho;
-    const int c = (i / ((long)Wo * Ho)) % C;
-    const int n = i / ((long)Wo * Ho * C);
-    const float* xp = x + ((long)n * C + c) * H * W;
-    float best = -INFINITY;
-    long best_hw = 0;
-    for (int kh = 0; kh < K; ++kh) {
-      const int h = ho * stride - pad + kh;
-      if (h < 0) continue;
-      if (h >= H) break;
-      for (int kw = 0; kw < K; ++kw) {
-        const int ww = wo * stride - pad + kw;
-        if (ww < 0) continue;
-        if (ww >= W) break;
-        const float v = xp[(long)h * W + ww];
-        if (v > best) {
-          best = v;
-          best_hw = (long)h * W + ww;
+  const long plane = blockIdx.x;  // n * C + c
+  const float* xp = x + plane * H * W;
+  float* yp = y + plane * Ho * Wo;
+  long* ip = idx + plane * Ho * Wo;
+  for (int ho = blockIdx.y; ho < Ho; ho += gridDim.y) {
+    const int h0 = ho * stride - pad;
+    for (int wo = threadIdx.x; wo < Wo; wo += blockDim.x) {
+      const int w0 = wo * stride - pad;
+      float best = -INFINITY;
+      long best_hw = 0;
+      for (int kh = 0; kh < K; ++kh) {
+        const int h = h0 + kh;
+        if (h < 0) continue;
+        if (h >= H) break;
+        for (int kw = 0; kw < K; ++kw) {
+          const int ww = w0 + kw;
+          if (ww < 0) continue;
+          if (ww >= W) break;
+          const float v = xp[(long)h * W + ww];
+          if (v > best) {
+            best = v;
+            best_hw = (long)h * W + ww;
+          }
         }
       }
+      yp[(long)ho * Wo + wo] = best;
+      // torch max_pool2d index convention: within the plane
+      ip[(long)ho * Wo + wo] = best_hw;
     }
-    y[i] = best;
-    idx[i] = best_hw;  // torch max_pool2d index convention: within the plane
   }
 }
 
@@ -188,13 +193,13 @@ __global__ void maxpool_bwd_kernel(const float* __restrict__ dy,
                                    const long* __restrict__ idx,
                                    float* __restrict__ dx, int N, int C,
                                    int HW, int HoWo) {
-  const long total = (long)N * C * HoWo;
-  const long gstride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += gstride) {
-    const int c = (i / HoWo) % C;
-    const int n = i / ((long)HoWo * C);
-    atomicAdd(&dx[((long)n * C + c) * HW + idx[i]], dy[i]);
+  const long plane = blockIdx.x;  // n * C + c — no div/mod in the sweep
+  const float* dyp = dy + plane * HoWo;
+  const long* ip = idx + plane * HoWo;
+  float* dxp = dx + plane * HW;
+  const int span = gridDim.y * blockDim.x;
+  for (int i = blockIdx.y * blockDim.x + threadIdx.x; i < HoWo; i += span) {
+    atomicAdd(&dxp[ip[i]], dyp[i]);
   }
 }
 
@@ -262,6 +267,41 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
   return dw;
 }
 
+// Zero-stuff dy for the strided-dgrad-as-convolution trick: out[...,
+// i*s, j*s] = dy[..., i, j] in a zeroed [N, C, Hd, Wd] buffer (extra
+// trailing rows/cols cover the stride remainder).  The dilated tensor
+// then feeds the fast implicit-GEMM forward with flipped weights — the
+// direct strided dgrad kernel was 25% of the ResNet-shape step.
+__global__ void dilate_kernel(const float* __restrict__ dy,
+                              float* __restrict__ out, int Ho, int Wo,
+                              int s, long plane_out, int Wd) {
+  const long plane = blockIdx.x;  // n * C + c
+  const float* src = dy + plane * (long)Ho * Wo;
+  float* dst = out + plane * plane_out;
+  for (int ho = blockIdx.y; ho < Ho; ho += gridDim.y) {
+    for (int wo = threadIdx.x; wo < Wo; wo += blockDim.x) {
+      dst[((long)ho * s) * Wd + (long)wo * s] = src[(long)ho * Wo + wo];
+    }
+  }
+}
+
+torch::Tensor conv_dilate(torch::Tensor dy, int64_t s, int64_t extra_h,
+                          int64_t extra_w) {
+  check_f32c(dy, "dy");
+  const int N = dy.size(0), C = dy.size(1), Ho = dy.size(2),
+            Wo = dy.size(3);
+  const long Hd = (long)(Ho - 1) * s + 1 + extra_h;
+  const long Wd = (long)(Wo - 1) * s + 1 + extra_w;
+  auto out = torch::zeros({N, C, Hd, Wd}, dy.options());
+  const int chunks = std::max(1, std::min(Ho, 2048 / (N * C) + 1));
+  hipLaunchKernelGGL(dilate_kernel, dim3(N * C, chunks),
+                     dim3(std::min(256, std::max(64, Wo))), 0,
+                     at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
+                     out.data_ptr<float>(), Ho, Wo, (int)s, Hd * Wd,
+                     (int)Wd);
+  return out;
+}
+
 torch::Tensor conv_bias_grad(torch::Tensor dy) {
   check_f32c(dy, "dy");
   const int N = dy.size(0), C = dy.size(1), HW = dy.size(2) * dy.size(3);
@@ -283,9 +323,9 @@ std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
   auto y = torch::empty({N, C, Ho, Wo}, x.options());
   auto idx = torch::empty({N, C, Ho, Wo},
                           x.options().dtype(torch::kInt64));
-  const long total = y.numel();
-  const int blocks = (int)std::min<long>(8192, (total + 255) / 256);
-  hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(blocks), dim3(256), 0,
+  const int ho_chunks = std::max(1, std::min(Ho, 2048 / (N * C) + 1));
+  hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(N * C, ho_chunks),
+                     dim3(std::min(256, std::max(64, Wo))), 0,
                      at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
                      y.data_ptr<float>(), idx.data_ptr<long>(), N, C, H, W,
                      (int)k, (int)stride, (int)pad, Ho, Wo);
@@ -298,9 +338,9 @@ torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
   const int N = dy.size(0), C = dy.size(1);
   const int HoWo = dy.size(2) * dy.size(3);
   auto dx = torch::zeros({N, C, H, W}, dy.options());
-  const long total = dy.numel();
-  const int blocks = (int)std::min<long>(8192, (total + 255) / 256);
-  hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(blocks), dim3(256), 0,
+  const int chunks = std::max(1, std::min((HoWo + 255) / 256,
+                                          4096 / (N * C) + 1));
+  hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(N * C, chunks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
                      idx.data_ptr<long>(), dx.data_ptr<float>(), N, C,
                      (int)(H * W), HoWo);

@@ -284,3 +284,33 @@ def test_gemm_act_fused_bwd_gpu(act):
     dwb = ops.hip.gemm_wgrad_bias_act(x, dy, y, act)
     torch.testing.assert_close(dx, dx_ref, atol=1e-4, rtol=1e-4)
     torch.testing.assert_close(dwb, dwb_ref, atol=1e-4, rtol=1e-4)
+
+
+@pytest.mark.parametrize("conf", [(2, 3, 33, 33, 8, 7, 2, 3),
+                                  (2, 8, 14, 14, 16, 3, 2, 1),
+                                  (2, 4, 28, 28, 8, 5, 1, 2)])
+def test_conv2d_autograd_strided_gpu(conf):
+    """Full Conv2dAct autograd (zero-stuffed strided dgrad path) vs the
+    plain torch fp32 reference."""
+    N, Cin, H, W, Cout, K, stride, pad = conf
+    torch.manual_seed(13)
+    x = torch.randn(N, Cin, H, W)
+    w = torch.randn(Cout, Cin, K, K) * 0.1
+    b = torch.randn(Cout)
+    proj = None
+
+    def run(device):
+        xi = x.clone().to(device).requires_grad_()
+        wi = w.clone().to(device).requires_grad_()
+        bi = b.clone().to(device).requires_grad_()
+        y = F.conv2d_act(xi, wi, bi, stride, pad, "relu")
+        nonlocal proj
+        if proj is None:
+            proj = torch.randn_like(y.detach().cpu())
+        (y * proj.to(device)).sum().backward()
+        return y.detach().cpu(), xi.grad.cpu(), wi.grad.cpu(), bi.grad.cpu()
+
+    outs_g = run(DEV)
+    outs_c = run("cpu")
+    for g, c in zip(outs_g, outs_c):
+        torch.testing.assert_close(g, c, atol=2e-3, rtol=2e-3)
