@@ -189,6 +189,44 @@ __global__ void maxpool_fwd_kernel(const float* __restrict__ x,
   }
 }
 
+// Flat-index variant for narrow planes (Wo < wavefront): the div/mod per
+// element costs less than idling most lanes of a row-sweep block.
+__global__ void maxpool_fwd_flat_kernel(const float* __restrict__ x,
+                                        float* __restrict__ y,
+                                        long* __restrict__ idx, int N, int C,
+                                        int H, int W, int K, int stride,
+                                        int pad, int Ho, int Wo) {
+  const long total = (long)N * C * Ho * Wo;
+  const long gstride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gstride) {
+    const int wo = i % Wo;
+    const long pc = i / Wo;         // (n*C+c)*Ho + ho
+    const int ho = pc % Ho;
+    const long plane = pc / Ho;     // n*C + c
+    const float* xp = x + plane * H * W;
+    float best = -INFINITY;
+    long best_hw = 0;
+    for (int kh = 0; kh < K; ++kh) {
+      const int h = ho * stride - pad + kh;
+      if (h < 0) continue;
+      if (h >= H) break;
+      for (int kw = 0; kw < K; ++kw) {
+        const int ww = wo * stride - pad + kw;
+        if (ww < 0) continue;
+        if (ww >= W) break;
+        const float v = xp[(long)h * W + ww];
+        if (v > best) {
+          best = v;
+          best_hw = (long)h * W + ww;
+        }
+      }
+    }
+    y[i] = best;
+    idx[i] = best_hw;
+  }
+}
+
 __global__ void maxpool_bwd_kernel(const float* __restrict__ dy,
                                    const long* __restrict__ idx,
                                    float* __restrict__ dx, int N, int C,
@@ -323,12 +361,21 @@ std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
   auto y = torch::empty({N, C, Ho, Wo}, x.options());
   auto idx = torch::empty({N, C, Ho, Wo},
                           x.options().dtype(torch::kInt64));
-  const int ho_chunks = std::max(1, std::min(Ho, 2048 / (N * C) + 1));
-  hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(N * C, ho_chunks),
-                     dim3(std::min(256, std::max(64, Wo))), 0,
-                     at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
-                     y.data_ptr<float>(), idx.data_ptr<long>(), N, C, H, W,
-                     (int)k, (int)stride, (int)pad, Ho, Wo);
+  if (Wo >= 64) {  // wide planes: div-free row sweeps
+    const int ho_chunks = std::max(1, std::min(Ho, 2048 / (N * C) + 1));
+    hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(N * C, ho_chunks),
+                       dim3(std::min(256, Wo)), 0,
+                       at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
+                       y.data_ptr<float>(), idx.data_ptr<long>(), N, C, H,
+                       W, (int)k, (int)stride, (int)pad, Ho, Wo);
+  } else {  // narrow planes: flat sweep keeps all lanes busy
+    const long total = y.numel();
+    const int blocks = (int)std::min<long>(8192, (total + 255) / 256);
+    hipLaunchKernelGGL(maxpool_fwd_flat_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
+                       y.data_ptr<float>(), idx.data_ptr<long>(), N, C, H,
+                       W, (int)k, (int)stride, (int)pad, Ho, Wo);
+  }
   return {y, idx};
 }
 
