@@ -4,8 +4,9 @@ MNISTCNN mirrors reference examples/cnn_example.py:13-21 (5x5x32 conv ->
 2x2 pool -> 3x3x64 conv -> 2x2 pool -> flatten -> dense 10).
 
 ResNet18Shape is the BASELINE.json config-5 model: a ResNet-18-shaped CNN
-for synthetic 3x224x224 inputs (large-batch PS stress).  Convs/GEMMs run on
-the CDNA4 HIP kernels; batch-norm uses torch-ROCm natives.
+for synthetic 3x224x224 inputs (large-batch PS stress).  Convs/GEMMs and batch-norm
+all run on the CDNA4 HIP kernels (snn.BatchNorm2d fuses the trailing
+ReLU where one follows).
 """
 from __future__ import annotations
 
@@ -49,20 +50,20 @@ class _BasicBlock(nn.Module):
         super().__init__()
         self.conv1 = snn.Conv2d(cin, cout, 3, stride=stride,
                                 padding=1, activation="none", bias=False)
-        self.bn1 = nn.BatchNorm2d(cout)
+        self.bn1 = snn.BatchNorm2d(cout, fuse_relu=True)
         self.conv2 = snn.Conv2d(cout, cout, 3, stride=1, padding=1,
                                 activation="none", bias=False)
-        self.bn2 = nn.BatchNorm2d(cout)
+        self.bn2 = snn.BatchNorm2d(cout)
         if stride != 1 or cin != cout:
             self.down_conv = snn.Conv2d(cin, cout, 1, stride=stride,
                                         padding=0, activation="none", bias=False)
-            self.down_bn = nn.BatchNorm2d(cout)
+            self.down_bn = snn.BatchNorm2d(cout)
         else:
             self.down_conv = None
 
     def forward(self, x):
         idt = x
-        y = torch.relu(self.bn1(self.conv1(x)))
+        y = self.bn1(self.conv1(x))  # fused bn+relu
         y = self.bn2(self.conv2(y))
         if self.down_conv is not None:
             idt = self.down_bn(self.down_conv(x))
@@ -76,7 +77,7 @@ class ResNet18Shape(nn.Module):
         super().__init__()
         self.stem_conv = snn.Conv2d(3, 64, 7, stride=2, padding=3,
                                     activation="none", bias=False)
-        self.stem_bn = nn.BatchNorm2d(64)
+        self.stem_bn = snn.BatchNorm2d(64, fuse_relu=True)
         self.stem_pool = snn.MaxPool2d(3, 2, padding=1)  # 112 -> 56
         stages = []
         cin = 64
@@ -91,7 +92,7 @@ class ResNet18Shape(nn.Module):
     def forward(self, x):
         if x.dim() == 2:
             x = x.reshape(-1, 3, 224, 224)
-        x = torch.relu(self.stem_bn(self.stem_conv(x)))
+        x = self.stem_bn(self.stem_conv(x))  # fused bn+relu
         x = self.stem_pool(x)
         for block in self.stages:
             x = block(x)

@@ -81,6 +81,31 @@ class Flatten(nn.Module):
         return x.reshape(x.shape[0], -1)
 
 
+class BatchNorm2d(nn.Module):
+    """BatchNorm2d on the custom CDNA4 kernels (training mode), with an
+    optional fused trailing ReLU (``fuse_relu=True`` replaces a separate
+    bn -> relu pair).  Eval/CPU fall back to torch batch_norm."""
+
+    def __init__(self, num_features: int, eps: float = 1e-5,
+                 momentum: float = 0.1, fuse_relu: bool = False):
+        super().__init__()
+        self.num_features = num_features
+        self.eps, self.momentum, self.fuse_relu = eps, momentum, fuse_relu
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+
+    def forward(self, x):
+        return F.batchnorm2d_act(x, self.weight, self.bias,
+                                 self.running_mean, self.running_var,
+                                 self.momentum, self.eps, self.fuse_relu,
+                                 self.training)
+
+    def extra_repr(self):
+        return f"{self.num_features}, fuse_relu={self.fuse_relu}"
+
+
 class Dropout(nn.Module):
     """Inverted dropout; inert in eval mode (the inference path — parity
     with the reference's keep-prob feed at prediction, ml_util.py:70-71)."""

@@ -77,6 +77,17 @@ std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
 torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
                             int64_t W);
 
+// norm.hip — training-mode BatchNorm2d (+ fused ReLU)
+std::vector<torch::Tensor> bn2d_fwd(torch::Tensor x, torch::Tensor gamma,
+                                    torch::Tensor beta,
+                                    torch::Tensor run_mean,
+                                    torch::Tensor run_var, double momentum,
+                                    double eps, bool relu);
+std::vector<torch::Tensor> bn2d_bwd(torch::Tensor x, torch::Tensor dy,
+                                    torch::Tensor y, torch::Tensor gamma,
+                                    torch::Tensor save_mean,
+                                    torch::Tensor save_invstd, bool relu);
+
 // dist_session.hip — native RCCL session loops (GIL-free)
 py::bytes nccl_unique_id();
 int64_t nccl_comm_init(py::bytes id_bytes, int64_t nranks, int64_t rank);
@@ -147,6 +158,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "zero-stuff dy by stride (strided dgrad as stride-1 conv)");
   m.def("maxpool2d_fwd", &maxpool2d_fwd, "maxpool fwd with argmax indices");
   m.def("maxpool2d_bwd", &maxpool2d_bwd, "maxpool bwd scatter");
+  m.def("bn2d_fwd", &bn2d_fwd,
+        "training BatchNorm2d fwd (+ fused ReLU); returns y, mean, invstd");
+  m.def("bn2d_bwd", &bn2d_bwd,
+        "BatchNorm2d bwd (+ fused ReLU bwd); returns dx, dgamma, dbeta");
   m.def("nccl_unique_id", &nccl_unique_id, "RCCL unique id for a pair comm");
   m.def("nccl_comm_init", &nccl_comm_init,
         "init this rank's side of a raw RCCL communicator");

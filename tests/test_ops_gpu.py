@@ -314,3 +314,33 @@ def test_conv2d_autograd_strided_gpu(conf):
     outs_c = run("cpu")
     for g, c in zip(outs_g, outs_c):
         torch.testing.assert_close(g, c, atol=2e-3, rtol=2e-3)
+
+
+@pytest.mark.parametrize("relu", [False, True])
+def test_batchnorm2d_gpu(relu):
+    """Custom BN fwd/bwd (+ fused ReLU) vs the torch fp32 reference."""
+    torch.manual_seed(14)
+    N, C, H, W = 8, 16, 14, 14
+    x = torch.randn(N, C, H, W)
+    g = torch.rand(C) + 0.5
+    b = torch.randn(C)
+    proj = torch.randn(N, C, H, W)
+
+    def run(device, hip):
+        from sparkflow_amd.ops.functional import batchnorm2d_act
+
+        xi = x.clone().to(device).requires_grad_()
+        gi = g.clone().to(device).requires_grad_()
+        bi = b.clone().to(device).requires_grad_()
+        rm = torch.zeros(C, device=device)
+        rv = torch.ones(C, device=device)
+        y = batchnorm2d_act(xi, gi, bi, rm, rv, 0.1, 1e-5, relu,
+                            training=True)
+        (y * proj.to(device)).sum().backward()
+        return (y.detach().cpu(), xi.grad.cpu(), gi.grad.cpu(),
+                bi.grad.cpu(), rm.cpu(), rv.cpu())
+
+    outs_g = run(DEV, True)
+    outs_c = run("cpu", False)
+    for gg, cc in zip(outs_g, outs_c):
+        torch.testing.assert_close(gg, cc, atol=2e-4, rtol=2e-4)
