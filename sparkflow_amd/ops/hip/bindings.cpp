@@ -23,7 +23,9 @@ torch::Tensor argmax_rows(torch::Tensor x);
 void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
                   torch::Tensor ctr, torch::Tensor done,
                   c10::optional<torch::Tensor> y,
-                  c10::optional<torch::Tensor> yb);
+                  c10::optional<torch::Tensor> yb,
+                  c10::optional<torch::Tensor> wsrc,
+                  c10::optional<torch::Tensor> wdst);
 void pack_segments(std::vector<torch::Tensor> srcs, torch::Tensor flat);
 
 // losses.hip
@@ -118,9 +120,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("argmax_rows", &argmax_rows, "rowwise argmax head");
   m.def("gather_batch", &gather_batch,
         "fused minibatch gather + pool-cursor advance (hipGraph step "
-        "prologue)", py::arg("x"), py::arg("xb"), py::arg("idx_pool"),
+        "prologue); optionally also performs the weight-pull copy",
+        py::arg("x"), py::arg("xb"), py::arg("idx_pool"),
         py::arg("ctr"), py::arg("done"), py::arg("y") = py::none(),
-        py::arg("yb") = py::none());
+        py::arg("yb") = py::none(), py::arg("wsrc") = py::none(),
+        py::arg("wdst") = py::none());
   m.def("pack_segments", &pack_segments,
         "pack <=16 grad tensors into the flat push buffer in one launch");
   m.def("softmax_xent_fwd", &softmax_xent_fwd, "fused softmax+xent fwd",

@@ -60,10 +60,23 @@ __global__ void gather_batch_kernel(const float* __restrict__ x,
                                     const long* __restrict__ idx_pool,
                                     long* __restrict__ ctr,
                                     int* __restrict__ done, int mbs, long dx,
-                                    long dy_row, int cap) {
+                                    long dy_row, int cap,
+                                    const float* __restrict__ wsrc,
+                                    float* __restrict__ wdst, long wn) {
   const long c = *ctr;  // every thread's gather depends on this read
   const long* row_idx = idx_pool + c * mbs;
   const bool vec4 = (dx & 3) == 0;
+  if (wsrc != nullptr) {
+    // fused weight pull: the step's flat-param copy rides this launch
+    // (grid-stride float4 over the whole grid; wn % 4 handled by tail)
+    const long w4 = wn >> 2;
+    const long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    const long stride = (long)gridDim.x * blockDim.x;
+    const float4* s4 = reinterpret_cast<const float4*>(wsrc);
+    float4* d4 = reinterpret_cast<float4*>(wdst);
+    for (long i = tid; i < w4; i += stride) d4[i] = s4[i];
+    for (long i = (w4 << 2) + tid; i < wn; i += stride) wdst[i] = wsrc[i];
+  }
   for (int r = blockIdx.x; r < mbs; r += gridDim.x) {
     const long src = row_idx[r];
     const float* xs = x + src * dx;
@@ -98,7 +111,9 @@ __global__ void gather_batch_kernel(const float* __restrict__ x,
 void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
                   torch::Tensor ctr, torch::Tensor done,
                   c10::optional<torch::Tensor> y,
-                  c10::optional<torch::Tensor> yb) {
+                  c10::optional<torch::Tensor> yb,
+                  c10::optional<torch::Tensor> wsrc,
+                  c10::optional<torch::Tensor> wdst) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && xb.is_contiguous(),
               "gather_batch: x/xb must be contiguous on GPU");
   TORCH_CHECK(idx_pool.scalar_type() == torch::kInt64 &&
@@ -121,12 +136,22 @@ void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
     ybp = yb->data_ptr<float>();
     dy_row = yb->numel() / mbs;
   }
+  const float* wsp = nullptr;
+  float* wdp = nullptr;
+  long wn = 0;
+  if (wsrc.has_value() && wdst.has_value()) {
+    TORCH_CHECK(wsrc->numel() == wdst->numel(), "weight pull size mismatch");
+    wsp = wsrc->data_ptr<float>();
+    wdp = wdst->data_ptr<float>();
+    wn = wsrc->numel();
+  }
   const int blocks = std::min(mbs, 64);  // few fat blocks: cheap ticket
   hipLaunchKernelGGL(gather_batch_kernel, dim3(blocks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
                      xb.data_ptr<float>(), yp, ybp,
                      idx_pool.data_ptr<long>(), ctr.data_ptr<long>(),
-                     done.data_ptr<int>(), mbs, dx, dy_row, cap);
+                     done.data_ptr<int>(), mbs, dx, dy_row, cap, wsp, wdp,
+                     wn);
 }
 
 // Pack up to 16 parameter-grad tensors into the flat push buffer in ONE

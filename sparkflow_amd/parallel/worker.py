@@ -350,13 +350,18 @@ class Worker:
         gather_inside = sub and (fused_gather or full_step)
 
         def body():
-            if full_step:
+            if full_step and not fused_gather:
                 self.flat_params.copy_(ps.weights)  # the weight pull
             if gather_inside:
                 if fused_gather:
+                    # full-step mode: the weight pull rides the gather
+                    # launch (one kernel stages the whole step's inputs)
                     _ops.hip.gather_batch(self.x, xb, idx_pool, ctr, done,
                                           self.y if yb is not None else None,
-                                          yb)
+                                          yb,
+                                          ps.weights if full_step else None,
+                                          self.flat_params if full_step
+                                          else None)
                 else:
                     row = idx_pool.index_select(0, ctr).view(-1)
                     torch.index_select(self.x, 0, row, out=xb)
