@@ -17,9 +17,7 @@
 
 #define BM 64
 #define BN 64
-#define BK 32     // staged as BK/16 sub-blocks of the 16-deep pattern
-#define BKSUB 16
-#define NSUB (BK / BKSUB)
+#define BK 16
 
 // Logical C[M,N] = act(A'[M,K] @ B'[K,N] + bias).
 // TA=0: A' = A_phys[M,K] (lda=K);  TA=1: A' = A_phys[K,M]^T (lda=M).
@@ -65,22 +63,22 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
   // Per-thread load/store split so the NEXT chunk's global loads issue
   // during the CURRENT chunk's MFMAs (register prefetch — the K loop is
   // load-latency-bound at these sizes, cdna_hip_programming.md "overlap
-  // global with MFMA").  Each chunk is NSUB repetitions of the 16-deep
-  // staging pattern (one float4 of A and one of B per thread each).
+  // global with MFMA").  Each thread owns one float4 of A and one of B
+  // per chunk.
   const int a_row = t >> 2, a_c4 = (t & 3) * 4;       // TA=0 path
   const int a_kk = t >> 4, a_m4 = (t & 15) * 4;       // TA=1 path
   const int b_kk = t >> 4, b_n4 = (t & 15) * 4;       // TB=0 path
   const int b_row = t >> 2, b_c4 = (t & 3) * 4;       // TB=1 path
 
-  auto load_a16 = [&](int k0s) -> float4 {  // one 16-deep sub-block
+  auto load_a = [&](int k0) -> float4 {
     float4 r;
     if (!TA) {
       const int gm = m0 + a_row;
-      if (gm < M && k0s + BKSUB <= K) {
-        r = *reinterpret_cast<const float4*>(&A[(long)gm * lda + k0s + a_c4]);
+      if (gm < M && k0 + BK <= K) {
+        r = *reinterpret_cast<const float4*>(&A[(long)gm * lda + k0 + a_c4]);
         if (DPRE_A) {
           const float4 y4 = *reinterpret_cast<const float4*>(
-              &preY[(long)gm * lda + k0s + a_c4]);
+              &preY[(long)gm * lda + k0 + a_c4]);
           float* rr = reinterpret_cast<float*>(&r);
           const float* yy = reinterpret_cast<const float*>(&y4);
 #pragma unroll
@@ -91,7 +89,7 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
         float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          const int gk = k0s + a_c4 + i;
+          const int gk = k0 + a_c4 + i;
           const bool ok = gm < M && gk < K;
           float v = ok ? A[(long)gm * lda + gk] : 0.0f;
           if (DPRE_A && ok)
@@ -100,7 +98,7 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
         }
       }
     } else {
-      const int gk = k0s + a_kk;
+      const int gk = k0 + a_kk;
       // BIASROW: logical row M-1 is a virtual all-ones row (bias grad
       // rides the wgrad GEMM); the fast path must not read it.
       const int m_phys = BIASROW ? M - 1 : M;
@@ -122,10 +120,10 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
     return r;
   };
 
-  auto load_b16 = [&](int k0s) -> float4 {
+  auto load_b = [&](int k0) -> float4 {
     float4 r;
     if (!TB) {
-      const int gk = k0s + b_kk;
+      const int gk = k0 + b_kk;
       if (gk < K && n0 + BN <= N) {
         r = *reinterpret_cast<const float4*>(&B[(long)gk * ldb + n0 + b_n4]);
         if (DPRE_B) {
@@ -151,13 +149,13 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
       }
     } else {
       const int gn = n0 + b_row;
-      if (gn < N && k0s + BKSUB <= K) {
-        r = *reinterpret_cast<const float4*>(&B[(long)gn * ldb + k0s + b_c4]);
+      if (gn < N && k0 + BK <= K) {
+        r = *reinterpret_cast<const float4*>(&B[(long)gn * ldb + k0 + b_c4]);
       } else {
         float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          const int gk = k0s + b_c4 + i;
+          const int gk = k0 + b_c4 + i;
           rr[i] = (gn < N && gk < K) ? B[(long)gn * ldb + gk] : 0.0f;
         }
       }
@@ -165,49 +163,38 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
     return r;
   };
 
-  auto load_chunk = [&](int k0, float4* ra, float4* rb) {
+  auto store_lds = [&](float4 ra, float4 rb) {
+    const float* pa = reinterpret_cast<const float*>(&ra);
+    const float* pb = reinterpret_cast<const float*>(&rb);
+    if (!TA) {
 #pragma unroll
-    for (int s = 0; s < NSUB; ++s) {
-      ra[s] = load_a16(k0 + s * BKSUB);
-      rb[s] = load_b16(k0 + s * BKSUB);
+      for (int i = 0; i < 4; ++i) As[a_row][a_c4 + i] = pa[i];
+    } else {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) As[a_m4 + i][a_kk] = pa[i];
     }
-  };
-
-  auto store_lds = [&](const float4* ra, const float4* rb) {
+    if (!TB) {
 #pragma unroll
-    for (int s = 0; s < NSUB; ++s) {
-      const float* pa = reinterpret_cast<const float*>(&ra[s]);
-      const float* pb = reinterpret_cast<const float*>(&rb[s]);
-      const int ks = s * BKSUB;
-      if (!TA) {
+      for (int i = 0; i < 4; ++i) Bs[b_kk][b_n4 + i] = pb[i];
+    } else {
 #pragma unroll
-        for (int i = 0; i < 4; ++i) As[a_row][ks + a_c4 + i] = pa[i];
-      } else {
-#pragma unroll
-        for (int i = 0; i < 4; ++i) As[a_m4 + i][ks + a_kk] = pa[i];
-      }
-      if (!TB) {
-#pragma unroll
-        for (int i = 0; i < 4; ++i) Bs[ks + b_kk][b_n4 + i] = pb[i];
-      } else {
-#pragma unroll
-        for (int i = 0; i < 4; ++i) Bs[ks + b_c4 + i][b_row] = pb[i];
-      }
+      for (int i = 0; i < 4; ++i) Bs[b_c4 + i][b_row] = pb[i];
     }
   };
 
   const int fr = lane & 15;   // fragment row/col position
   const int fk = lane >> 4;   // k sub-index 0..3
 
-  float4 ra[NSUB], rb[NSUB];
-  load_chunk(k_begin, ra, rb);
+  float4 ra = load_a(k_begin);
+  float4 rb = load_b(k_begin);
   for (int k0 = k_begin; k0 < k_end; k0 += BK) {
     store_lds(ra, rb);
     __syncthreads();
     if (k0 + BK < k_end) {  // prefetch next chunk over this chunk's MFMAs
-      load_chunk(k0 + BK, ra, rb);
+      ra = load_a(k0 + BK);
+      rb = load_b(k0 + BK);
     }
-    // ---- 16x16x4 MFMA over the BK-deep chunk ----
+    // ---- 16x16x4 MFMA over the 16-deep chunk ----
 #pragma unroll
     for (int kk = 0; kk < BK / 4; ++kk) {
       const float a0 = As[wm + fr][kk * 4 + fk];
