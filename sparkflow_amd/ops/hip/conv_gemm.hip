@@ -19,13 +19,7 @@
 
 #define BM 64
 #define BN 64
-// Conv K loops run 18-36+ iterations (K = Cin*Kh*Kw or N*Ho*Wo), so a
-// 32-deep chunk staged as two 16-deep sub-blocks halves the iteration
-// and __syncthreads count (unlike the tiny dense GEMMs, where a BK=32
-// experiment measured slower).
-#define BK 32
-#define BKSUB 16
-#define NSUB (BK / BKSUB)
+#define BK 16
 
 struct ConvShape {
   int N, Cin, H, W, Cout, Kh, Kw, Ho, Wo, stride, pad;
@@ -69,11 +63,10 @@ __global__ __launch_bounds__(256) void conv_gemm_fwd_kernel(
   f32x4 acc[2][2] = {};
 
   // Register-prefetched staging (next chunk's gathers overlap this
-  // chunk's MFMAs — same pattern as gemm.hip), BK staged as NSUB 16-deep
-  // sub-blocks.  The output pixel (n, ho, wo) is FIXED per thread, so
-  // its decode happens once; the filter coordinate advances
-  // incrementally (+16/+BK with carry), leaving no div/mod in the K
-  // loop.
+  // chunk's MFMAs — same pattern as gemm.hip).  The output pixel (n, ho,
+  // wo) is FIXED per thread, so its decode happens once; the filter
+  // coordinate advances incrementally (+BK with carry), leaving no
+  // div/mod in the K loop.
   const int st_row = t >> 2, st_c4 = (t & 3) * 4;
 
   const int gm = m0 + st_row;
@@ -83,79 +76,69 @@ __global__ __launch_bounds__(256) void conv_gemm_fwd_kernel(
   const long a_nbase = (long)(m_dec / (s.Wo * s.Ho)) * s.Cin * s.H * s.W;
   const int h0 = a_ho * s.stride - s.pad;  // + kh
   const int w0 = a_wo * s.stride - s.pad;  // + kw
-  struct FC { int kw, kh, ci; };
-  FC a0 = {st_c4 % s.Kw, (st_c4 / s.Kw) % s.Kh, st_c4 / (s.Kw * s.Kh)};
+  int a_kw = st_c4 % s.Kw;
+  int a_kh = (st_c4 / s.Kw) % s.Kh;
+  int a_ci = st_c4 / (s.Kw * s.Kh);
 
-  auto fc_advance = [&](FC& c, int step) {  // += step, carry, no div
-    c.kw += step;
-    while (c.kw >= s.Kw) {
-      c.kw -= s.Kw;
-      if (++c.kh >= s.Kh) { c.kh -= s.Kh; ++c.ci; }
-    }
-  };
-  auto load_a_at = [&](int k0s, FC c) -> float4 {
+  auto load_a = [&](int k0) -> float4 {
     float4 r;
     float* rr = reinterpret_cast<float*>(&r);
+    int kw = a_kw, kh = a_kh, ci = a_ci;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const int h = h0 + c.kh, w = w0 + c.kw;
+      const int h = h0 + kh, w = w0 + kw;
       float v = 0.0f;
-      if (gm < M && k0s + st_c4 + i < K && h >= 0 && h < s.H && w >= 0 &&
+      if (gm < M && k0 + st_c4 + i < K && h >= 0 && h < s.H && w >= 0 &&
           w < s.W) {
-        v = x[a_nbase + ((long)c.ci * s.H + h) * s.W + w];
+        v = x[a_nbase + ((long)ci * s.H + h) * s.W + w];
       }
       rr[i] = v;
-      if (++c.kw == s.Kw) { c.kw = 0; if (++c.kh == s.Kh) { c.kh = 0; ++c.ci; } }
+      if (++kw == s.Kw) { kw = 0; if (++kh == s.Kh) { kh = 0; ++ci; } }
     }
     return r;
   };
-  auto load_b_at = [&](int k0s) -> float4 {
+  auto advance_a = [&]() {  // filter coordinate += BK, carry, no div
+    a_kw += BK;
+    while (a_kw >= s.Kw) {
+      a_kw -= s.Kw;
+      if (++a_kh >= s.Kh) { a_kh -= s.Kh; ++a_ci; }
+    }
+  };
+  auto load_b = [&](int k0) -> float4 {
     float4 r;
     const int gn = n0 + st_row;
-    if (gn < Nc && k0s + BKSUB <= K) {
-      r = *reinterpret_cast<const float4*>(&wgt[(long)gn * K + k0s + st_c4]);
+    if (gn < Nc && k0 + BK <= K) {
+      r = *reinterpret_cast<const float4*>(&wgt[(long)gn * K + k0 + st_c4]);
     } else {
       float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        const int gk = k0s + st_c4 + i;
+        const int gk = k0 + st_c4 + i;
         rr[i] = (gn < Nc && gk < K) ? wgt[(long)gn * K + gk] : 0.0f;
       }
     }
     return r;
   };
-  auto load_chunk = [&](int k0, float4* ra, float4* rb) {
-    FC c = a0;
-#pragma unroll
-    for (int su = 0; su < NSUB; ++su) {
-      ra[su] = load_a_at(k0 + su * BKSUB, c);
-      if (su + 1 < NSUB) fc_advance(c, BKSUB);
-      rb[su] = load_b_at(k0 + su * BKSUB);
-    }
-  };
 
   const int fr = lane & 15;
   const int fk = lane >> 4;
 
-  float4 ra[NSUB], rb[NSUB];
-  load_chunk(0, ra, rb);
+  float4 ra = load_a(0);
+  float4 rb = load_b(0);
   for (int k0 = 0; k0 < K; k0 += BK) {
     {
+      const float* pa = reinterpret_cast<const float*>(&ra);
+      const float* pb = reinterpret_cast<const float*>(&rb);
 #pragma unroll
-      for (int su = 0; su < NSUB; ++su) {
-        const float* pa = reinterpret_cast<const float*>(&ra[su]);
-        const float* pb = reinterpret_cast<const float*>(&rb[su]);
-        const int ks = su * BKSUB;
+      for (int i = 0; i < 4; ++i) As[st_row][st_c4 + i] = pa[i];
 #pragma unroll
-        for (int i = 0; i < 4; ++i) As[st_row][ks + st_c4 + i] = pa[i];
-#pragma unroll
-        for (int i = 0; i < 4; ++i) Bs[ks + st_c4 + i][st_row] = pb[i];
-      }
+      for (int i = 0; i < 4; ++i) Bs[st_c4 + i][st_row] = pb[i];
     }
     __syncthreads();
     if (k0 + BK < K) {
-      fc_advance(a0, BK);
-      load_chunk(k0 + BK, ra, rb);
+      advance_a();
+      ra = load_a(k0 + BK);
+      rb = load_b(k0 + BK);
     }
 #pragma unroll
     for (int kk = 0; kk < BK / 4; ++kk) {
@@ -256,41 +239,29 @@ __global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
   }
   const long n_stride = (long)s.Cin * s.H * s.W;
 
-  struct PixA { int n, hw; };
-  struct PixB { int n, ho, wo; };
-  auto a_step = [&](PixA& p, int step) {
-    p.hw += step;
-    while (p.hw >= HoWo) { p.hw -= HoWo; ++p.n; }
-  };
-  auto b_step = [&](PixB& p, int step) {
-    p.wo += step;
-    while (p.wo >= s.Wo) { p.wo -= s.Wo; ++p.ho; }
-    while (p.ho >= s.Ho) { p.ho -= s.Ho; ++p.n; }
-  };
-
-  auto load_a_at = [&](int gp0, PixA p) -> float4 {
+  auto load_a = [&](int gp0) -> float4 {
     float4 r;
     float* rr = reinterpret_cast<float*>(&r);
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      int hw = p.hw + i, n = p.n;
+      int hw = a_hw + i, n = a_n;
       while (hw >= HoWo) { hw -= HoWo; ++n; }  // HoWo may be tiny
       rr[i] = (gm < M && gp0 + i < K)
                   ? dy[((long)n * M + gm) * HoWo + hw] : 0.0f;
     }
     return r;
   };
-  auto load_b_at = [&](int gp0, PixB p) -> float4 {
+  auto load_b = [&](int gp0) -> float4 {
     float4 r;
     float* rr = reinterpret_cast<float*>(&r);
     const bool p_ok = gp0 < K;
-    const long nbase = (long)p.n * n_stride;
+    const long nbase = (long)b_n * n_stride;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       float v = 0.0f;
       if (p_ok && gk_ok[i]) {
-        const int h = p.ho * s.stride - s.pad + kh_i[i];
-        const int w = p.wo * s.stride - s.pad + kw_i[i];
+        const int h = b_ho * s.stride - s.pad + kh_i[i];
+        const int w = b_wo * s.stride - s.pad + kw_i[i];
         if (h >= 0 && h < s.H && w >= 0 && w < s.W)
           v = x[nbase + coff_i[i] + (long)h * s.W + w];
       }
@@ -298,46 +269,33 @@ __global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
     }
     return r;
   };
-
-  PixA pa0 = {a_n, a_hw};
-  PixB pb0 = {b_n, b_ho, b_wo};
-  auto load_chunk = [&](int k0, float4* ra, float4* rb) {
-    PixA pa = pa0;
-    PixB pb = pb0;
-#pragma unroll
-    for (int su = 0; su < NSUB; ++su) {
-      ra[su] = load_a_at(k0 + su * BKSUB + a_c4, pa);
-      rb[su] = load_b_at(k0 + su * BKSUB + b_kk, pb);
-      if (su + 1 < NSUB) {
-        a_step(pa, BKSUB);
-        b_step(pb, BKSUB);
-      }
-    }
+  auto advance = [&]() {  // both cursors move +BK pixels, carry, no div
+    a_hw += BK;
+    while (a_hw >= HoWo) { a_hw -= HoWo; ++a_n; }
+    b_wo += BK;
+    while (b_wo >= s.Wo) { b_wo -= s.Wo; ++b_ho; }
+    while (b_ho >= s.Ho) { b_ho -= s.Ho; ++b_n; }
   };
 
   const int fr = lane & 15;
   const int fk = lane >> 4;
 
-  float4 ra[NSUB], rb[NSUB];
-  load_chunk(k_begin, ra, rb);
+  float4 ra = load_a(k_begin + a_c4);
+  float4 rb = load_b(k_begin + b_kk);
   for (int k0 = k_begin; k0 < k_end; k0 += BK) {
     {
+      const float* pa = reinterpret_cast<const float*>(&ra);
+      const float* pb = reinterpret_cast<const float*>(&rb);
 #pragma unroll
-      for (int su = 0; su < NSUB; ++su) {
-        const float* pa = reinterpret_cast<const float*>(&ra[su]);
-        const float* pb = reinterpret_cast<const float*>(&rb[su]);
-        const int ks = su * BKSUB;
+      for (int i = 0; i < 4; ++i) As[a_row][a_c4 + i] = pa[i];
 #pragma unroll
-        for (int i = 0; i < 4; ++i) As[a_row][ks + a_c4 + i] = pa[i];
-#pragma unroll
-        for (int i = 0; i < 4; ++i) Bs[ks + b_kk][b_n4 + i] = pb[i];
-      }
+      for (int i = 0; i < 4; ++i) Bs[b_kk][b_n4 + i] = pb[i];
     }
     __syncthreads();
     if (k0 + BK < k_end) {
-      a_step(pa0, BK);
-      b_step(pb0, BK);
-      load_chunk(k0 + BK, ra, rb);
+      advance();
+      ra = load_a(k0 + BK + a_c4);
+      rb = load_b(k0 + BK + b_kk);
     }
 #pragma unroll
     for (int kk = 0; kk < BK / 4; ++kk) {
