@@ -272,8 +272,21 @@ class DistPairs:
                     idbuf = torch.frombuffer(bytearray(raw),
                                              dtype=torch.uint8).clone()
                     dist.send(idbuf, dst=w, group=cg)
-                    self.native[(w, i)] = ops.hip.nccl_comm_init(
-                        bytes(idbuf.numpy().tobytes()), 2, 0)
+                    handle = None
+                    try:
+                        handle = ops.hip.nccl_comm_init(
+                            bytes(idbuf.numpy().tobytes()), 2, 0)
+                    except Exception as e:  # peer fails symmetrically
+                        print(f"[sparkflow_amd] raw comm init (0,{w},{i}) "
+                              f"failed ({e}); torch pair fallback")
+                    # confirm: both sides must agree the pair is usable
+                    ok = torch.tensor([1 if handle is not None else 0],
+                                      dtype=torch.int64)
+                    peer_ok = torch.zeros(1, dtype=torch.int64)
+                    dist.send(ok, dst=w, group=cg)
+                    dist.recv(peer_ok, src=w, group=cg)
+                    if handle is not None and int(peer_ok.item()):
+                        self.native[(w, i)] = handle
         else:
             for i in range(self.workers_per_gpu):
                 cg = self.ctrl[(self.rank, i)]
@@ -284,8 +297,21 @@ class DistPairs:
                     continue
                 idbuf = torch.zeros(128, dtype=torch.uint8)
                 dist.recv(idbuf, src=0, group=cg)
-                self.native[(self.rank, i)] = ops.hip.nccl_comm_init(
-                    bytes(idbuf.numpy().tobytes()), 2, 1)
+                handle = None
+                try:
+                    handle = ops.hip.nccl_comm_init(
+                        bytes(idbuf.numpy().tobytes()), 2, 1)
+                except Exception as e:
+                    print(f"[sparkflow_amd] raw comm init "
+                          f"({self.rank},{i}) failed ({e}); torch pair "
+                          "fallback")
+                ok = torch.tensor([1 if handle is not None else 0],
+                                  dtype=torch.int64)
+                peer_ok = torch.zeros(1, dtype=torch.int64)
+                dist.recv(peer_ok, src=0, group=cg)
+                dist.send(ok, dst=0, group=cg)
+                if handle is not None and int(peer_ok.item()):
+                    self.native[(self.rank, i)] = handle
 
 
 class DistTransport(WorkerTransport):
