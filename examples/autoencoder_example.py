@@ -1,5 +1,10 @@
 """Unsupervised autoencoder on Spark (port of reference
 examples/autoencoder_example.py — ``tfLabel=None``)."""
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
 from pyspark.ml.feature import VectorAssembler
 from pyspark.sql import SparkSession
 

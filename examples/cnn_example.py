@@ -1,5 +1,10 @@
 """MNIST CNN on Spark (port of reference examples/cnn_example.py) using
 the async-with-lock mode (BASELINE.json config 3)."""
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
 from pyspark.ml.feature import VectorAssembler
 from pyspark.sql import SparkSession
 

@@ -5,6 +5,11 @@ synthetic blobs against a parameter-server process over the raw-tensor
 TCP transport.  For multi-GPU RCCL runs use bench.py / the dist_engine
 (torchrun) path instead.
 """
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
 import numpy as np
 
 from sparkflow_amd.graph_utils import build_graph

@@ -4,6 +4,11 @@ Requires: pyspark, an mnist_train.csv with label in column 0.
 Each Spark executor should own one MI355X (set HIP_VISIBLE_DEVICES per
 executor); the driver hosts the parameter server.
 """
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
 from pyspark.ml.feature import OneHotEncoder, VectorAssembler
 from pyspark.ml.pipeline import Pipeline, PipelineModel
 from pyspark.sql import SparkSession
