@@ -123,34 +123,23 @@ __global__ __launch_bounds__(256) void conv_gemm_fwd_kernel(
   const int fr = lane & 15;
   const int fk = lane >> 4;
 
-  // Depth-2 software pipeline: TWO chunks' gathers in flight (the 18-72
-  // iteration conv K loops are gather-latency-bound; one-deep prefetch
-  // only hides latency behind a single chunk's MFMAs).
-  float4 ra[2], rb[2];
-  ra[0] = load_a(0);
-  rb[0] = load_b(0);
-  if (BK < K) {
-    advance_a();
-    ra[1] = load_a(BK);
-    rb[1] = load_b(BK);
-  }
-  int buf = 0;
+  float4 ra = load_a(0);
+  float4 rb = load_b(0);
   for (int k0 = 0; k0 < K; k0 += BK) {
     {
-      const float* pa = reinterpret_cast<const float*>(&ra[buf]);
-      const float* pb = reinterpret_cast<const float*>(&rb[buf]);
+      const float* pa = reinterpret_cast<const float*>(&ra);
+      const float* pb = reinterpret_cast<const float*>(&rb);
 #pragma unroll
       for (int i = 0; i < 4; ++i) As[st_row][st_c4 + i] = pa[i];
 #pragma unroll
       for (int i = 0; i < 4; ++i) Bs[st_c4 + i][st_row] = pb[i];
     }
     __syncthreads();
-    if (k0 + 2 * BK < K) {  // refill this buffer two chunks ahead
+    if (k0 + BK < K) {
       advance_a();
-      ra[buf] = load_a(k0 + 2 * BK);
-      rb[buf] = load_b(k0 + 2 * BK);
+      ra = load_a(k0 + BK);
+      rb = load_b(k0 + BK);
     }
-    buf ^= 1;
 #pragma unroll
     for (int kk = 0; kk < BK / 4; ++kk) {
       const float a0 = As[wm + fr][kk * 4 + fk];
@@ -291,32 +280,23 @@ __global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
   const int fr = lane & 15;
   const int fk = lane >> 4;
 
-  // Depth-2 software pipeline (see the forward kernel).
-  float4 ra[2], rb[2];
-  ra[0] = load_a(k_begin + a_c4);
-  rb[0] = load_b(k_begin + b_kk);
-  if (k_begin + BK < k_end) {
-    advance();
-    ra[1] = load_a(k_begin + BK + a_c4);
-    rb[1] = load_b(k_begin + BK + b_kk);
-  }
-  int buf = 0;
+  float4 ra = load_a(k_begin + a_c4);
+  float4 rb = load_b(k_begin + b_kk);
   for (int k0 = k_begin; k0 < k_end; k0 += BK) {
     {
-      const float* pa = reinterpret_cast<const float*>(&ra[buf]);
-      const float* pb = reinterpret_cast<const float*>(&rb[buf]);
+      const float* pa = reinterpret_cast<const float*>(&ra);
+      const float* pb = reinterpret_cast<const float*>(&rb);
 #pragma unroll
       for (int i = 0; i < 4; ++i) As[a_row][a_c4 + i] = pa[i];
 #pragma unroll
       for (int i = 0; i < 4; ++i) Bs[b_kk][b_n4 + i] = pb[i];
     }
     __syncthreads();
-    if (k0 + 2 * BK < k_end) {
+    if (k0 + BK < k_end) {
       advance();
-      ra[buf] = load_a(k0 + 2 * BK + a_c4);
-      rb[buf] = load_b(k0 + 2 * BK + b_kk);
+      ra = load_a(k0 + BK + a_c4);
+      rb = load_b(k0 + BK + b_kk);
     }
-    buf ^= 1;
 #pragma unroll
     for (int kk = 0; kk < BK / 4; ++kk) {
       const float a0 = As[wm + fr][kk * 4 + fk];
