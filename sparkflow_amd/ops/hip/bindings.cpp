@@ -4,8 +4,6 @@
 #include <vector>
 
 // gemm.hip
-int sf_parallel_hint();
-void sf_set_parallel_hint(int n);
 torch::Tensor gemm_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
                        int64_t act);
 torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w);
@@ -140,10 +138,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mse_fwd", &mse_fwd, "fused MSE loss", py::arg("pred"),
         py::arg("target"), py::arg("want_value") = true);
   m.def("mse_bwd", &mse_bwd, "MSE grad");
-  m.def("set_parallel_hint", &sf_set_parallel_hint,
-        "number of concurrent worker streams sharing the GPU (shrinks "
-        "split-K)");
-  m.def("parallel_hint", &sf_parallel_hint, "current concurrency hint");
   m.def("fused_sgd", &fused_sgd, "single-launch SGD apply");
   m.def("fused_momentum", &fused_momentum, "single-launch momentum apply");
   m.def("fused_adam", &fused_adam, "single-launch Adam apply");

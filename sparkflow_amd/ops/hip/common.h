@@ -55,11 +55,6 @@ __device__ __forceinline__ float wave_max(float v) {
 
 static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
 
-// How many concurrent worker streams share the GPU (set by the engine).
-// Split-K exists to fill idle CUs in the latency-bound single-stream
-// case; with several Hogwild workers racing, their kernels already
-// overlap, so the K-split (and its reduce launch + workspace traffic)
-// only adds overhead — the cap shrinks with the hint.
-int sf_parallel_hint();               // defined in gemm.hip
-void sf_set_parallel_hint(int n);
-int sf_splitk_env_max();              // SPARKFLOW_SPLITK_MAX override
+// SPARKFLOW_SPLITK_MAX env override for split-K experiments (unset = no
+// cap; structural limits at the call sites govern).  Defined in gemm.hip.
+int sf_splitk_env_max();

@@ -6,9 +6,10 @@
 // GEMM views (all fp32, NCHW):
 //   fwd:   C[M=N*Ho*Wo, N'=Cout]       = im2col(x) @ W^T          (+bias+act)
 //   wgrad: C[M=Cout, N'=Cin*Kh*Kw]     = dyT @ im2col(x)
-//   dgrad: falls back to conv.hip's direct kernel (the gather predicate
-//          with stride divisibility does not tile as cleanly; direct dgrad
-//          has the same MAC count as fwd).
+//   dgrad: runs as a stride-1 convolution of (zero-stuffed) dy with
+//          flipped weights on this same fwd pipeline (conv_dilate in
+//          conv.hip); conv.hip's direct dgrad kernel remains as the
+//          tested reference.
 //
 // The MFMA inner loop is the same 64x64-tile / 16-deep structure as
 // gemm.hip; only the A/B staging differs (indexed gathers).

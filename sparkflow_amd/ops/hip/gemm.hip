@@ -306,11 +306,6 @@ static inline void check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
 }
 
-static int g_parallel_hint = 1;
-
-int sf_parallel_hint() { return g_parallel_hint; }
-void sf_set_parallel_hint(int n) { g_parallel_hint = n < 1 ? 1 : n; }
-
 int sf_splitk_env_max() {
   static int cached = -1;
   if (cached < 0) {
