@@ -371,7 +371,8 @@ class DistTransport(WorkerTransport):
         return self.native_comm is not None
 
     def pull_into(self, out: torch.Tensor) -> None:
-        raw = self._use_raw()
+        # inside a session, the channel is whatever run_begin declared
+        raw = self._session_native if self._session > 0 else self._use_raw()
         if self._session <= 0:
             self._ctrl(OP_PULL, chan=1 if raw else 0)
         if raw:
@@ -382,7 +383,7 @@ class DistTransport(WorkerTransport):
             self.dist.recv(out, src=0, group=self.pg)
 
     def push(self, flat_grads: torch.Tensor) -> None:
-        raw = self._use_raw()
+        raw = self._session_native if self._session > 0 else self._use_raw()
         if self._session <= 0:
             self._ctrl(OP_PUSH, chan=1 if raw else 0)
         else:
