@@ -109,3 +109,47 @@ def test_argmax_rows():
     out = F.argmax_rows(y)
     assert out.dtype == torch.float32
     np.testing.assert_array_equal(out.numpy(), [1.0, 0.0])
+
+
+def test_grad_only_cpu_keeps_value():
+    """On CPU the loss value stays exact even inside grad_only() (the
+    value-skip is a GPU-kernel optimization only)."""
+    import torch
+
+    from sparkflow_amd.ops import functional as F
+
+    torch.manual_seed(3)
+    logits = torch.randn(32, 10, requires_grad=True)
+    labels = torch.randint(0, 10, (32,))
+    with F.grad_only():
+        loss = F.softmax_xent(logits, labels)
+    ref = torch.nn.functional.cross_entropy(logits.detach(), labels)
+    torch.testing.assert_close(loss.detach(), ref, atol=1e-5, rtol=1e-5)
+    loss.backward(gradient=torch.ones(()))
+    assert torch.isfinite(logits.grad).all()
+
+
+def test_batchnorm2d_module_cpu():
+    """snn.BatchNorm2d CPU fallback matches torch batch_norm + relu and
+    updates running stats."""
+    import torch
+
+    from sparkflow_amd import nn as snn
+
+    torch.manual_seed(4)
+    bn = snn.BatchNorm2d(6, fuse_relu=True)
+    x = torch.randn(4, 6, 5, 5)
+    y = bn(x)
+    ref_bn = torch.nn.BatchNorm2d(6)
+    ref = torch.relu(ref_bn(x))
+    torch.testing.assert_close(y, ref, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(bn.running_mean, ref_bn.running_mean,
+                               atol=1e-6, rtol=1e-5)
+    torch.testing.assert_close(bn.running_var, ref_bn.running_var,
+                               atol=1e-6, rtol=1e-5)
+    # eval mode uses running stats
+    bn.eval()
+    y2 = bn(x)
+    ref_bn.eval()
+    torch.testing.assert_close(y2, torch.relu(ref_bn(x)), atol=1e-5,
+                               rtol=1e-5)
