@@ -145,7 +145,9 @@ void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
     wdp = wdst->data_ptr<float>();
     wn = wsrc->numel();
   }
-  const int blocks = std::min(mbs, 64);  // few fat blocks: cheap ticket
+  // enough blocks to stream the weight copy + rows, few enough that the
+  // same-address ticket atomics stay ~1us
+  const int blocks = std::min(mbs, 128);
   hipLaunchKernelGGL(gather_batch_kernel, dim3(blocks), dim3(256), 0,
                      at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
                      xb.data_ptr<float>(), yp, ybp,
