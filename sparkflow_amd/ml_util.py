@@ -115,10 +115,22 @@ def _best_device() -> torch.device:
 
 
 def predict_batch(module: torch.nn.Module, x: np.ndarray,
-                  device: Optional[torch.device] = None) -> np.ndarray:
-    """One batched inference forward on the best device."""
+                  device: Optional[torch.device] = None,
+                  keep_dropout: bool = False) -> np.ndarray:
+    """One batched inference forward on the best device.
+
+    ``keep_dropout`` preserves the reference's ``to_keep_dropout``
+    semantics (ml_util.py:70-71: the dropout feed keeps its TRAINING
+    keep-prob at inference instead of 1.0): Dropout modules stay in
+    train mode while everything else runs eval."""
     device = device or _best_device()
     module = module.to(device).eval()
+    if keep_dropout:
+        from sparkflow_amd.nn.modules import Dropout
+
+        for m in module.modules():
+            if isinstance(m, (Dropout, torch.nn.Dropout)):
+                m.train()
     with torch.no_grad():
         out = module(torch.as_tensor(x, dtype=torch.float32, device=device))
     return out.detach().cpu().numpy()
@@ -131,9 +143,10 @@ def predict_func(rows, graph_json, prediction, graph_weights, inp,
     ml_util.predict_func, ml_util.py:54-83).
 
     Differences by design: the module is rebuilt once per partition and the
-    whole partition runs as ONE batched forward on the GPU; dropout is
-    handled by module eval() mode (the tf_dropout / to_keep_dropout params
-    are accepted for API compatibility).  Scalar outputs become floats,
+    whole partition runs as ONE batched forward on the GPU; dropout
+    follows the reference keep/drop semantics (``to_keep_dropout=True``
+    keeps Dropout modules active at inference, ml_util.py:70-71;
+    otherwise eval mode makes them inert).  Scalar outputs become floats,
     vector outputs become ``pyspark.ml.linalg.Vectors.dense`` — identical
     row-shape contract.
     """
@@ -149,7 +162,7 @@ def predict_func(rows, graph_json, prediction, graph_weights, inp,
     module = module_from_json(graph_json)
     torch_set_weights(module, convert_json_to_weights(graph_weights))
     x = np.asarray([np.asarray(row[inp], dtype=np.float32) for row in rows])
-    pred = predict_batch(module, x)
+    pred = predict_batch(module, x, keep_dropout=bool(to_keep_dropout))
 
     out_rows = []
     for i, row in enumerate(rows):

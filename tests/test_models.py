@@ -64,3 +64,30 @@ def test_resnet18_shape_small_batch():
     y = torch.randint(0, 10, (2,))
     out, _ = _fwd_bwd(m, x, y)
     assert out.shape == (2, 10)
+
+
+def test_keep_dropout_inference():
+    """to_keep_dropout=True keeps Dropout active at inference (reference
+    ml_util.py:70-71 keep/drop feed semantics)."""
+    import numpy as np
+    import torch
+
+    from sparkflow_amd import nn as snn
+    from sparkflow_amd.ml_util import predict_batch
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.drop = snn.Dropout(0.5)
+
+        def forward(self, x):
+            return self.drop(x)
+
+    m = M()
+    x = np.ones((64, 32), dtype=np.float32)
+    out_eval = predict_batch(m, x, device=torch.device("cpu"))
+    np.testing.assert_array_equal(out_eval, x)  # dropout inert
+    torch.manual_seed(0)
+    out_keep = predict_batch(m, x, device=torch.device("cpu"),
+                             keep_dropout=True)
+    assert (out_keep == 0).any() and (out_keep > 1.0).any()  # active
