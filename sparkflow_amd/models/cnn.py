@@ -46,18 +46,21 @@ class MNISTCNN(nn.Module):
 
 
 class _BasicBlock(nn.Module):
-    def __init__(self, cin, cout, stride):
+    def __init__(self, cin, cout, stride, layout="nhwc"):
         super().__init__()
-        self.conv1 = snn.Conv2d(cin, cout, 3, stride=stride,
-                                padding=1, activation="none", bias=False)
-        self.bn1 = snn.BatchNorm2d(cout, fuse_relu=True)
+        self.conv1 = snn.Conv2d(cin, cout, 3, stride=stride, padding=1,
+                                activation="none", bias=False,
+                                layout=layout)
+        self.bn1 = snn.BatchNorm2d(cout, fuse_relu=True, layout=layout)
         self.conv2 = snn.Conv2d(cout, cout, 3, stride=1, padding=1,
-                                activation="none", bias=False)
-        self.bn2 = snn.BatchNorm2d(cout)
+                                activation="none", bias=False,
+                                layout=layout)
+        self.bn2 = snn.BatchNorm2d(cout, layout=layout)
         if stride != 1 or cin != cout:
             self.down_conv = snn.Conv2d(cin, cout, 1, stride=stride,
-                                        padding=0, activation="none", bias=False)
-            self.down_bn = snn.BatchNorm2d(cout)
+                                        padding=0, activation="none",
+                                        bias=False, layout=layout)
+            self.down_bn = snn.BatchNorm2d(cout, layout=layout)
         else:
             self.down_conv = None
 
@@ -71,19 +74,25 @@ class _BasicBlock(nn.Module):
 
 
 class ResNet18Shape(nn.Module):
-    """ResNet-18 layer shapes for 3x224x224 synthetic inputs."""
+    """ResNet-18 layer shapes for 3x224x224 synthetic inputs.
 
-    def __init__(self, num_classes: int = 1000):
+    Internally channel-last (NHWC): one permute at the input, then every
+    conv/bn runs on the channel-contiguous kernel family."""
+
+    def __init__(self, num_classes: int = 1000, layout: str = "nhwc"):
         super().__init__()
+        self.layout = layout
         self.stem_conv = snn.Conv2d(3, 64, 7, stride=2, padding=3,
-                                    activation="none", bias=False)
-        self.stem_bn = snn.BatchNorm2d(64, fuse_relu=True)
-        self.stem_pool = snn.MaxPool2d(3, 2, padding=1)  # 112 -> 56
+                                    activation="none", bias=False,
+                                    layout=layout)
+        self.stem_bn = snn.BatchNorm2d(64, fuse_relu=True, layout=layout)
+        self.stem_pool = snn.MaxPool2d(3, 2, padding=1,
+                                       layout=layout)  # 112 -> 56
         stages = []
         cin = 64
         for cout, stride in [(64, 1), (64, 1), (128, 2), (128, 1),
                              (256, 2), (256, 1), (512, 2), (512, 1)]:
-            stages.append(_BasicBlock(cin, cout, stride))
+            stages.append(_BasicBlock(cin, cout, stride, layout=layout))
             cin = cout
         self.stages = nn.ModuleList(stages)
         self.fc = snn.Linear(512, num_classes)
@@ -92,11 +101,16 @@ class ResNet18Shape(nn.Module):
     def forward(self, x):
         if x.dim() == 2:
             x = x.reshape(-1, 3, 224, 224)
+        if self.layout == "nhwc":
+            x = x.permute(0, 2, 3, 1).contiguous()
+            sdims = (1, 2)
+        else:
+            sdims = (2, 3)
         x = self.stem_bn(self.stem_conv(x))  # fused bn+relu
         x = self.stem_pool(x)
         for block in self.stages:
             x = block(x)
-        x = x.mean(dim=(2, 3))
+        x = x.mean(dim=sdims)
         return self.fc(x)
 
     def compute_loss(self, out, target):

@@ -44,9 +44,14 @@ class Linear(nn.Module):
 
 
 class Conv2d(nn.Module):
+    """``layout="nchw"`` (default) keeps weights [Cout, Cin, Kh, Kw] and
+    NCHW tensors; ``layout="nhwc"`` keeps weights [Cout, Kh, Kw, Cin] and
+    channel-last tensors (the ResNet-config fast path — channel-
+    contiguous float4 im2col reads)."""
+
     def __init__(self, in_channels: int, out_channels: int, kernel_size: int,
                  stride: int = 1, padding="same", activation: str = "none",
-                 bias: bool = True):
+                 bias: bool = True, layout: str = "nchw"):
         super().__init__()
         k = kernel_size
         if padding == "same":
@@ -54,25 +59,37 @@ class Conv2d(nn.Module):
                 raise ValueError("padding='same' requires stride=1")
             padding = (k - 1) // 2
         self.stride, self.padding, self.activation = stride, int(padding), activation
+        self.layout = layout
         fan_in = in_channels * k * k
         fan_out = out_channels * k * k
-        self.weight = nn.Parameter(
-            _glorot((out_channels, in_channels, k, k), fan_in, fan_out))
+        shape = ((out_channels, k, k, in_channels) if layout == "nhwc"
+                 else (out_channels, in_channels, k, k))
+        self.weight = nn.Parameter(_glorot(shape, fan_in, fan_out))
         self.bias = nn.Parameter(torch.zeros(out_channels)) if bias else None
 
     def forward(self, x):
+        if self.layout == "nhwc":
+            return F.conv2d_act_nhwc(x, self.weight, self.bias, self.stride,
+                                     self.padding, self.activation)
         return F.conv2d_act(x, self.weight, self.bias, self.stride,
                             self.padding, self.activation)
 
 
 class MaxPool2d(nn.Module):
-    def __init__(self, kernel_size: int, stride: int = None, padding: int = 0):
+    def __init__(self, kernel_size: int, stride: int = None, padding: int = 0,
+                 layout: str = "nchw"):
         super().__init__()
         self.kernel_size = kernel_size
         self.stride = stride or kernel_size
         self.padding = padding
+        self.layout = layout
 
     def forward(self, x):
+        if self.layout == "nhwc":  # permute through the NCHW kernel (one
+            # pool in the ResNet stem — not worth its own kernel family)
+            y = F.maxpool2d(x.permute(0, 3, 1, 2).contiguous(),
+                            self.kernel_size, self.stride, self.padding)
+            return y.permute(0, 2, 3, 1).contiguous()
         return F.maxpool2d(x, self.kernel_size, self.stride, self.padding)
 
 
@@ -87,16 +104,24 @@ class BatchNorm2d(nn.Module):
     bn -> relu pair).  Eval/CPU fall back to torch batch_norm."""
 
     def __init__(self, num_features: int, eps: float = 1e-5,
-                 momentum: float = 0.1, fuse_relu: bool = False):
+                 momentum: float = 0.1, fuse_relu: bool = False,
+                 layout: str = "nchw"):
         super().__init__()
         self.num_features = num_features
         self.eps, self.momentum, self.fuse_relu = eps, momentum, fuse_relu
+        self.layout = layout
         self.weight = nn.Parameter(torch.ones(num_features))
         self.bias = nn.Parameter(torch.zeros(num_features))
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
 
     def forward(self, x):
+        if self.layout == "nhwc":
+            return F.batchnorm2d_act_nhwc(x, self.weight, self.bias,
+                                          self.running_mean,
+                                          self.running_var, self.momentum,
+                                          self.eps, self.fuse_relu,
+                                          self.training)
         return F.batchnorm2d_act(x, self.weight, self.bias,
                                  self.running_mean, self.running_var,
                                  self.momentum, self.eps, self.fuse_relu,

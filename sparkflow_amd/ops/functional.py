@@ -293,6 +293,113 @@ def conv2d_act(x, w, b=None, stride=1, pad=0, act="none"):
                            ACT_CODES[act] if isinstance(act, str) else act)
 
 
+class Conv2dActNHWC(torch.autograd.Function):
+    """Channel-last conv: x [N,H,W,Cin], w [Cout,Kh,Kw,Cin] — im2col
+    gathers and output stores are channel-contiguous float4s (the NCHW
+    kernels were gather-latency-bound on ResNet shapes).  CPU fallback
+    permutes through torch's NCHW conv."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, stride: int, pad: int, act: int):
+        if ops.use_hip(x):
+            y = ops.hip.conv_nhwc_fwd(
+                x, w, b if b is not None else torch.empty(0, device=x.device),
+                stride, pad, act)
+        else:
+            xc = x.permute(0, 3, 1, 2)
+            wc = w.permute(0, 3, 1, 2)
+            pre = torch.nn.functional.conv2d(xc, wc, b, stride=stride,
+                                             padding=pad)
+            y = _act_fwd(pre, act).permute(0, 2, 3, 1).contiguous()
+        ctx.save_for_backward(x, w, y)
+        ctx.conf = (stride, pad, act, b is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        stride, pad, act, has_bias = ctx.conf
+        need_dx = ctx.needs_input_grad[0]
+        dy = dy.contiguous()
+        kk = w.shape[1]
+        if ops.use_hip(x):
+            dpre = ops.hip.act_bwd(y, dy, act) if act != ACT_NONE else dy
+            if not need_dx:
+                dx = None
+            else:
+                # dgrad IS a stride-1 conv of (zero-stuffed) dpre with
+                # spatially-flipped, in/out-swapped weights
+                wt = w.flip(1, 2).permute(3, 1, 2, 0).contiguous()
+                if stride == 1:
+                    src = dpre
+                else:
+                    rh = (x.shape[1] + 2 * pad - kk) % stride
+                    rw = (x.shape[2] + 2 * pad - kk) % stride
+                    src = ops.hip.conv_nhwc_dilate(dpre, stride, rh, rw)
+                dx = ops.hip.conv_nhwc_fwd(
+                    src, wt, torch.empty(0, device=x.device), 1,
+                    kk - 1 - pad, ACT_NONE)
+            dw = ops.hip.conv_nhwc_wgrad(x, dpre, kk, w.shape[2], stride,
+                                         pad)
+            db = ops.hip.conv_nhwc_bias_grad(dpre) if has_bias else None
+        else:
+            dpre = _act_bwd_from_out(y, dy, act)
+            xc = x.permute(0, 3, 1, 2)
+            wc = w.permute(0, 3, 1, 2)
+            dpc = dpre.permute(0, 3, 1, 2)
+            dx = None
+            if need_dx:
+                dx = torch.nn.grad.conv2d_input(
+                    xc.shape, wc, dpc, stride=stride, padding=pad) \
+                    .permute(0, 2, 3, 1).contiguous()
+            dw = torch.nn.grad.conv2d_weight(
+                xc, wc.shape, dpc, stride=stride, padding=pad) \
+                .permute(0, 2, 3, 1).contiguous()
+            db = dpre.sum(dim=(0, 1, 2)) if has_bias else None
+        return dx, dw, db, None, None, None
+
+
+def conv2d_act_nhwc(x, w, b=None, stride=1, pad=0, act="none"):
+    return Conv2dActNHWC.apply(x, w, b, stride, pad,
+                               ACT_CODES[act] if isinstance(act, str)
+                               else act)
+
+
+class BatchNorm2dActNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, run_mean, run_var, momentum, eps,
+                relu):
+        y, mean, invstd = ops.hip.bn2d_nhwc_fwd(x, gamma, beta, run_mean,
+                                                run_var, momentum, eps,
+                                                relu)
+        ctx.save_for_backward(x, y, gamma, mean, invstd)
+        ctx.relu = relu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y, gamma, mean, invstd = ctx.saved_tensors
+        dx, dgamma, dbeta = ops.hip.bn2d_nhwc_bwd(x, dy.contiguous(), y,
+                                                  gamma, mean, invstd,
+                                                  ctx.relu)
+        return dx, dgamma, dbeta, None, None, None, None, None
+
+
+def batchnorm2d_act_nhwc(x, gamma, beta, run_mean, run_var, momentum=0.1,
+                         eps=1e-5, relu=False, training=True):
+    """x: [N, H, W, C] channel-last."""
+    if training and ops.use_hip(x):
+        return BatchNorm2dActNHWC.apply(x.contiguous(), gamma, beta,
+                                        run_mean, run_var, momentum, eps,
+                                        relu)
+    xc = x.permute(0, 3, 1, 2)
+    y = torch.nn.functional.batch_norm(xc, run_mean, run_var, gamma, beta,
+                                       training, momentum, eps)
+    if relu:
+        y = torch.relu(y)
+    return y.permute(0, 2, 3, 1).contiguous()
+
+
 class MaxPool2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, k: int, stride: int, pad: int):

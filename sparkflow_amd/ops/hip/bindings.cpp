@@ -77,6 +77,16 @@ std::vector<torch::Tensor> maxpool2d_fwd(torch::Tensor x, int64_t k,
 torch::Tensor maxpool2d_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
                             int64_t W);
 
+// conv_nhwc.hip — channel-last implicit-GEMM conv family
+torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w,
+                            torch::Tensor bias, int64_t stride, int64_t pad,
+                            int64_t act);
+torch::Tensor conv_nhwc_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
+                              int64_t Kw, int64_t stride, int64_t pad);
+torch::Tensor conv_nhwc_dilate(torch::Tensor dy, int64_t s, int64_t extra_h,
+                               int64_t extra_w);
+torch::Tensor conv_nhwc_bias_grad(torch::Tensor dy);
+
 // norm.hip — training-mode BatchNorm2d (+ fused ReLU)
 std::vector<torch::Tensor> bn2d_fwd(torch::Tensor x, torch::Tensor gamma,
                                     torch::Tensor beta,
@@ -87,6 +97,19 @@ std::vector<torch::Tensor> bn2d_bwd(torch::Tensor x, torch::Tensor dy,
                                     torch::Tensor y, torch::Tensor gamma,
                                     torch::Tensor save_mean,
                                     torch::Tensor save_invstd, bool relu);
+std::vector<torch::Tensor> bn2d_nhwc_fwd(torch::Tensor x,
+                                         torch::Tensor gamma,
+                                         torch::Tensor beta,
+                                         torch::Tensor run_mean,
+                                         torch::Tensor run_var,
+                                         double momentum, double eps,
+                                         bool relu);
+std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
+                                         torch::Tensor y,
+                                         torch::Tensor gamma,
+                                         torch::Tensor save_mean,
+                                         torch::Tensor save_invstd,
+                                         bool relu);
 
 // dist_session.hip — native RCCL session loops (GIL-free)
 py::bytes nccl_unique_id();
@@ -156,10 +179,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "zero-stuff dy by stride (strided dgrad as stride-1 conv)");
   m.def("maxpool2d_fwd", &maxpool2d_fwd, "maxpool fwd with argmax indices");
   m.def("maxpool2d_bwd", &maxpool2d_bwd, "maxpool bwd scatter");
+  m.def("conv_nhwc_fwd", &conv_nhwc_fwd,
+        "NHWC implicit-GEMM conv fwd (+bias+act); w [Cout,Kh,Kw,Cin]");
+  m.def("conv_nhwc_wgrad", &conv_nhwc_wgrad,
+        "NHWC implicit-GEMM conv weight grad (split-K)");
+  m.def("conv_nhwc_dilate", &conv_nhwc_dilate,
+        "zero-stuff NHWC dy by stride (strided dgrad as stride-1 conv)");
+  m.def("conv_nhwc_bias_grad", &conv_nhwc_bias_grad,
+        "NHWC conv bias grad (channel-contiguous column sums)");
   m.def("bn2d_fwd", &bn2d_fwd,
         "training BatchNorm2d fwd (+ fused ReLU); returns y, mean, invstd");
   m.def("bn2d_bwd", &bn2d_bwd,
         "BatchNorm2d bwd (+ fused ReLU bwd); returns dx, dgamma, dbeta");
+  m.def("bn2d_nhwc_fwd", &bn2d_nhwc_fwd, "NHWC training BatchNorm2d fwd");
+  m.def("bn2d_nhwc_bwd", &bn2d_nhwc_bwd, "NHWC BatchNorm2d bwd");
   m.def("nccl_unique_id", &nccl_unique_id, "RCCL unique id for a pair comm");
   m.def("nccl_comm_init", &nccl_comm_init,
         "init this rank's side of a raw RCCL communicator");

@@ -17,6 +17,7 @@
 #include <ATen/cuda/CUDAContext.h>
 
 #include "common.h"
+#include "conv_reduce.h"
 
 #define BM 64
 #define BN 64
@@ -336,59 +337,6 @@ __global__ __launch_bounds__(256) void conv_gemm_wgrad_kernel(
   }
 }
 
-// One block per output element, threads parallel over the z planes (the
-// deep conv split — up to 256 slices — made a per-thread z loop a serial
-// latency chain).  Used when splitk >= 64; shallow splits use the
-// grid-stride variant below (a 256-thread block per element would idle
-// most threads).
-__global__ void conv_splitk_reduce_kernel(const float* __restrict__ W,
-                                          float* __restrict__ C, long n,
-                                          int splitk) {
-  for (long i = blockIdx.x; i < n; i += gridDim.x) {
-    float sum = 0.0f;
-    for (int z = threadIdx.x; z < splitk; z += blockDim.x)
-      sum += W[(long)z * n + i];
-    sum = wave_sum(sum);
-    __shared__ float partials[4];
-    const int wid = threadIdx.x / WAVE;
-    if ((threadIdx.x & (WAVE - 1)) == 0) partials[wid] = sum;
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      float b = 0.0f;
-      for (int k = 0; k < (int)(blockDim.x / WAVE); ++k) b += partials[k];
-      C[i] = b;
-    }
-    __syncthreads();
-  }
-}
-
-__global__ void conv_splitk_reduce_flat_kernel(const float4* __restrict__ W,
-                                               float4* __restrict__ C,
-                                               long n4, int splitk) {
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
-       i += stride) {
-    float4 s = {0.0f, 0.0f, 0.0f, 0.0f};
-    for (int z = 0; z < splitk; ++z) {
-      const float4 w = W[(long)z * n4 + i];
-      s.x += w.x; s.y += w.y; s.z += w.z; s.w += w.w;
-    }
-    C[i] = s;
-  }
-}
-
-__global__ void conv_splitk_reduce_flat1_kernel(const float* __restrict__ W,
-                                                float* __restrict__ C,
-                                                long n, int splitk) {
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    float s = 0.0f;
-    for (int z = 0; z < splitk; ++z) s += W[(long)z * n + i];
-    C[i] = s;
-  }
-}
-
 // ---------------------------------------------------------------------------
 
 static inline void check_f32g(const torch::Tensor& t, const char* name) {
@@ -460,24 +408,8 @@ torch::Tensor conv_gemm_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
                        stream, x.data_ptr<float>(), dy.data_ptr<float>(),
                        ws.data_ptr<float>(), s, M, Nc, K, k_chunk);
     const long n = (long)M * Nc;
-    if (splitk >= 64) {
-      const int blocks = (int)std::min<long>(8192, n);
-      hipLaunchKernelGGL(conv_splitk_reduce_kernel, dim3(blocks), dim3(256),
-                         0, stream, ws.data_ptr<float>(),
-                         dw.data_ptr<float>(), n, splitk);
-    } else if ((n & 3) == 0) {
-      const long n4 = n >> 2;
-      const int blocks = (int)std::min<long>(4096, (n4 + 255) / 256);
-      hipLaunchKernelGGL(
-          conv_splitk_reduce_flat_kernel, dim3(blocks), dim3(256), 0,
-          stream, reinterpret_cast<const float4*>(ws.data_ptr<float>()),
-          reinterpret_cast<float4*>(dw.data_ptr<float>()), n4, splitk);
-    } else {
-      const int blocks = (int)std::min<long>(4096, (n + 255) / 256);
-      hipLaunchKernelGGL(conv_splitk_reduce_flat1_kernel, dim3(blocks),
-                         dim3(256), 0, stream, ws.data_ptr<float>(),
-                         dw.data_ptr<float>(), n, splitk);
-    }
+    launch_conv_splitk_reduce(ws.data_ptr<float>(), dw.data_ptr<float>(), n,
+                              splitk, stream);
   } else {
     dim3 grid(ceil_div(Nc, BN), ceil_div(M, BM));
     hipLaunchKernelGGL((conv_gemm_wgrad_kernel<false>), grid, dim3(256), 0,

@@ -1,0 +1,474 @@
+// NHWC implicit-GEMM convolution (fp32, MFMA) — the channel-last layout
+// makes every im2col gather a CONTIGUOUS float4 read along Cin and every
+// output store contiguous along Cout, fixing the two things that bound
+// the NCHW conv kernels on ResNet shapes (scalar gathers in the K loop,
+// scattered epilogue).  Used by the layout="nhwc" modules; weights live
+// as [Cout, Kh, Kw, Cin] so the k = (kh, kw, ci) GEMM dimension is the
+// weight row's native order (no per-step permutes).
+//
+// GEMM views:
+//   fwd:   C[m=(n,ho,wo)][co] = sum_k x[n, ho*s-p+kh, wo*s-p+kw, ci]
+//                                     * w[co][k],   k = (kh, kw, ci)
+//   wgrad: dw[co][k]          = sum_p dy[p, co] * im2col(x)[p, k]
+// Same 64x64-tile / 16-deep MFMA pipeline as gemm.hip / conv_gemm.hip.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+#include "conv_reduce.h"
+
+#define BM 64
+#define BN 64
+#define BK 16
+
+struct ConvShapeHW {  // NHWC shapes
+  int N, H, W, Cin, Cout, Kh, Kw, Ho, Wo, stride, pad;
+};
+
+// ---------------- forward ----------------
+template <bool HAS_BIAS>
+__global__ __launch_bounds__(256) void conv_nhwc_fwd_kernel(
+    const float* __restrict__ x, const float* __restrict__ wgt,
+    const float* __restrict__ bias, float* __restrict__ y, ConvShapeHW s,
+    int M, int K, int act) {
+  __shared__ float As[BM][BK + 1];
+  __shared__ float Bs[BK][BN + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = (wv >> 1) * 32;
+  const int wn = (wv & 1) * 32;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+  const int Nc = s.Cout;
+
+  f32x4 acc[2][2] = {};
+
+  // A staging: thread owns output pixel gm = m0 + (t>>2) and k-slot
+  // (t&3)*4.  The pixel decode happens once; the (khkw, ci) cursor
+  // advances incrementally (+BK with carry) — when Cin % 4 == 0 the 4
+  // consecutive k's sit inside one (kh, kw) cell, so the gather is ONE
+  // float4 from x.
+  const int st_row = t >> 2, st_c4 = (t & 3) * 4;
+  const int gm = m0 + st_row;
+  const int m_dec = gm < M ? gm : 0;
+  const int a_wo = m_dec % s.Wo;
+  const int a_ho = (m_dec / s.Wo) % s.Ho;
+  const long a_nbase =
+      (long)(m_dec / (s.Wo * s.Ho)) * s.H * s.W * s.Cin;
+  const int h0 = a_ho * s.stride - s.pad;
+  const int w0 = a_wo * s.stride - s.pad;
+  int a_ci = st_c4 % s.Cin;
+  int a_khkw = st_c4 / s.Cin;
+  const bool vec = (s.Cin & 3) == 0;
+
+  auto load_a = [&](int k0) -> float4 {
+    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gm >= M) return r;
+    if (vec) {  // 4 k's inside one (kh,kw) cell
+      const int kh = a_khkw / s.Kw, kw = a_khkw % s.Kw;
+      const int h = h0 + kh, w = w0 + kw;
+      if (k0 + st_c4 + 3 < K && h >= 0 && h < s.H && w >= 0 && w < s.W) {
+        r = *reinterpret_cast<const float4*>(
+            &x[a_nbase + ((long)h * s.W + w) * s.Cin + a_ci]);
+      }
+      return r;
+    }
+    float* rr = reinterpret_cast<float*>(&r);
+    int ci = a_ci, khkw = a_khkw;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int kh = khkw / s.Kw, kw = khkw % s.Kw;
+      const int h = h0 + kh, w = w0 + kw;
+      if (k0 + st_c4 + i < K && h >= 0 && h < s.H && w >= 0 && w < s.W) {
+        rr[i] = x[a_nbase + ((long)h * s.W + w) * s.Cin + ci];
+      }
+      if (++ci == s.Cin) { ci = 0; ++khkw; }
+    }
+    return r;
+  };
+  auto advance_a = [&]() {
+    a_ci += BK;
+    while (a_ci >= s.Cin) { a_ci -= s.Cin; ++a_khkw; }
+  };
+  // B staging: w[co][k] rows are contiguous in k — float4 straight read.
+  auto load_b = [&](int k0) -> float4 {
+    float4 r;
+    const int gn = n0 + st_row;
+    if (gn < Nc && k0 + BK <= K) {
+      r = *reinterpret_cast<const float4*>(&wgt[(long)gn * K + k0 + st_c4]);
+    } else {
+      float* rr = reinterpret_cast<float*>(&r);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int gk = k0 + st_c4 + i;
+        rr[i] = (gn < Nc && gk < K) ? wgt[(long)gn * K + gk] : 0.0f;
+      }
+    }
+    return r;
+  };
+
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  float4 ra = load_a(0);
+  float4 rb = load_b(0);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    {
+      const float* pa = reinterpret_cast<const float*>(&ra);
+      const float* pb = reinterpret_cast<const float*>(&rb);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) As[st_row][st_c4 + i] = pa[i];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) Bs[st_c4 + i][st_row] = pb[i];
+    }
+    __syncthreads();
+    if (k0 + BK < K) {
+      advance_a();
+      ra = load_a(k0 + BK);
+      rb = load_b(k0 + BK);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      const float a0 = As[wm + fr][kk * 4 + fk];
+      const float a1 = As[wm + 16 + fr][kk * 4 + fk];
+      const float b0 = Bs[kk * 4 + fk][wn + fr];
+      const float b1 = Bs[kk * 4 + fk][wn + 16 + fr];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: y[m][co] — contiguous along co (coalesced store)
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int co = n0 + wn + ni * 16 + fcol;
+      if (co >= Nc) continue;
+      const float bv = HAS_BIAS ? bias[co] : 0.0f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int m = m0 + wm + mi * 16 + frow + i;
+        if (m < M) {
+          y[(long)m * Nc + co] = act_apply(acc[mi][ni][i] + bv, act);
+        }
+      }
+    }
+  }
+}
+
+// ---------------- wgrad ----------------
+// dw[co][k'] = sum_p dy[p, co] * im2col(x)[p, k'] — A staged transposed
+// (dy contiguous along co), B's im2col contiguous along ci.
+template <bool SPLITK>
+__global__ __launch_bounds__(256) void conv_nhwc_wgrad_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    float* __restrict__ dw, ConvShapeHW s, int M, int Nc, int K,
+    int k_chunk) {
+  __shared__ float As[BM][BK + 1];
+  __shared__ float Bs[BK][BN + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = (wv >> 1) * 32;
+  const int wn = (wv & 1) * 32;
+  const int m0 = blockIdx.y * BM;  // co tile
+  const int n0 = blockIdx.x * BN;  // k' tile
+
+  f32x4 acc[2][2] = {};
+
+  int k_begin = 0, k_end = K;
+  if (SPLITK) {
+    k_begin = blockIdx.z * k_chunk;
+    k_end = min(K, k_begin + k_chunk);
+  }
+
+  // A: As[co][p] = dy[p][co] — thread owns pixel-row kk = t>>4 and 4
+  // consecutive co's, ONE float4 per chunk (dy NHWC = [p][co]).
+  const int a_kk = t >> 4, a_m4 = (t & 15) * 4;
+  // B: Bs[p][k'] — thread owns pixel-row b_kk = t>>4 and 4 consecutive
+  // k' = (kh,kw,ci): float4 from x when Cin % 4 == 0.
+  const int b_kk = t >> 4, b_n4 = (t & 15) * 4;
+  const int HoWo = s.Ho * s.Wo;
+
+  int pa_n = (k_begin + a_kk) / HoWo;
+  int pa_hw = (k_begin + a_kk) % HoWo;
+  int pb_n = (k_begin + b_kk) / HoWo;
+  int pb_ho = ((k_begin + b_kk) % HoWo) / s.Wo;
+  int pb_wo = ((k_begin + b_kk) % HoWo) % s.Wo;
+  // decode the 4 k' columns once (fixed for the whole K loop)
+  int kw_i[4], kh_i[4], ci_i[4];
+  bool gk_ok[4];
+  bool bvec = (s.Cin & 3) == 0;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int gk = n0 + b_n4 + i;
+    gk_ok[i] = gk < Nc;
+    const int g = gk_ok[i] ? gk : 0;
+    ci_i[i] = g % s.Cin;
+    const int khkw = g / s.Cin;
+    kw_i[i] = khkw % s.Kw;
+    kh_i[i] = khkw / s.Kw;
+  }
+  // the float4 fast path additionally needs the 4 k's in one (kh,kw)
+  bvec = bvec && gk_ok[3] && kh_i[0] == kh_i[3] && kw_i[0] == kw_i[3];
+  const long n_stride = (long)s.H * s.W * s.Cin;
+
+  auto load_a = [&](int gp0) -> float4 {
+    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gp0 < K) {
+      int n = pa_n, hw = pa_hw;
+      if ((s.Cout & 3) == 0 && m0 + a_m4 + 3 < s.Cout) {
+        r = *reinterpret_cast<const float4*>(
+            &dy[((long)n * HoWo + hw) * s.Cout + m0 + a_m4]);
+      } else {
+        float* rr = reinterpret_cast<float*>(&r);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int co = m0 + a_m4 + i;
+          rr[i] = co < s.Cout
+                      ? dy[((long)n * HoWo + hw) * s.Cout + co] : 0.0f;
+        }
+      }
+    }
+    return r;
+  };
+  auto load_b = [&](int gp0) -> float4 {
+    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gp0 >= K) return r;
+    const long nbase = (long)pb_n * n_stride;
+    if (bvec) {
+      const int h = pb_ho * s.stride - s.pad + kh_i[0];
+      const int w = pb_wo * s.stride - s.pad + kw_i[0];
+      if (h >= 0 && h < s.H && w >= 0 && w < s.W) {
+        r = *reinterpret_cast<const float4*>(
+            &x[nbase + ((long)h * s.W + w) * s.Cin + ci_i[0]]);
+      }
+      return r;
+    }
+    float* rr = reinterpret_cast<float*>(&r);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      if (gk_ok[i]) {
+        const int h = pb_ho * s.stride - s.pad + kh_i[i];
+        const int w = pb_wo * s.stride - s.pad + kw_i[i];
+        if (h >= 0 && h < s.H && w >= 0 && w < s.W)
+          rr[i] = x[nbase + ((long)h * s.W + w) * s.Cin + ci_i[i]];
+      }
+    }
+    return r;
+  };
+  auto advance = [&]() {  // both pixel cursors += BK
+    pa_hw += BK;
+    while (pa_hw >= HoWo) { pa_hw -= HoWo; ++pa_n; }
+    pb_wo += BK;
+    while (pb_wo >= s.Wo) { pb_wo -= s.Wo; ++pb_ho; }
+    while (pb_ho >= s.Ho) { pb_ho -= s.Ho; ++pb_n; }
+  };
+
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  float4 ra = load_a(k_begin + a_kk);
+  float4 rb = load_b(k_begin + b_kk);
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+    {
+      const float* pa = reinterpret_cast<const float*>(&ra);
+      const float* pb = reinterpret_cast<const float*>(&rb);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) As[a_m4 + i][a_kk] = pa[i];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) Bs[b_kk][b_n4 + i] = pb[i];
+    }
+    __syncthreads();
+    if (k0 + BK < k_end) {
+      advance();
+      ra = load_a(k0 + BK + a_kk);
+      rb = load_b(k0 + BK + b_kk);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      const float a0 = As[wm + fr][kk * 4 + fk];
+      const float a1 = As[wm + 16 + fr][kk * 4 + fk];
+      const float b0 = Bs[kk * 4 + fk][wn + fr];
+      const float b1 = Bs[kk * 4 + fk][wn + 16 + fr];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int kcol = n0 + wn + ni * 16 + fcol;
+      if (kcol >= Nc) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int co = m0 + wm + mi * 16 + frow + i;
+        if (co < M) {
+          if (SPLITK) {
+            dw[(long)blockIdx.z * M * Nc + (long)co * Nc + kcol] =
+                acc[mi][ni][i];
+          } else {
+            dw[(long)co * Nc + kcol] = acc[mi][ni][i];
+          }
+        }
+      }
+    }
+  }
+}
+
+// Zero-stuff dy (NHWC) for strided dgrad-as-convolution: rows stay
+// channel-contiguous (a row copy is C contiguous floats).
+__global__ void dilate_nhwc_kernel(const float* __restrict__ dy,
+                                   float* __restrict__ out, int Ho, int Wo,
+                                   int C, int s, long plane_out, int Wd) {
+  const long n = blockIdx.x;
+  const float* src = dy + n * (long)Ho * Wo * C;
+  float* dst = out + n * plane_out;
+  for (int ho = blockIdx.y; ho < Ho; ho += gridDim.y) {
+    for (int wo = 0; wo < Wo; ++wo) {
+      const float* srow = src + ((long)ho * Wo + wo) * C;
+      float* drow = dst + (((long)ho * s) * Wd + (long)wo * s) * C;
+      for (int c = threadIdx.x; c < C; c += blockDim.x) drow[c] = srow[c];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+
+static inline void check_f32w(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                  t.is_contiguous(),
+              name, " must be contiguous fp32 on GPU");
+}
+
+// x: [N, H, W, Cin], w: [Cout, Kh, Kw, Cin] -> y: [N, Ho, Wo, Cout]
+torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w,
+                            torch::Tensor bias, int64_t stride, int64_t pad,
+                            int64_t act) {
+  check_f32w(x, "x");
+  check_f32w(w, "w");
+  ConvShapeHW s;
+  s.N = x.size(0); s.H = x.size(1); s.W = x.size(2); s.Cin = x.size(3);
+  s.Cout = w.size(0); s.Kh = w.size(1); s.Kw = w.size(2);
+  TORCH_CHECK(w.size(3) == s.Cin, "Cin mismatch");
+  s.stride = stride; s.pad = pad;
+  s.Ho = (s.H + 2 * s.pad - s.Kh) / s.stride + 1;
+  s.Wo = (s.W + 2 * s.pad - s.Kw) / s.stride + 1;
+  const int M = s.N * s.Ho * s.Wo;
+  const int K = s.Kh * s.Kw * s.Cin;
+  auto y = torch::empty({s.N, s.Ho, s.Wo, s.Cout}, x.options());
+  dim3 grid(ceil_div(s.Cout, BN), ceil_div(M, BM));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (bias.numel() > 0) {
+    hipLaunchKernelGGL((conv_nhwc_fwd_kernel<true>), grid, dim3(256), 0,
+                       stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                       bias.data_ptr<float>(), y.data_ptr<float>(), s, M, K,
+                       (int)act);
+  } else {
+    hipLaunchKernelGGL((conv_nhwc_fwd_kernel<false>), grid, dim3(256), 0,
+                       stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                       nullptr, y.data_ptr<float>(), s, M, K, (int)act);
+  }
+  return y;
+}
+
+// x: [N, H, W, Cin], dy: [N, Ho, Wo, Cout] -> dw: [Cout, Kh, Kw, Cin]
+torch::Tensor conv_nhwc_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
+                              int64_t Kw, int64_t stride, int64_t pad) {
+  check_f32w(x, "x");
+  check_f32w(dy, "dy");
+  ConvShapeHW s;
+  s.N = x.size(0); s.H = x.size(1); s.W = x.size(2); s.Cin = x.size(3);
+  s.Cout = dy.size(3); s.Kh = Kh; s.Kw = Kw;
+  s.stride = stride; s.pad = pad;
+  s.Ho = dy.size(1); s.Wo = dy.size(2);
+  const int M = s.Cout;
+  const int Nc = s.Kh * s.Kw * s.Cin;
+  const int K = s.N * s.Ho * s.Wo;
+  auto dw = torch::empty({s.Cout, (long)Kh, (long)Kw, s.Cin}, x.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int base = ceil_div(Nc, BN) * ceil_div(M, BM);
+  int splitk = 1;
+  const int max_by_k = std::max(1, K / 256);
+  const int cap = sf_splitk_env_max();
+  while (splitk < 256 && splitk < cap && base * splitk <= 1024 &&
+         splitk * 2 <= max_by_k) {
+    splitk *= 2;
+  }
+  if (splitk > 1) {
+    const int k_chunk = ceil_div(K, splitk * BK) * BK;
+    auto ws = torch::empty({splitk, M, Nc}, x.options());
+    dim3 grid(ceil_div(Nc, BN), ceil_div(M, BM), splitk);
+    hipLaunchKernelGGL((conv_nhwc_wgrad_kernel<true>), grid, dim3(256), 0,
+                       stream, x.data_ptr<float>(), dy.data_ptr<float>(),
+                       ws.data_ptr<float>(), s, M, Nc, K, k_chunk);
+    const long n = (long)M * Nc;
+    launch_conv_splitk_reduce(ws.data_ptr<float>(), dw.data_ptr<float>(), n,
+                              splitk, stream);
+  } else {
+    dim3 grid(ceil_div(Nc, BN), ceil_div(M, BM));
+    hipLaunchKernelGGL((conv_nhwc_wgrad_kernel<false>), grid, dim3(256), 0,
+                       stream, x.data_ptr<float>(), dy.data_ptr<float>(),
+                       dw.data_ptr<float>(), s, M, Nc, K, K);
+  }
+  return dw;
+}
+
+torch::Tensor conv_nhwc_dilate(torch::Tensor dy, int64_t s, int64_t extra_h,
+                               int64_t extra_w) {
+  check_f32w(dy, "dy");
+  const int N = dy.size(0), Ho = dy.size(1), Wo = dy.size(2),
+            C = dy.size(3);
+  const long Hd = (long)(Ho - 1) * s + 1 + extra_h;
+  const long Wd = (long)(Wo - 1) * s + 1 + extra_w;
+  auto out = torch::zeros({N, Hd, Wd, C}, dy.options());
+  const int chunks = std::max(1, std::min(Ho, 2048 / std::max(1, N)));
+  hipLaunchKernelGGL(dilate_nhwc_kernel, dim3(N, chunks),
+                     dim3(std::min(256, std::max(64, C))), 0,
+                     at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
+                     out.data_ptr<float>(), Ho, Wo, C, (int)s, Hd * Wd * C,
+                     (int)Wd);
+  return out;
+}
+
+// db[co] = sum over (n, ho, wo) of dy[..., co] — rows are co-contiguous.
+__global__ void conv_nhwc_bias_grad_kernel(const float* __restrict__ dy,
+                                           float* __restrict__ db, long rows,
+                                           int C) {
+  const long r0 = blockIdx.x;
+  float s = 0.0f;
+  const int c = threadIdx.x;
+  if (c >= C) return;
+  for (long r = r0; r < rows; r += gridDim.x) s += dy[r * C + c];
+  atomicAdd(&db[c], s);
+}
+
+torch::Tensor conv_nhwc_bias_grad(torch::Tensor dy) {
+  check_f32w(dy, "dy");
+  const int C = dy.size(3);
+  const long rows = dy.numel() / C;
+  auto db = torch::zeros({C}, dy.options());
+  TORCH_CHECK(C <= 1024, "bias grad: C too large for one block");
+  const int blocks = (int)std::min<long>(512, rows);
+  hipLaunchKernelGGL(conv_nhwc_bias_grad_kernel, dim3(blocks),
+                     dim3(std::min(1024, ((C + 63) / 64) * 64)), 0,
+                     at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
+                     db.data_ptr<float>(), rows, C);
+  return db;
+}

@@ -233,3 +233,170 @@ std::vector<torch::Tensor> bn2d_bwd(torch::Tensor x, torch::Tensor dy,
   // (contiguous views into the reduce buffer — no extra copies)
   return {dx, red.narrow(0, C, C), red.narrow(0, 0, C)};
 }
+
+// ---- NHWC variants (x: [N, H, W, C] = [rows, C]) -------------------------
+// Channel is the contiguous dim: every sweep is `for c = tid; c < C`
+// over row-strided blocks — fully coalesced, stats combined with one
+// atomic per (block, channel).
+
+__global__ void bn_nhwc_stats_kernel(const float* __restrict__ x,
+                                     float* __restrict__ stats, long rows,
+                                     int C) {
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float s = 0.0f, s2 = 0.0f;
+    for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+      const float v = x[r * C + c];
+      s += v;
+      s2 += v * v;
+    }
+    atomicAdd(&stats[c], s);
+    atomicAdd(&stats[C + c], s2);
+  }
+}
+
+__global__ void bn_nhwc_norm_kernel(const float* __restrict__ x,
+                                    float* __restrict__ y,
+                                    const float* __restrict__ stats,
+                                    const float* __restrict__ gamma,
+                                    const float* __restrict__ beta,
+                                    float* __restrict__ save_mean,
+                                    float* __restrict__ save_invstd,
+                                    float* __restrict__ run_mean,
+                                    float* __restrict__ run_var,
+                                    float momentum, float eps, long rows,
+                                    int C, int relu) {
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    const float cnt = (float)rows;
+    const float mean = stats[c] / cnt;
+    const float var = stats[C + c] / cnt - mean * mean;
+    const float invstd = __frsqrt_rn(var + eps);
+    const float g = gamma ? gamma[c] : 1.0f;
+    const float b = beta ? beta[c] : 0.0f;
+    const float scale = g * invstd;
+    const float shift = b - mean * scale;
+    if (blockIdx.x == 0) {
+      save_mean[c] = mean;
+      save_invstd[c] = invstd;
+      if (run_mean) {
+        run_mean[c] = (1.0f - momentum) * run_mean[c] + momentum * mean;
+        const float ub = var * cnt / fmaxf(cnt - 1.0f, 1.0f);
+        run_var[c] = (1.0f - momentum) * run_var[c] + momentum * ub;
+      }
+    }
+    for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+      const float v = x[r * C + c] * scale + shift;
+      y[r * C + c] = relu ? fmaxf(v, 0.0f) : v;
+    }
+  }
+}
+
+__global__ void bn_nhwc_bwd_reduce_kernel(const float* __restrict__ x,
+                                          const float* __restrict__ dy,
+                                          const float* __restrict__ y,
+                                          const float* __restrict__ sm,
+                                          const float* __restrict__ si,
+                                          float* __restrict__ red, long rows,
+                                          int C, int relu) {
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    const float mean = sm[c], invstd = si[c];
+    float sd = 0.0f, sdx = 0.0f;
+    for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+      float d = dy[r * C + c];
+      if (relu && y[r * C + c] <= 0.0f) d = 0.0f;
+      sd += d;
+      sdx += d * (x[r * C + c] - mean) * invstd;
+    }
+    atomicAdd(&red[c], sd);
+    atomicAdd(&red[C + c], sdx);
+  }
+}
+
+__global__ void bn_nhwc_bwd_dx_kernel(const float* __restrict__ x,
+                                      const float* __restrict__ dy,
+                                      const float* __restrict__ y,
+                                      float* __restrict__ dx,
+                                      const float* __restrict__ sm,
+                                      const float* __restrict__ si,
+                                      const float* __restrict__ gamma,
+                                      const float* __restrict__ red,
+                                      long rows, int C, int relu) {
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    const float mean = sm[c], invstd = si[c];
+    const float g = gamma ? gamma[c] : 1.0f;
+    const float cnt = (float)rows;
+    const float mean_dy = red[c] / cnt;
+    const float mean_dyx = red[C + c] / cnt;
+    const float scale = g * invstd;
+    for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+      float d = dy[r * C + c];
+      if (relu && y[r * C + c] <= 0.0f) d = 0.0f;
+      const float xhat = (x[r * C + c] - mean) * invstd;
+      dx[r * C + c] = scale * (d - mean_dy - xhat * mean_dyx);
+    }
+  }
+}
+
+static inline dim3 bn_nhwc_grid(long rows) {
+  return dim3((int)std::min<long>(256, rows));
+}
+
+std::vector<torch::Tensor> bn2d_nhwc_fwd(torch::Tensor x,
+                                         torch::Tensor gamma,
+                                         torch::Tensor beta,
+                                         torch::Tensor run_mean,
+                                         torch::Tensor run_var,
+                                         double momentum, double eps,
+                                         bool relu) {
+  check_f32n(x, "x");
+  const int C = x.size(-1);
+  const long rows = x.numel() / C;
+  auto y = torch::empty_like(x);
+  auto stats = torch::zeros({2 * C}, x.options());
+  auto save_mean = torch::empty({C}, x.options());
+  auto save_invstd = torch::empty({C}, x.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const dim3 grid = bn_nhwc_grid(rows);
+  const int thr = std::min(256, ((C + 63) / 64) * 64);
+  hipLaunchKernelGGL(bn_nhwc_stats_kernel, grid, dim3(thr), 0, stream,
+                     x.data_ptr<float>(), stats.data_ptr<float>(), rows, C);
+  hipLaunchKernelGGL(
+      bn_nhwc_norm_kernel, grid, dim3(thr), 0, stream, x.data_ptr<float>(),
+      y.data_ptr<float>(), stats.data_ptr<float>(),
+      gamma.numel() ? gamma.data_ptr<float>() : nullptr,
+      beta.numel() ? beta.data_ptr<float>() : nullptr,
+      save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
+      run_mean.numel() ? run_mean.data_ptr<float>() : nullptr,
+      run_var.numel() ? run_var.data_ptr<float>() : nullptr,
+      (float)momentum, (float)eps, rows, C, (int)relu);
+  return {y, save_mean, save_invstd};
+}
+
+std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
+                                         torch::Tensor y,
+                                         torch::Tensor gamma,
+                                         torch::Tensor save_mean,
+                                         torch::Tensor save_invstd,
+                                         bool relu) {
+  check_f32n(x, "x");
+  check_f32n(dy, "dy");
+  const int C = x.size(-1);
+  const long rows = x.numel() / C;
+  auto red = torch::zeros({2 * C}, x.options());
+  auto dx = torch::empty_like(x);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const dim3 grid = bn_nhwc_grid(rows);
+  const int thr = std::min(256, ((C + 63) / 64) * 64);
+  const float* yp = relu ? y.data_ptr<float>() : nullptr;
+  hipLaunchKernelGGL(bn_nhwc_bwd_reduce_kernel, grid, dim3(thr), 0, stream,
+                     x.data_ptr<float>(), dy.data_ptr<float>(), yp,
+                     save_mean.data_ptr<float>(),
+                     save_invstd.data_ptr<float>(), red.data_ptr<float>(),
+                     rows, C, (int)relu);
+  hipLaunchKernelGGL(
+      bn_nhwc_bwd_dx_kernel, grid, dim3(thr), 0, stream,
+      x.data_ptr<float>(), dy.data_ptr<float>(), yp, dx.data_ptr<float>(),
+      save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
+      gamma.numel() ? gamma.data_ptr<float>() : nullptr,
+      red.data_ptr<float>(), rows, C, (int)relu);
+  return {dx, red.narrow(0, C, C), red.narrow(0, 0, C)};
+}

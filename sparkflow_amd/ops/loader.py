@@ -22,6 +22,7 @@ SOURCES = [
     HERE / "hip" / "optim.hip",
     HERE / "hip" / "conv.hip",
     HERE / "hip" / "conv_gemm.hip",
+    HERE / "hip" / "conv_nhwc.hip",
     HERE / "hip" / "norm.hip",
     HERE / "hip" / "dist_session.hip",
 ]
