@@ -234,21 +234,37 @@ std::vector<torch::Tensor> bn2d_bwd(torch::Tensor x, torch::Tensor dy,
   return {dx, red.narrow(0, C, C), red.narrow(0, 0, C)};
 }
 
-// ---- NHWC variants (x: [N, H, W, C] = [rows, C]) -------------------------
-// Channel is the contiguous dim: every sweep is `for c = tid; c < C`
-// over row-strided blocks — fully coalesced, stats combined with one
+// ---- NHWC variants (x: [rows, C], rows = N*H*W) --------------------------
+// grid (ceil(C/64), row_chunks); 256-thread blocks = 64 channel lanes x 4
+// row groups.  Reads are 64-channel-wide coalesced row segments; the
+// stats/reduce kernels combine the 4 row groups in LDS and emit ONE
 // atomic per (block, channel).
+
+#define BNC 64  // channel lanes per block
 
 __global__ void bn_nhwc_stats_kernel(const float* __restrict__ x,
                                      float* __restrict__ stats, long rows,
                                      int C) {
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float s = 0.0f, s2 = 0.0f;
-    for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+  const int cl = threadIdx.x & (BNC - 1);
+  const int rg = threadIdx.x >> 6;       // row group 0..3
+  const int c = blockIdx.x * BNC + cl;
+  const bool ok = c < C;
+  float s = 0.0f, s2 = 0.0f;
+  if (ok) {
+    for (long r = (long)blockIdx.y * 4 + rg; r < rows;
+         r += (long)gridDim.y * 4) {
       const float v = x[r * C + c];
       s += v;
       s2 += v * v;
     }
+  }
+  __shared__ float ls[4][BNC], ls2[4][BNC];
+  ls[rg][cl] = s;
+  ls2[rg][cl] = s2;
+  __syncthreads();
+  if (rg == 0 && ok) {
+    s = ls[0][cl] + ls[1][cl] + ls[2][cl] + ls[3][cl];
+    s2 = ls2[0][cl] + ls2[1][cl] + ls2[2][cl] + ls2[3][cl];
     atomicAdd(&stats[c], s);
     atomicAdd(&stats[C + c], s2);
   }
@@ -265,28 +281,31 @@ __global__ void bn_nhwc_norm_kernel(const float* __restrict__ x,
                                     float* __restrict__ run_var,
                                     float momentum, float eps, long rows,
                                     int C, int relu) {
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    const float cnt = (float)rows;
-    const float mean = stats[c] / cnt;
-    const float var = stats[C + c] / cnt - mean * mean;
-    const float invstd = __frsqrt_rn(var + eps);
-    const float g = gamma ? gamma[c] : 1.0f;
-    const float b = beta ? beta[c] : 0.0f;
-    const float scale = g * invstd;
-    const float shift = b - mean * scale;
-    if (blockIdx.x == 0) {
-      save_mean[c] = mean;
-      save_invstd[c] = invstd;
-      if (run_mean) {
-        run_mean[c] = (1.0f - momentum) * run_mean[c] + momentum * mean;
-        const float ub = var * cnt / fmaxf(cnt - 1.0f, 1.0f);
-        run_var[c] = (1.0f - momentum) * run_var[c] + momentum * ub;
-      }
+  const int cl = threadIdx.x & (BNC - 1);
+  const int rg = threadIdx.x >> 6;
+  const int c = blockIdx.x * BNC + cl;
+  if (c >= C) return;
+  const float cnt = (float)rows;
+  const float mean = stats[c] / cnt;
+  const float var = stats[C + c] / cnt - mean * mean;
+  const float invstd = __frsqrt_rn(var + eps);
+  const float g = gamma ? gamma[c] : 1.0f;
+  const float b = beta ? beta[c] : 0.0f;
+  const float scale = g * invstd;
+  const float shift = b - mean * scale;
+  if (blockIdx.y == 0 && rg == 0) {
+    save_mean[c] = mean;
+    save_invstd[c] = invstd;
+    if (run_mean) {  // torch semantics: unbiased var into running stats
+      run_mean[c] = (1.0f - momentum) * run_mean[c] + momentum * mean;
+      const float ub = var * cnt / fmaxf(cnt - 1.0f, 1.0f);
+      run_var[c] = (1.0f - momentum) * run_var[c] + momentum * ub;
     }
-    for (long r = blockIdx.x; r < rows; r += gridDim.x) {
-      const float v = x[r * C + c] * scale + shift;
-      y[r * C + c] = relu ? fmaxf(v, 0.0f) : v;
-    }
+  }
+  for (long r = (long)blockIdx.y * 4 + rg; r < rows;
+       r += (long)gridDim.y * 4) {
+    const float v = x[r * C + c] * scale + shift;
+    y[r * C + c] = relu ? fmaxf(v, 0.0f) : v;
   }
 }
 
@@ -297,15 +316,28 @@ __global__ void bn_nhwc_bwd_reduce_kernel(const float* __restrict__ x,
                                           const float* __restrict__ si,
                                           float* __restrict__ red, long rows,
                                           int C, int relu) {
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+  const int cl = threadIdx.x & (BNC - 1);
+  const int rg = threadIdx.x >> 6;
+  const int c = blockIdx.x * BNC + cl;
+  const bool ok = c < C;
+  float sd = 0.0f, sdx = 0.0f;
+  if (ok) {
     const float mean = sm[c], invstd = si[c];
-    float sd = 0.0f, sdx = 0.0f;
-    for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+    for (long r = (long)blockIdx.y * 4 + rg; r < rows;
+         r += (long)gridDim.y * 4) {
       float d = dy[r * C + c];
       if (relu && y[r * C + c] <= 0.0f) d = 0.0f;
       sd += d;
       sdx += d * (x[r * C + c] - mean) * invstd;
     }
+  }
+  __shared__ float ls[4][BNC], ls2[4][BNC];
+  ls[rg][cl] = sd;
+  ls2[rg][cl] = sdx;
+  __syncthreads();
+  if (rg == 0 && ok) {
+    sd = ls[0][cl] + ls[1][cl] + ls[2][cl] + ls[3][cl];
+    sdx = ls2[0][cl] + ls2[1][cl] + ls2[2][cl] + ls2[3][cl];
     atomicAdd(&red[c], sd);
     atomicAdd(&red[C + c], sdx);
   }
@@ -320,24 +352,30 @@ __global__ void bn_nhwc_bwd_dx_kernel(const float* __restrict__ x,
                                       const float* __restrict__ gamma,
                                       const float* __restrict__ red,
                                       long rows, int C, int relu) {
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    const float mean = sm[c], invstd = si[c];
-    const float g = gamma ? gamma[c] : 1.0f;
-    const float cnt = (float)rows;
-    const float mean_dy = red[c] / cnt;
-    const float mean_dyx = red[C + c] / cnt;
-    const float scale = g * invstd;
-    for (long r = blockIdx.x; r < rows; r += gridDim.x) {
-      float d = dy[r * C + c];
-      if (relu && y[r * C + c] <= 0.0f) d = 0.0f;
-      const float xhat = (x[r * C + c] - mean) * invstd;
-      dx[r * C + c] = scale * (d - mean_dy - xhat * mean_dyx);
-    }
+  const int cl = threadIdx.x & (BNC - 1);
+  const int rg = threadIdx.x >> 6;
+  const int c = blockIdx.x * BNC + cl;
+  if (c >= C) return;
+  const float mean = sm[c], invstd = si[c];
+  const float g = gamma ? gamma[c] : 1.0f;
+  const float cnt = (float)rows;
+  const float mean_dy = red[c] / cnt;
+  const float mean_dyx = red[C + c] / cnt;
+  const float scale = g * invstd;
+  for (long r = (long)blockIdx.y * 4 + rg; r < rows;
+       r += (long)gridDim.y * 4) {
+    float d = dy[r * C + c];
+    if (relu && y[r * C + c] <= 0.0f) d = 0.0f;
+    const float xhat = (x[r * C + c] - mean) * invstd;
+    dx[r * C + c] = scale * (d - mean_dy - xhat * mean_dyx);
   }
 }
 
-static inline dim3 bn_nhwc_grid(long rows) {
-  return dim3((int)std::min<long>(256, rows));
+static inline dim3 bn_nhwc_grid(long rows, int C) {
+  const int cgrids = (C + BNC - 1) / BNC;
+  long chunks = (rows + 3) / 4;
+  chunks = std::min(chunks, (long)std::max(1, 2048 / cgrids));
+  return dim3(cgrids, (int)chunks);
 }
 
 std::vector<torch::Tensor> bn2d_nhwc_fwd(torch::Tensor x,
@@ -355,12 +393,11 @@ std::vector<torch::Tensor> bn2d_nhwc_fwd(torch::Tensor x,
   auto save_mean = torch::empty({C}, x.options());
   auto save_invstd = torch::empty({C}, x.options());
   auto stream = at::cuda::getCurrentCUDAStream();
-  const dim3 grid = bn_nhwc_grid(rows);
-  const int thr = std::min(256, ((C + 63) / 64) * 64);
-  hipLaunchKernelGGL(bn_nhwc_stats_kernel, grid, dim3(thr), 0, stream,
+  const dim3 grid = bn_nhwc_grid(rows, C);
+  hipLaunchKernelGGL(bn_nhwc_stats_kernel, grid, dim3(256), 0, stream,
                      x.data_ptr<float>(), stats.data_ptr<float>(), rows, C);
   hipLaunchKernelGGL(
-      bn_nhwc_norm_kernel, grid, dim3(thr), 0, stream, x.data_ptr<float>(),
+      bn_nhwc_norm_kernel, grid, dim3(256), 0, stream, x.data_ptr<float>(),
       y.data_ptr<float>(), stats.data_ptr<float>(),
       gamma.numel() ? gamma.data_ptr<float>() : nullptr,
       beta.numel() ? beta.data_ptr<float>() : nullptr,
@@ -384,16 +421,15 @@ std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
   auto red = torch::zeros({2 * C}, x.options());
   auto dx = torch::empty_like(x);
   auto stream = at::cuda::getCurrentCUDAStream();
-  const dim3 grid = bn_nhwc_grid(rows);
-  const int thr = std::min(256, ((C + 63) / 64) * 64);
+  const dim3 grid = bn_nhwc_grid(rows, C);
   const float* yp = relu ? y.data_ptr<float>() : nullptr;
-  hipLaunchKernelGGL(bn_nhwc_bwd_reduce_kernel, grid, dim3(thr), 0, stream,
+  hipLaunchKernelGGL(bn_nhwc_bwd_reduce_kernel, grid, dim3(256), 0, stream,
                      x.data_ptr<float>(), dy.data_ptr<float>(), yp,
                      save_mean.data_ptr<float>(),
                      save_invstd.data_ptr<float>(), red.data_ptr<float>(),
                      rows, C, (int)relu);
   hipLaunchKernelGGL(
-      bn_nhwc_bwd_dx_kernel, grid, dim3(thr), 0, stream,
+      bn_nhwc_bwd_dx_kernel, grid, dim3(256), 0, stream,
       x.data_ptr<float>(), dy.data_ptr<float>(), yp, dx.data_ptr<float>(),
       save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
       gamma.numel() ? gamma.data_ptr<float>() : nullptr,
