@@ -33,8 +33,8 @@ def main():
     ap.add_argument("--mbs", type=int, default=300,
                     help="minibatch per step (reference config: 300)")
     ap.add_argument("--model", default="mnist_mlp",
-                    choices=["mnist_mlp", "mnist_cnn", "tabular_autoencoder",
-                             "resnet18_shape"])
+                    choices=["mnist_mlp", "mnist_cnn", "mnist_cnn_nhwc",
+                             "tabular_autoencoder", "resnet18_shape"])
     ap.add_argument("--optimizer", default="adam")
     ap.add_argument("--lr", type=float, default=0.001)
     ap.add_argument("--lock", action="store_true",
@@ -71,7 +71,7 @@ def main():
         n_rows = max(2 * args.mbs, 64)
         x = rng.random((n_rows, 3 * 224 * 224), dtype=np.float32)
         y = rng.integers(0, 1000, n_rows).astype(np.float32).reshape(-1, 1)
-    elif args.model == "mnist_cnn" or args.model == "mnist_mlp":
+    elif args.model in ("mnist_cnn", "mnist_cnn_nhwc", "mnist_mlp"):
         x = rng.random((n_rows, 784), dtype=np.float32)
         y = rng.integers(0, 10, n_rows).astype(np.float32).reshape(-1, 1)
     else:  # autoencoder: unsupervised
@@ -126,6 +126,7 @@ def main():
             "config": {
                 "model": {"mnist_mlp": "mnist_mlp_784_256_256_10",
                           "mnist_cnn": "mnist_cnn_5x5x32_3x3x64",
+                          "mnist_cnn_nhwc": "mnist_cnn_5x5x32_3x3x64_nhwc",
                           "tabular_autoencoder": "autoencoder_784_32_784",
                           "resnet18_shape": "resnet18_shape_3x224x224"}[args.model],
                 "global_batch": mbs_eff * world * args.workers,

@@ -3,6 +3,7 @@ from sparkflow_amd.models.cnn import MNISTCNN, ResNet18Shape  # noqa: F401
 from sparkflow_amd.models.zoo import (  # noqa: F401
     mnist_mlp,
     mnist_cnn,
+    mnist_cnn_nhwc,
     mnist_autoencoder,
     tabular_autoencoder,
     resnet18_shape,

@@ -18,20 +18,26 @@ from sparkflow_amd.ops import functional as F
 
 
 class MNISTCNN(nn.Module):
-    """Input: flat [B, 784] (Spark vector rows), reshaped to NCHW."""
+    """Input: flat [B, 784] (Spark vector rows), reshaped to NCHW (or
+    channel-last with layout="nhwc")."""
 
-    def __init__(self, num_classes: int = 10):
+    def __init__(self, num_classes: int = 10, layout: str = "nchw"):
         super().__init__()
-        self.conv1 = snn.Conv2d(1, 32, 5, padding="same", activation="relu")
-        self.pool1 = snn.MaxPool2d(2, 2)
-        self.conv2 = snn.Conv2d(32, 64, 3, padding="same", activation="relu")
-        self.pool2 = snn.MaxPool2d(2, 2)
+        self.layout = layout
+        self.conv1 = snn.Conv2d(1, 32, 5, padding="same", activation="relu",
+                                layout=layout)
+        self.pool1 = snn.MaxPool2d(2, 2, layout=layout)
+        self.conv2 = snn.Conv2d(32, 64, 3, padding="same", activation="relu",
+                                layout=layout)
+        self.pool2 = snn.MaxPool2d(2, 2, layout=layout)
         self.fc = snn.Linear(7 * 7 * 64, num_classes)
         self.loss_fn = snn.SoftmaxCrossEntropy()
 
     def forward(self, x):
         if x.dim() == 2:
             x = x.reshape(-1, 1, 28, 28)
+        if self.layout == "nhwc":
+            x = x.permute(0, 2, 3, 1).contiguous()
         x = self.pool1(self.conv1(x))
         x = self.pool2(self.conv2(x))
         x = x.reshape(x.shape[0], -1)

@@ -19,6 +19,11 @@ def mnist_cnn():
     return MNISTCNN()
 
 
+def mnist_cnn_nhwc():
+    """Channel-last variant of the MNIST CNN."""
+    return MNISTCNN(layout="nhwc")
+
+
 def mnist_autoencoder():
     """784-256-128-256-784 AE (reference examples/autoencoder_example.py)."""
     return Autoencoder([784, 256, 128, 256, 784])
