@@ -85,8 +85,12 @@ class MaxPool2d(nn.Module):
         self.layout = layout
 
     def forward(self, x):
-        if self.layout == "nhwc":  # permute through the NCHW kernel (one
-            # pool in the ResNet stem — not worth its own kernel family)
+        if self.layout == "nhwc":
+            from sparkflow_amd import ops as _ops
+
+            if x.is_cuda and _ops.available():
+                return F.maxpool2d_nhwc(x, self.kernel_size, self.stride,
+                                        self.padding)
             y = F.maxpool2d(x.permute(0, 3, 1, 2).contiguous(),
                             self.kernel_size, self.stride, self.padding)
             return y.permute(0, 2, 3, 1).contiguous()

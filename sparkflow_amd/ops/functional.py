@@ -430,6 +430,29 @@ def maxpool2d(x, k, stride=None, pad=0):
     return MaxPool2d.apply(x, k, stride or k, pad)
 
 
+class MaxPool2dNHWC(torch.autograd.Function):
+    """Channel-last maxpool (GPU only; the module's CPU path permutes
+    through the torch NCHW pool)."""
+
+    @staticmethod
+    def forward(ctx, x, k: int, stride: int, pad: int):
+        y, idx = ops.hip.maxpool_nhwc_fwd(x, k, stride, pad)
+        ctx.save_for_backward(idx)
+        ctx.conf = (x.shape[1], x.shape[2])
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        H, W = ctx.conf
+        return ops.hip.maxpool_nhwc_bwd(dy.contiguous(), idx, H, W), \
+            None, None, None
+
+
+def maxpool2d_nhwc(x, k, stride=None, pad=0):
+    return MaxPool2dNHWC.apply(x, k, stride or k, pad)
+
+
 class BatchNorm2dAct(torch.autograd.Function):
     """Training-mode BatchNorm2d with optional fused ReLU on the custom
     CDNA4 kernels (MIOpen's spatial BN was 27% of the ResNet-shape step).

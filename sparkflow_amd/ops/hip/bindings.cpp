@@ -86,6 +86,10 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
 torch::Tensor conv_nhwc_dilate(torch::Tensor dy, int64_t s, int64_t extra_h,
                                int64_t extra_w);
 torch::Tensor conv_nhwc_bias_grad(torch::Tensor dy);
+std::vector<torch::Tensor> maxpool_nhwc_fwd(torch::Tensor x, int64_t k,
+                                            int64_t stride, int64_t pad);
+torch::Tensor maxpool_nhwc_bwd(torch::Tensor dy, torch::Tensor idx,
+                               int64_t H, int64_t W);
 
 // norm.hip — training-mode BatchNorm2d (+ fused ReLU)
 std::vector<torch::Tensor> bn2d_fwd(torch::Tensor x, torch::Tensor gamma,
@@ -187,6 +191,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "zero-stuff NHWC dy by stride (strided dgrad as stride-1 conv)");
   m.def("conv_nhwc_bias_grad", &conv_nhwc_bias_grad,
         "NHWC conv bias grad (channel-contiguous column sums)");
+  m.def("maxpool_nhwc_fwd", &maxpool_nhwc_fwd,
+        "NHWC maxpool fwd with argmax indices");
+  m.def("maxpool_nhwc_bwd", &maxpool_nhwc_bwd, "NHWC maxpool bwd scatter");
   m.def("bn2d_fwd", &bn2d_fwd,
         "training BatchNorm2d fwd (+ fused ReLU); returns y, mean, invstd");
   m.def("bn2d_bwd", &bn2d_bwd,

@@ -472,3 +472,93 @@ torch::Tensor conv_nhwc_bias_grad(torch::Tensor dy) {
                      db.data_ptr<float>(), rows, C);
   return db;
 }
+
+// ---- NHWC maxpool --------------------------------------------------------
+// out[n,ho,wo,c]: threads sweep channels (contiguous), wo in an inner
+// loop, (n,ho) on the grid.  idx stores h*W+w (per-plane convention,
+// layout-independent) for the backward scatter.
+__global__ void maxpool_nhwc_fwd_kernel(const float* __restrict__ x,
+                                        float* __restrict__ y,
+                                        long* __restrict__ idx, int N, int C,
+                                        int H, int W, int K, int stride,
+                                        int pad, int Ho, int Wo) {
+  const int n = blockIdx.x;
+  const float* xp = x + (long)n * H * W * C;
+  for (int ho = blockIdx.y; ho < Ho; ho += gridDim.y) {
+    const int h0 = ho * stride - pad;
+    for (int wo = 0; wo < Wo; ++wo) {
+      const int w0 = wo * stride - pad;
+      const long obase = (((long)n * Ho + ho) * Wo + wo) * C;
+      for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float best = -INFINITY;
+        long best_hw = 0;
+        for (int kh = 0; kh < K; ++kh) {
+          const int h = h0 + kh;
+          if (h < 0) continue;
+          if (h >= H) break;
+          for (int kw = 0; kw < K; ++kw) {
+            const int ww = w0 + kw;
+            if (ww < 0) continue;
+            if (ww >= W) break;
+            const float v = xp[((long)h * W + ww) * C + c];
+            if (v > best) {
+              best = v;
+              best_hw = (long)h * W + ww;
+            }
+          }
+        }
+        y[obase + c] = best;
+        idx[obase + c] = best_hw;
+      }
+    }
+  }
+}
+
+__global__ void maxpool_nhwc_bwd_kernel(const float* __restrict__ dy,
+                                        const long* __restrict__ idx,
+                                        float* __restrict__ dx, int N, int C,
+                                        int HW, int HoWo) {
+  const int n = blockIdx.x;
+  const float* dyp = dy + (long)n * HoWo * C;
+  const long* ip = idx + (long)n * HoWo * C;
+  float* dxp = dx + (long)n * HW * C;
+  const int span = gridDim.y * blockDim.x;
+  for (long i = (long)blockIdx.y * blockDim.x + threadIdx.x;
+       i < (long)HoWo * C; i += span) {
+    const int c = i % C;
+    atomicAdd(&dxp[ip[i] * C + c], dyp[i]);
+  }
+}
+
+std::vector<torch::Tensor> maxpool_nhwc_fwd(torch::Tensor x, int64_t k,
+                                            int64_t stride, int64_t pad) {
+  check_f32w(x, "x");
+  const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  const int Ho = (H + 2 * pad - k) / stride + 1;
+  const int Wo = (W + 2 * pad - k) / stride + 1;
+  auto y = torch::empty({N, Ho, Wo, C}, x.options());
+  auto idx = torch::empty({N, Ho, Wo, C}, x.options().dtype(torch::kInt64));
+  const int chunks = std::max(1, std::min(Ho, 2048 / std::max(1, N)));
+  hipLaunchKernelGGL(maxpool_nhwc_fwd_kernel, dim3(N, chunks),
+                     dim3(std::min(256, std::max(64, C))), 0,
+                     at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
+                     y.data_ptr<float>(), idx.data_ptr<long>(), N, C, H, W,
+                     (int)k, (int)stride, (int)pad, Ho, Wo);
+  return {y, idx};
+}
+
+torch::Tensor maxpool_nhwc_bwd(torch::Tensor dy, torch::Tensor idx,
+                               int64_t H, int64_t W) {
+  check_f32w(dy, "dy");
+  const int N = dy.size(0), C = dy.size(3);
+  const int HoWo = dy.size(1) * dy.size(2);
+  auto dx = torch::zeros({N, (long)H, (long)W, C}, dy.options());
+  const int chunks =
+      std::max(1, std::min((HoWo * C + 255) / 256,
+                           4096 / std::max(1, N)));
+  hipLaunchKernelGGL(maxpool_nhwc_bwd_kernel, dim3(N, chunks), dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
+                     idx.data_ptr<long>(), dx.data_ptr<float>(), N, C,
+                     (int)(H * W), HoWo);
+  return dx;
+}

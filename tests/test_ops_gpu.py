@@ -426,3 +426,25 @@ def test_resnet_nhwc_step_gpu():
     torch.cuda.synchronize()
     assert ps.update_count >= 1
     assert torch.isfinite(ps.weights).all()
+
+
+@pytest.mark.parametrize("k,s,p,hw,c", [(3, 2, 1, 33, 16), (2, 2, 0, 14, 8)])
+def test_maxpool_nhwc_gpu(k, s, p, hw, c):
+    torch.manual_seed(18)
+    x = torch.randn(2, hw, hw, c)
+    proj = None
+
+    def run(device):
+        from sparkflow_amd.nn.modules import MaxPool2d
+
+        m = MaxPool2d(k, s, padding=p, layout="nhwc")
+        xi = x.clone().to(device).requires_grad_()
+        y = m(xi)
+        nonlocal proj
+        if proj is None:
+            proj = torch.randn_like(y.detach().cpu())
+        (y * proj.to(device)).sum().backward()
+        return y.detach().cpu(), xi.grad.cpu()
+
+    for g, cc in zip(run(DEV), run("cpu")):
+        torch.testing.assert_close(g, cc, atol=1e-5, rtol=1e-5)
