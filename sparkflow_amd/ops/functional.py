@@ -420,9 +420,13 @@ class MaxPool2d(torch.autograd.Function):
         if ops.use_hip(dy):
             dx = ops.hip.maxpool2d_bwd(dy, idx, shape[2], shape[3])
         else:
-            dx = torch.nn.functional.max_unpool2d(dy, idx, k, stride,
-                                                  padding=pad,
-                                                  output_size=shape[2:])
+            # scatter_add, NOT max_unpool2d: with overlapping windows
+            # (stride < k) one input can be the max of several windows
+            # and unpool would overwrite instead of accumulating
+            n, c = shape[0], shape[1]
+            dx = torch.zeros(n, c, shape[2] * shape[3], dtype=dy.dtype)
+            dx.scatter_add_(2, idx.reshape(n, c, -1), dy.reshape(n, c, -1))
+            dx = dx.reshape(shape)
         return dx, None, None, None
 
 
