@@ -202,3 +202,28 @@ def test_multi_replay_native_gpu():
     assert eng.ps.update_count == 14
     assert torch.isfinite(eng.ps.weights).all()
     eng.close()
+
+
+def test_hogwild_tcp_gpu_workers():
+    """The Spark-shaped path on a GPU box: PS in a spawned process
+    (auto -> cuda), workers on cuda over the raw-tensor TCP transport."""
+    from tests.conftest import free_port
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import gaussian_mlp
+    from sparkflow_amd.parallel.hogwild import HogwildSparkModel
+    from sparkflow_amd.parallel.ps import OptimizerSpec
+
+    rng = np.random.default_rng(21)
+    rows = [(rng.normal(-2 if i % 2 == 0 else 2, 0.7, 10), float(i % 2))
+            for i in range(200)]
+    port = free_port()
+    trainer = HogwildSparkModel(
+        tensorflowGraph=build_graph(gaussian_mlp), iters=4,
+        tfInput="x", tfLabel="y",
+        optimizer=OptimizerSpec("adam", 0.01),
+        master_url="127.0.0.1:%d" % port, port=port,
+        mini_batch=32, mini_stochastic_iters=1)
+    weights = trainer.train([rows[0::2], rows[1::2]])
+    assert len(weights) == 4  # 2 layers x (w, b)
+    assert all(np.isfinite(w).all() for w in weights)
