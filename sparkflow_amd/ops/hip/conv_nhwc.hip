@@ -357,13 +357,6 @@ static inline void check_f32w(const torch::Tensor& t, const char* name) {
               name, " must be contiguous fp32 on GPU");
 }
 
-// defined below (BM=128 tile variant)
-template <bool HAS_BIAS>
-__global__ __launch_bounds__(256) void conv_nhwc_fwd128_kernel(
-    const float* __restrict__ x, const float* __restrict__ wgt,
-    const float* __restrict__ bias, float* __restrict__ y, ConvShapeHW s,
-    int M, int K, int act);
-
 // x: [N, H, W, Cin], w: [Cout, Kh, Kw, Cin] -> y: [N, Ho, Wo, Cout]
 torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w,
                             torch::Tensor bias, int64_t stride, int64_t pad,
@@ -380,24 +373,8 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w,
   const int M = s.N * s.Ho * s.Wo;
   const int K = s.Kh * s.Kw * s.Cin;
   auto y = torch::empty({s.N, s.Ho, s.Wo, s.Cout}, x.options());
-  auto stream = at::cuda::getCurrentCUDAStream();
-  // 128-row tiles when the grid still fills the chip at that size
-  const bool big = (long)ceil_div(M, 128) * ceil_div(s.Cout, BN) >= 512;
-  if (big) {
-    dim3 grid(ceil_div(s.Cout, BN), ceil_div(M, 128));
-    if (bias.numel() > 0) {
-      hipLaunchKernelGGL((conv_nhwc_fwd128_kernel<true>), grid, dim3(256),
-                         0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
-                         bias.data_ptr<float>(), y.data_ptr<float>(), s, M,
-                         K, (int)act);
-    } else {
-      hipLaunchKernelGGL((conv_nhwc_fwd128_kernel<false>), grid, dim3(256),
-                         0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
-                         nullptr, y.data_ptr<float>(), s, M, K, (int)act);
-    }
-    return y;
-  }
   dim3 grid(ceil_div(s.Cout, BN), ceil_div(M, BM));
+  auto stream = at::cuda::getCurrentCUDAStream();
   if (bias.numel() > 0) {
     hipLaunchKernelGGL((conv_nhwc_fwd_kernel<true>), grid, dim3(256), 0,
                        stream, x.data_ptr<float>(), w.data_ptr<float>(),
@@ -584,162 +561,4 @@ torch::Tensor maxpool_nhwc_bwd(torch::Tensor dy, torch::Tensor idx,
                      idx.data_ptr<long>(), dx.data_ptr<float>(), N, C,
                      (int)(H * W), HoWo);
   return dx;
-}
-
-// ---- BM=128 forward variant ----------------------------------------------
-// Twice the output rows per block: each wave computes a 64m x 32n patch
-// (4x2 MFMA fragments), so one LDS stage + sync covers twice the
-// arithmetic — picked when M is large enough to still fill the chip
-// (every ResNet mainline conv).
-#define BM2 128
-
-template <bool HAS_BIAS>
-__global__ __launch_bounds__(256) void conv_nhwc_fwd128_kernel(
-    const float* __restrict__ x, const float* __restrict__ wgt,
-    const float* __restrict__ bias, float* __restrict__ y, ConvShapeHW s,
-    int M, int K, int act) {
-  __shared__ float As[BM2][BK + 1];
-  __shared__ float Bs[BK][BN + 4];
-
-  const int t = threadIdx.x;
-  const int lane = t & 63;
-  const int wv = t >> 6;
-  const int wm = (wv >> 1) * 64;  // wave owns 64 m-rows
-  const int wn = (wv & 1) * 32;
-  const int m0 = blockIdx.y * BM2;
-  const int n0 = blockIdx.x * BN;
-  const int Nc = s.Cout;
-
-  f32x4 acc[4][2] = {};
-
-  // A staging: 128 rows x 16 k = 2 float4 per thread.
-  // row = t >> 1 (2 threads per row), k-half = (t & 1) * 8.
-  const int a_row = t >> 1, a_c8 = (t & 1) * 8;
-  const int gm = m0 + a_row;
-  const int m_dec = gm < M ? gm : 0;
-  const int a_wo = m_dec % s.Wo;
-  const int a_ho = (m_dec / s.Wo) % s.Ho;
-  const long a_nbase = (long)(m_dec / (s.Wo * s.Ho)) * s.H * s.W * s.Cin;
-  const int h0 = a_ho * s.stride - s.pad;
-  const int w0 = a_wo * s.stride - s.pad;
-  int a_ci = a_c8 % s.Cin;
-  int a_khkw = a_c8 / s.Cin;
-  const bool vec = (s.Cin & 3) == 0;
-
-  auto load_a_one = [&](int k0s, int ci, int khkw) -> float4 {
-    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
-    if (gm >= M) return r;
-    if (vec) {
-      const int kh = khkw / s.Kw, kw = khkw % s.Kw;
-      const int h = h0 + kh, w = w0 + kw;
-      if (k0s + 3 < K && h >= 0 && h < s.H && w >= 0 && w < s.W) {
-        r = *reinterpret_cast<const float4*>(
-            &x[a_nbase + ((long)h * s.W + w) * s.Cin + ci]);
-      }
-      return r;
-    }
-    float* rr = reinterpret_cast<float*>(&r);
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int kh = khkw / s.Kw, kw = khkw % s.Kw;
-      const int h = h0 + kh, w = w0 + kw;
-      if (k0s + i < K && h >= 0 && h < s.H && w >= 0 && w < s.W) {
-        rr[i] = x[a_nbase + ((long)h * s.W + w) * s.Cin + ci];
-      }
-      if (++ci == s.Cin) { ci = 0; ++khkw; }
-    }
-    return r;
-  };
-  auto ci_adv = [&](int ci, int khkw, int step, int* co, int* ko) {
-    ci += step;
-    while (ci >= s.Cin) { ci -= s.Cin; ++khkw; }
-    *co = ci;
-    *ko = khkw;
-  };
-  auto load_a = [&](int k0, float4* r) {
-    int c1, k1;
-    r[0] = load_a_one(k0 + a_c8, a_ci, a_khkw);
-    ci_adv(a_ci, a_khkw, 4, &c1, &k1);
-    r[1] = load_a_one(k0 + a_c8 + 4, c1, k1);
-  };
-  auto advance_a = [&]() {
-    int c1, k1;
-    ci_adv(a_ci, a_khkw, BK, &c1, &k1);
-    a_ci = c1;
-    a_khkw = k1;
-  };
-  // B staging identical to the BM=64 kernel.
-  const int b_row = t >> 2, b_c4 = (t & 3) * 4;
-  auto load_b = [&](int k0) -> float4 {
-    float4 r;
-    const int gn = n0 + b_row;
-    if (gn < Nc && k0 + BK <= K) {
-      r = *reinterpret_cast<const float4*>(&wgt[(long)gn * K + k0 + b_c4]);
-    } else {
-      float* rr = reinterpret_cast<float*>(&r);
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int gk = k0 + b_c4 + i;
-        rr[i] = (gn < Nc && gk < K) ? wgt[(long)gn * K + gk] : 0.0f;
-      }
-    }
-    return r;
-  };
-
-  const int fr = lane & 15;
-  const int fk = lane >> 4;
-
-  float4 ra[2], rb;
-  load_a(0, ra);
-  rb = load_b(0);
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    {
-      const float* pa0 = reinterpret_cast<const float*>(&ra[0]);
-      const float* pa1 = reinterpret_cast<const float*>(&ra[1]);
-      const float* pb = reinterpret_cast<const float*>(&rb);
-#pragma unroll
-      for (int i = 0; i < 4; ++i) As[a_row][a_c8 + i] = pa0[i];
-#pragma unroll
-      for (int i = 0; i < 4; ++i) As[a_row][a_c8 + 4 + i] = pa1[i];
-#pragma unroll
-      for (int i = 0; i < 4; ++i) Bs[b_c4 + i][b_row] = pb[i];
-    }
-    __syncthreads();
-    if (k0 + BK < K) {
-      advance_a();
-      load_a(k0 + BK, ra);
-      rb = load_b(k0 + BK);
-    }
-#pragma unroll
-    for (int kk = 0; kk < BK / 4; ++kk) {
-      const float b0 = Bs[kk * 4 + fk][wn + fr];
-      const float b1 = Bs[kk * 4 + fk][wn + 16 + fr];
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
-        const float a = As[wm + mi * 16 + fr][kk * 4 + fk];
-        acc[mi][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b0, acc[mi][0], 0, 0, 0);
-        acc[mi][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b1, acc[mi][1], 0, 0, 0);
-      }
-    }
-    __syncthreads();
-  }
-
-  const int fcol = lane & 15;
-  const int frow = (lane >> 4) * 4;
-#pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
-#pragma unroll
-    for (int ni = 0; ni < 2; ++ni) {
-      const int co = n0 + wn + ni * 16 + fcol;
-      if (co >= Nc) continue;
-      const float bv = HAS_BIAS ? bias[co] : 0.0f;
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int m = m0 + wm + mi * 16 + frow + i;
-        if (m < M) {
-          y[(long)m * Nc + co] = act_apply(acc[mi][ni][i] + bv, act);
-        }
-      }
-    }
-  }
 }
