@@ -18,6 +18,8 @@ import torch
 from sparkflow_amd import ops
 
 ACT_NONE, ACT_RELU, ACT_SIGMOID = 0, 1, 2
+ACT_CODES = {"none": ACT_NONE, "linear": ACT_NONE, "relu": ACT_RELU,
+             "sigmoid": ACT_SIGMOID}
 
 _TLS = threading.local()
 
@@ -39,8 +41,6 @@ def grad_only():
         yield
     finally:
         _TLS.grad_only = prev
-ACT_CODES = {"none": ACT_NONE, "linear": ACT_NONE, "relu": ACT_RELU,
-             "sigmoid": ACT_SIGMOID}
 
 
 def _dev_scale(dloss: torch.Tensor) -> torch.Tensor:
