@@ -405,27 +405,34 @@ def test_batchnorm2d_nhwc_gpu():
 
 
 @pytest.mark.timeout(600)
-def test_resnet_nhwc_step_gpu():
-    """One ResNet-18-shape NHWC training step end to end on the HIP
-    kernels."""
+def test_resnet_nhwc_train_converges_gpu():
+    """ResNet-18-shape NHWC training on the HIP kernels must DRIVE THE
+    LOSS DOWN on a small memorizable set (guards against silently wrong
+    conv/BN numerics that stay finite)."""
     from sparkflow_amd.graph_utils import build_graph
     from sparkflow_amd.models import resnet18_shape
     from sparkflow_amd.parallel.ps import OptimizerSpec, PSCore, TrainConfig
     from sparkflow_amd.parallel.transport import LocalTransport
     from sparkflow_amd.parallel.worker import Worker
 
-    graph_json = build_graph(resnet18_shape)
+    graph_json = build_graph(lambda: __import__(
+        "sparkflow_amd.models", fromlist=["cnn"]).cnn.ResNet18Shape(8))
     ps = PSCore(graph_json, OptimizerSpec("adam", 0.001), device=DEV)
+    losses = []
     w = Worker(graph_json, LocalTransport(ps, DEV),
-               TrainConfig(iters=1, mini_batch=8, mini_stochastic_iters=1),
-               device=DEV)
+               TrainConfig(iters=12, mini_batch=8, mini_stochastic_iters=1,
+                           shuffle=False, seed=3),
+               device=DEV,
+               loss_callback=lambda l, it, pid: losses.append(l))
     rng = np.random.default_rng(17)
-    x = rng.random((16, 3 * 224 * 224), dtype=np.float32)
-    y = rng.integers(0, 1000, (16, 1)).astype(np.float32)
+    x = rng.random((8, 3 * 224 * 224), dtype=np.float32)
+    y = rng.integers(0, 8, (8, 1)).astype(np.float32)
+    w.loss_callback = lambda l, it, pid: losses.append(l)
     w.run(x, y)
     torch.cuda.synchronize()
-    assert ps.update_count >= 1
+    assert ps.update_count >= 12
     assert torch.isfinite(ps.weights).all()
+    assert losses[-1] < losses[0] * 0.9, losses
 
 
 @pytest.mark.parametrize("k,s,p,hw,c", [(3, 2, 1, 33, 16), (2, 2, 0, 14, 8)])
