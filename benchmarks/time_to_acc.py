@@ -64,7 +64,7 @@ def main():
     graph_json = build_graph(mnist_mlp)
     cfg = TrainConfig(iters=args.max_rounds * args.round_steps,
                       mini_batch=args.mbs, mini_stochastic_iters=1,
-                      shuffle=False)
+                      shuffle=False, seed=7)  # seeded minibatch stream
     engine = build_engine(graph_json, OptimizerSpec("adam", args.lr), cfg)
     device = engine.device
 
