@@ -79,10 +79,17 @@ def main():
         y = None
     engine.stage_all(x, y)
 
-    # warmup (also initializes the RCCL pair communicators)
+    # warmup (also initializes the RCCL pair communicators).  Runs at
+    # least ~2 s of wall time: a cold GPU takes that long to ramp its
+    # clocks, and a short warmup would leave the first timed steps
+    # measured at low sclk (observed 4x on first-process runs).
+    wt0 = time.perf_counter()
     engine.run_steps(args.warmup, args.mbs)
     if device.type == "cuda":
         torch.cuda.synchronize()
+        while time.perf_counter() - wt0 < 2.0:
+            engine.run_steps(max(1, args.warmup // 4), args.mbs)
+            torch.cuda.synchronize()
     engine.barrier()
     if device.type == "cuda":
         torch.cuda.synchronize()

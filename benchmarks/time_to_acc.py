@@ -82,10 +82,20 @@ def main():
         ey_t = torch.as_tensor(ey, device=device)
         eval_module = engine.worker.module  # weights loaded from PS below
 
-    # a couple of warmup steps to build communicators / graphs
+    # warmup builds communicators/graphs AND holds the GPU busy ~2 s so
+    # the clocks ramp before the timer starts (first-process runs
+    # otherwise measure the ramp, not the training)
+    wt0 = time.perf_counter()
     engine.worker.run_steps(3, args.mbs)
     if device.type == "cuda":
         torch.cuda.synchronize()
+        # clock-ramp burn on DUMMY work (extra training steps here would
+        # corrupt the time-to-accuracy semantics)
+        a = torch.randn(4096, 4096, device=device)
+        while time.perf_counter() - wt0 < 2.0:
+            a = a @ a * 1e-6
+            torch.cuda.synchronize()
+        del a
     engine.barrier()
 
     flag = torch.zeros(1, dtype=torch.int64)
