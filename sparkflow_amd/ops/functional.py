@@ -25,6 +25,10 @@ _TLS = threading.local()
 
 
 def _want_loss_value() -> bool:
+    import os
+
+    if os.environ.get("SPARKFLOW_GRAD_ONLY", "1") == "0":
+        return True  # debug toggle: always compute the scalar
     return not getattr(_TLS, "grad_only", False)
 
 
@@ -93,7 +97,10 @@ class LinearAct(torch.autograd.Function):
         need_dx = ctx.needs_input_grad[0]
         dy = dy.contiguous()
         if ops.use_hip(x):
-            if act != ACT_NONE and ctx.has_bias:
+            import os
+
+            if act != ACT_NONE and ctx.has_bias and \
+                    os.environ.get("SPARKFLOW_FUSED_ACTBWD", "1") != "0":
                 # activation backward fused into BOTH backward GEMMs'
                 # operand loads — no standalone act_bwd pass or dpre buf
                 dx = ops.hip.gemm_dgrad_act(dy, y, w, act) \

@@ -339,9 +339,13 @@ class Worker:
         ps = self.transport.ps if full_step else None
         from sparkflow_amd import ops as _ops
 
+        import os
+
         fused_gather = (sub and _ops.available()
                         and self.device.type == "cuda"
-                        and (yb is None or yb.dtype == torch.float32))
+                        and (yb is None or yb.dtype == torch.float32)
+                        and os.environ.get("SPARKFLOW_FUSED_GATHER",
+                                           "1") != "0")
 
         # the device-pool gather is captured in the graph whenever the
         # fused kernel applies (full-step AND remote compute graphs);
