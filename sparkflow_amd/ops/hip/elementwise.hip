@@ -60,10 +60,15 @@ __global__ void gather_batch_kernel(const float* __restrict__ x,
                                     const long* __restrict__ idx_pool,
                                     long* __restrict__ ctr,
                                     int* __restrict__ done, int mbs, long dx,
-                                    long dy_row, int cap,
+                                    long dy_row, int cap, long n_rows,
                                     const float* __restrict__ wsrc,
                                     float* __restrict__ wdst, long wn) {
-  const long c = *ctr;  // every thread's gather depends on this read
+  // atomic read: a plain load of a location that a racing/previous
+  // launch plain-stored was observed (rarely) returning garbage on
+  // gfx950, which sent `src` gathers to wild addresses
+  long c = (long)atomicAdd(reinterpret_cast<unsigned long long*>(ctr),
+                           0ull);
+  if (c < 0 || c >= cap) c = 0;  // defensive: never index off the pool
   const long* row_idx = idx_pool + c * mbs;
   const bool vec4 = (dx & 3) == 0;
   if (wsrc != nullptr) {
@@ -78,7 +83,8 @@ __global__ void gather_batch_kernel(const float* __restrict__ x,
     for (long i = (w4 << 2) + tid; i < wn; i += stride) wdst[i] = wsrc[i];
   }
   for (int r = blockIdx.x; r < mbs; r += gridDim.x) {
-    const long src = row_idx[r];
+    long src = row_idx[r];
+    if (src < 0 || src >= n_rows) src = 0;  // defensive
     const float* xs = x + src * dx;
     float* xd = xb + (long)r * dx;
     if (vec4) {
@@ -101,9 +107,10 @@ __global__ void gather_batch_kernel(const float* __restrict__ x,
     // ticket, so the kernel replays identically inside a hipGraph.
     // Plain stores suffice: nothing reads ctr/done again until the next
     // launch, and the kernel boundary orders that.
-    if (atomicAdd(done, 1) == gridDim.x - 1) {
-      *ctr = (c + 1) % cap;
-      *done = 0;
+    if (atomicAdd(done, 1) == (int)gridDim.x - 1) {
+      atomicExch(reinterpret_cast<unsigned long long*>(ctr),
+                 (unsigned long long)((c + 1) % cap));
+      atomicExch(done, 0);
     }
   }
 }
@@ -122,6 +129,7 @@ void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
               "gather_batch: idx_pool/ctr int64, done int32");
   const int mbs = xb.size(0);
   const long dx = xb.numel() / mbs;
+  const long n_rows = x.numel() / dx;
   const int cap = idx_pool.size(0);
   TORCH_CHECK(idx_pool.size(1) == mbs, "pool width != mbs");
   const float* yp = nullptr;
@@ -152,8 +160,8 @@ void gather_batch(torch::Tensor x, torch::Tensor xb, torch::Tensor idx_pool,
                      at::cuda::getCurrentCUDAStream(), x.data_ptr<float>(),
                      xb.data_ptr<float>(), yp, ybp,
                      idx_pool.data_ptr<long>(), ctr.data_ptr<long>(),
-                     done.data_ptr<int>(), mbs, dx, dy_row, cap, wsp, wdp,
-                     wn);
+                     done.data_ptr<int>(), mbs, dx, dy_row, cap, n_rows,
+                     wsp, wdp, wn);
 }
 
 // Pack up to 16 parameter-grad tensors into the flat push buffer in ONE
