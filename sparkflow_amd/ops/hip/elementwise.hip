@@ -63,12 +63,20 @@ __global__ void gather_batch_kernel(const float* __restrict__ x,
                                     long dy_row, int cap, long n_rows,
                                     const float* __restrict__ wsrc,
                                     float* __restrict__ wdst, long wn) {
-  // atomic read: a plain load of a location that a racing/previous
-  // launch plain-stored was observed (rarely) returning garbage on
-  // gfx950, which sent `src` gathers to wild addresses
-  long c = (long)atomicAdd(reinterpret_cast<unsigned long long*>(ctr),
-                           0ull);
-  if (c < 0 || c >= cap) c = 0;  // defensive: never index off the pool
+  // One ATOMIC cursor read per block (thread 0), broadcast through LDS:
+  // a plain per-thread load of this cross-launch-stored location was
+  // observed (rarely) returning garbage on gfx950, sending the row
+  // gather to wild addresses — and a per-THREAD atomic read serializes
+  // the whole grid on one cache line.
+  __shared__ long c_sh;
+  if (threadIdx.x == 0) {
+    long cv = (long)atomicAdd(reinterpret_cast<unsigned long long*>(ctr),
+                              0ull);
+    if (cv < 0 || cv >= cap) cv = 0;  // defensive: never index off pool
+    c_sh = cv;
+  }
+  __syncthreads();
+  const long c = c_sh;
   const long* row_idx = idx_pool + c * mbs;
   const bool vec4 = (dx & 3) == 0;
   if (wsrc != nullptr) {
