@@ -76,3 +76,32 @@ def test_load_torch_model_api(tmp_path):
     torch.save(m, str(path))
     model = load_torch_model(str(path))
     assert model.getOrDefault(model.modelWeights)
+
+
+def test_payload_byte_compatible_with_reference_encoding():
+    """Our stopwords payload must match the reference's _to_java encoding
+    EXACTLY (dill -> zlib -> 'b0,b1,...,' string + GUID sentinel,
+    reference pipeline_util.py:109-127), and decode with the reference's
+    load_byte_array logic (34-45)."""
+    import zlib
+
+    import dill
+
+    from sparkflow_amd.spark.pipeline_util import (
+        PY_OBJ_GUID,
+        decode_py_obj,
+        encode_py_obj,
+    )
+
+    obj = {"stage": [1, 2, 3], "name": "payload", "nested": {"k": (4, 5)}}
+    ours = encode_py_obj(obj)
+
+    ref_payload = "".join(
+        str(i) + "," for i in bytearray(zlib.compress(dill.dumps(obj))))
+    assert ours[0] == ref_payload
+    assert ours[1] == PY_OBJ_GUID == "4c1740b00d3c4ff6806a1402321572cb"
+
+    # decode OUR payload with the REFERENCE's algorithm
+    ints = ours[0].split(",")[:-1]
+    assert dill.loads(zlib.decompress(bytes(int(i) for i in ints))) == obj
+    assert decode_py_obj(ours) == obj
