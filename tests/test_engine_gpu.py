@@ -227,3 +227,17 @@ def test_hogwild_tcp_gpu_workers():
     weights = trainer.train([rows[0::2], rows[1::2]])
     assert len(weights) == 4  # 2 layers x (w, b)
     assert all(np.isfinite(w).all() for w in weights)
+
+
+def test_raw_rccl_comm_bootstrap_gpu():
+    """dist_session's raw communicator bootstrap (ncclGetUniqueId /
+    ncclCommInitRank / destroy) works on a real GPU — the full session
+    loops need >1 GPU and first run in the scaling bench, but this
+    catches RCCL linkage/runtime issues early."""
+    from sparkflow_amd import ops
+
+    uid = ops.hip.nccl_unique_id()
+    assert isinstance(uid, bytes) and len(uid) == 128
+    h = ops.hip.nccl_comm_init(uid, 1, 0)  # single-rank communicator
+    assert h >= 0
+    ops.hip.nccl_comm_destroy(h)
