@@ -162,6 +162,15 @@ class EngineHandle:
                 w.transport.close()
         if self.service is not None:
             self.service.join()
+        if self.world > 1 and self.device.type == "cuda":
+            from sparkflow_amd.parallel.transport import destroy_native
+
+            pairs = getattr(self.workers[0].transport, "_pairs", None)
+            if pairs is None and self.service is not None:
+                pairs = self.service.pairs
+            if pairs is not None:
+                torch.cuda.synchronize(self.device)
+                destroy_native(pairs)
 
 
 def build_engine(graph_json: str, optimizer: OptimizerSpec,

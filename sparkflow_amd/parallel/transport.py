@@ -314,6 +314,19 @@ class DistPairs:
                     self.native[(self.rank, i)] = handle
 
 
+def destroy_native(pairs: DistPairs) -> None:
+    """Tear down this rank's raw RCCL communicators (call after every
+    session is closed / service thread joined)."""
+    from sparkflow_amd import ops
+
+    for key, h in list(pairs.native.items()):
+        try:
+            ops.hip.nccl_comm_destroy(h)
+        except Exception:
+            pass
+    pairs.native.clear()
+
+
 class DistTransport(WorkerTransport):
     """Worker side of the RCCL star.
 
@@ -333,6 +346,7 @@ class DistTransport(WorkerTransport):
 
         self.dist = dist
         self.rank = pairs.rank
+        self._pairs = pairs
         self.pg = pairs.payload[(self.rank, widx)]
         self.cg = pairs.ctrl[(self.rank, widx)]
         self.native_comm = pairs.native.get((self.rank, widx))
