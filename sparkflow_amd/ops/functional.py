@@ -47,6 +47,28 @@ def grad_only():
         _TLS.grad_only = prev
 
 
+@contextlib.contextmanager
+def defer_wgrad(records: list):
+    """Inside this context LinearAct.backward appends
+    ``(weight_data_ptr, workspace)`` to ``records`` instead of reducing
+    the wgrad split-K workspace to per-tensor grads (the weight/bias
+    grads it returns are then None).  The captured step feeds the raw
+    workspaces to ``reduce_apply_adam`` — one launch replaces every
+    wgrad reduce, pack_segments, and the Adam apply.  GPU biased-Linear
+    layers only; other ops ignore the context (the caller checks
+    coverage and falls back)."""
+    prev = getattr(_TLS, "defer_wgrad", None)
+    _TLS.defer_wgrad = records
+    try:
+        yield
+    finally:
+        _TLS.defer_wgrad = prev
+
+
+def _defer_records():
+    return getattr(_TLS, "defer_wgrad", None)
+
+
 def _dev_scale(dloss: torch.Tensor) -> torch.Tensor:
     """Upstream loss grad as a device fp32 scalar for the HIP bwd kernels.
 
@@ -88,6 +110,10 @@ class LinearAct(torch.autograd.Function):
         ctx.save_for_backward(x, w, y)
         ctx.act = act
         ctx.has_bias = b is not None
+        # Captured at FORWARD time (defer_wgrad is entered on the thread
+        # running the step; autograd executes backward on its own engine
+        # thread where the thread-local would be invisible).
+        ctx.defer_sink = _defer_records()
         return y
 
     @staticmethod
@@ -99,6 +125,22 @@ class LinearAct(torch.autograd.Function):
         if ops.use_hip(x):
             import os
 
+            recs = ctx.defer_sink
+            if recs is not None and ctx.has_bias:
+                # deferred-reduce mode: hand the raw split-K workspace to
+                # the fused reduce+apply; weight/bias grads stay None
+                if act != ACT_NONE and \
+                        os.environ.get("SPARKFLOW_FUSED_ACTBWD", "1") != "0":
+                    dx = ops.hip.gemm_dgrad_act(dy, y, w, act) \
+                        if need_dx else None
+                    ws = ops.hip.gemm_wgrad_bias_act_ws(x, dy, y, act)
+                else:
+                    dpre = ops.hip.act_bwd(y, dy, act) \
+                        if act != ACT_NONE else dy
+                    dx = ops.hip.gemm_dgrad(dpre, w) if need_dx else None
+                    ws = ops.hip.gemm_wgrad_bias_ws(x, dpre)
+                recs.append((w.data_ptr(), ws))
+                return dx, None, None, None
             if act != ACT_NONE and ctx.has_bias and \
                     os.environ.get("SPARKFLOW_FUSED_ACTBWD", "1") != "0":
                 # activation backward fused into BOTH backward GEMMs'

@@ -13,6 +13,9 @@ torch::Tensor gemm_wgrad(torch::Tensor x, torch::Tensor dy);
 torch::Tensor gemm_wgrad_bias(torch::Tensor x, torch::Tensor dy);
 torch::Tensor gemm_wgrad_bias_act(torch::Tensor x, torch::Tensor dy,
                                   torch::Tensor y, int64_t act);
+torch::Tensor gemm_wgrad_bias_ws(torch::Tensor x, torch::Tensor dy);
+torch::Tensor gemm_wgrad_bias_act_ws(torch::Tensor x, torch::Tensor dy,
+                                     torch::Tensor y, int64_t act);
 torch::Tensor bias_grad(torch::Tensor dy);
 
 // elementwise.hip
@@ -56,6 +59,10 @@ void fused_adam_dev(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
                    torch::Tensor mom, double lr, double decay,
                    double momentum, double eps);
+void reduce_apply_adam(std::vector<torch::Tensor> ws,
+                       std::vector<int64_t> offsets, torch::Tensor p,
+                       torch::Tensor m, torch::Tensor v, torch::Tensor state,
+                       double lr, double b1, double b2, double eps);
 
 // conv.hip
 torch::Tensor conv2d_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
@@ -140,6 +147,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "dWb = [x;ones]^T @ dy (bias grad fused as last row)");
   m.def("gemm_wgrad_bias_act", &gemm_wgrad_bias_act,
         "dWb = [x;ones]^T @ act_grad(y, dy) (act bwd fused into B load)");
+  m.def("gemm_wgrad_bias_ws", &gemm_wgrad_bias_ws,
+        "wgrad+bias split-K workspace, reduce deferred");
+  m.def("gemm_wgrad_bias_act_ws", &gemm_wgrad_bias_act_ws,
+        "wgrad+bias+actbwd split-K workspace, reduce deferred");
   m.def("bias_grad", &bias_grad, "column-sum bias gradient");
   m.def("act_bwd", &act_bwd, "activation backward from output");
   m.def("argmax_rows", &argmax_rows, "rowwise argmax head");
@@ -171,6 +182,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam_dev", &fused_adam_dev,
         "Adam apply with device-resident step state (hipGraph-safe)");
   m.def("fused_rmsprop", &fused_rmsprop, "single-launch RMSProp apply");
+  m.def("reduce_apply_adam", &reduce_apply_adam,
+        "sum per-layer wgrad workspaces + device-state Adam apply in one "
+        "launch (replaces split-K reduces + pack_segments + adam)");
   m.def("conv2d_act", &conv2d_act, "conv2d + bias + activation fwd (direct)");
   m.def("conv_gemm_fwd", &conv_gemm_fwd,
         "implicit-GEMM MFMA conv2d + bias + activation fwd");

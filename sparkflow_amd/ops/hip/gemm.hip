@@ -502,6 +502,52 @@ static torch::Tensor wgrad_bias_impl(torch::Tensor x, torch::Tensor dy,
   return dwb;
 }
 
+// Deferred-reduce variant: return the raw [splitk, Kin+1, Nout] split-K
+// workspace WITHOUT the reduce epilogue.  The fused reduce+Adam apply
+// (optim.hip reduce_apply_adam) consumes the partials of every layer in
+// one launch; a plane of this workspace is exactly the layer's
+// contiguous (weight, bias) slice of the flat buffer.  splitk == 1
+// degenerates to the direct GEMM writing plane 0.
+template <bool DPRE_B>
+static torch::Tensor wgrad_bias_ws_impl(torch::Tensor x, torch::Tensor dy,
+                                        const float* preY, int pre_act) {
+  check_f32(x, "x");
+  check_f32(dy, "dy");
+  const int Mb = x.size(0), Kin = x.size(1), Nout = dy.size(1);
+  TORCH_CHECK(dy.size(0) == Mb, "shape mismatch wgrad_bias_ws");
+  const int M = Kin + 1;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int splitk = pick_splitk(M, Nout, Mb);
+  auto ws = torch::empty({splitk, M, Nout}, x.options());
+  if (splitk > 1) {
+    const int k_chunk = ceil_div(Mb, splitk * BK) * BK;
+    dim3 grid(ceil_div(Nout, BN), ceil_div(M, BM), splitk);
+    hipLaunchKernelGGL(
+        (gemm_f32_kernel<true, false, false, true, true, false, DPRE_B>),
+        grid, dim3(256), 0, stream, x.data_ptr<float>(),
+        dy.data_ptr<float>(), nullptr, ws.data_ptr<float>(), M, Nout, Mb,
+        Kin, Nout, ACT_NONE, k_chunk, preY, pre_act);
+  } else {
+    dim3 grid(ceil_div(Nout, BN), ceil_div(M, BM));
+    hipLaunchKernelGGL(
+        (gemm_f32_kernel<true, false, false, false, true, false, DPRE_B>),
+        grid, dim3(256), 0, stream, x.data_ptr<float>(),
+        dy.data_ptr<float>(), nullptr, ws.data_ptr<float>(), M, Nout, Mb,
+        Kin, Nout, ACT_NONE, Mb, preY, pre_act);
+  }
+  return ws;
+}
+
+torch::Tensor gemm_wgrad_bias_ws(torch::Tensor x, torch::Tensor dy) {
+  return wgrad_bias_ws_impl<false>(x, dy, nullptr, 0);
+}
+
+torch::Tensor gemm_wgrad_bias_act_ws(torch::Tensor x, torch::Tensor dy,
+                                     torch::Tensor y, int64_t act) {
+  check_f32(y, "y");
+  return wgrad_bias_ws_impl<true>(x, dy, y.data_ptr<float>(), (int)act);
+}
+
 torch::Tensor gemm_wgrad_bias(torch::Tensor x, torch::Tensor dy) {
   return wgrad_bias_impl<false>(x, dy, nullptr, 0);
 }

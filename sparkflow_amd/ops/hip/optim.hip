@@ -187,3 +187,107 @@ void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
                      mom.data_ptr<float>(), (float)lr, (float)decay,
                      (float)momentum, (float)eps, n);
 }
+
+// ---------------------------------------------------------------------------
+// Fused split-K-reduce + Adam apply.
+//
+// The MLP step's tail was 6 bandwidth-trivial launches — three wgrad
+// split-K reduces, a scalar reduce, pack_segments, adam — each paying
+// the ~4-7 us launch floor (profiles/mlp_w4_kernels.md).  Because every
+// wgrad workspace is [splits, Kin+1, Nout] whose plane is EXACTLY the
+// layer's contiguous (weight, bias) slice of the flat buffer, one kernel
+// can sum the partials of every layer and feed them straight into the
+// device-state Adam update on the PS weights — the per-tensor grads, the
+// packed flat-grad buffer, and five launches disappear.  The z-loop add
+// order and the Adam math are identical to splitk_reduce_kernel +
+// adam_dev_kernel, so the fused path is numerically identical too.
+#define RA_MAX_SEG 16
+struct RASeg {
+  const float* ws;  // [splits, plane] workspace, plane == end - base
+  long plane;       // elements per split slice
+  long end;         // exclusive flat-buffer end of this segment
+  int splits;
+};
+struct RAArgs {
+  RASeg seg[RA_MAX_SEG];
+  int nseg;
+};
+
+__global__ void reduce_apply_adam_kernel(RAArgs a, float* __restrict__ p,
+                                         float* __restrict__ m,
+                                         float* __restrict__ v,
+                                         float* __restrict__ state, float lr,
+                                         float b1, float b2, float eps) {
+  const float lr_t = state[1];
+  long base = 0;
+  for (int k = 0; k < a.nseg; ++k) {  // uniform: no divergent seg lookup
+    const float* __restrict__ ws = a.seg[k].ws;
+    const long plane = a.seg[k].plane;
+    const int splits = a.seg[k].splits;
+    GS_LOOP(off, plane) {
+      float gi = 0.0f;
+      for (int z = 0; z < splits; ++z) gi += ws[(long)z * plane + off];
+      const long i = base + off;
+      const float mi = b1 * m[i] + (1.0f - b1) * gi;
+      const float vi = b2 * v[i] + (1.0f - b2) * gi * gi;
+      m[i] = mi;
+      v[i] = vi;
+      p[i] -= lr_t * mi / (__fsqrt_rn(vi) + eps);
+    }
+    base = a.seg[k].end;
+  }
+  __syncthreads();  // every thread of this block has consumed lr_t
+  if (threadIdx.x == 0) {
+    int* tick = reinterpret_cast<int*>(state + 2);
+    if (atomicAdd(tick, 1) == (int)gridDim.x - 1) {
+      *tick = 0;
+      const float tdone = state[0] + 1.0f;
+      state[0] = tdone;
+      const float tn = tdone + 1.0f;
+      state[1] = lr * __fsqrt_rn(1.0f - __powf(b2, tn)) /
+                 (1.0f - __powf(b1, tn));
+    }
+  }
+}
+
+// ws[k]: [splits_k, plane_k] fp32 workspaces ordered by flat offset;
+// offsets[k] is segment k's start and segments must tile [0, p.numel())
+// contiguously (the caller derives offsets from the params' positions in
+// the flat buffer and falls back to the unfused path when they don't).
+void reduce_apply_adam(std::vector<torch::Tensor> ws,
+                       std::vector<int64_t> offsets, torch::Tensor p,
+                       torch::Tensor m, torch::Tensor v, torch::Tensor state,
+                       double lr, double b1, double b2, double eps) {
+  const long n = p.numel();
+  check_flat(p, "p", n);
+  check_flat(m, "m", n);
+  check_flat(v, "v", n);
+  TORCH_CHECK(state.numel() >= 3 && state.is_cuda(), "bad adam state");
+  TORCH_CHECK(ws.size() == offsets.size() && !ws.empty() &&
+                  ws.size() <= RA_MAX_SEG,
+              "reduce_apply_adam: 1..16 segments");
+  RAArgs a;
+  a.nseg = (int)ws.size();
+  long expect = 0;
+  for (int k = 0; k < a.nseg; ++k) {
+    const auto& t = ws[k];
+    TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                    t.is_contiguous() && t.dim() >= 2,
+                "bad workspace ", k);
+    const int splits = (int)t.size(0);
+    const long plane = t.numel() / splits;
+    TORCH_CHECK(offsets[k] == expect,
+                "workspaces must tile the flat buffer contiguously");
+    a.seg[k].ws = t.data_ptr<float>();
+    a.seg[k].plane = plane;
+    a.seg[k].splits = splits;
+    expect += plane;
+    a.seg[k].end = expect;
+  }
+  TORCH_CHECK(expect == n, "workspaces must cover the whole flat buffer");
+  hipLaunchKernelGGL(reduce_apply_adam_kernel, dim3(adam_blocks(n)),
+                     dim3(256), 0, at::cuda::getCurrentCUDAStream(), a,
+                     p.data_ptr<float>(), m.data_ptr<float>(),
+                     v.data_ptr<float>(), state.data_ptr<float>(),
+                     (float)lr, (float)b1, (float)b2, (float)eps);
+}

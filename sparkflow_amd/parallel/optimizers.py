@@ -165,6 +165,18 @@ class Adam(FlatOptimizer):
     def state_tensors(self):
         return {"m": self.m, "v": self.v}
 
+    def apply_from_workspaces(self, param, ws_list, offsets):
+        """Fused split-K-reduce + Adam apply: `ws_list` are the raw wgrad
+        workspaces (deferred reduces, ops.functional.defer_wgrad) whose
+        planes tile the flat buffer at `offsets`.  One launch replaces
+        the per-layer reduces, pack_segments, and fused_adam_dev; math
+        and add order are identical (optim.hip reduce_apply_adam)."""
+        from sparkflow_amd import ops
+
+        ops.hip.reduce_apply_adam(ws_list, offsets, param, self.m, self.v,
+                                  self.dev_state, self.lr, self.beta1,
+                                  self.beta2, self.eps)
+
     def native_spec(self):
         if self.dev_state is None:
             return None

@@ -103,7 +103,12 @@ class Worker:
             self._ones_root = torch.ones_like(loss)
         loss.backward(gradient=self._ones_root)
         from sparkflow_amd import ops as _ops
+        from sparkflow_amd.ops.functional import _defer_records
 
+        if _defer_records() is not None:
+            # deferred-reduce mode: grads stay in the raw split-K
+            # workspaces (consumed by the fused reduce+apply launch)
+            return None
         if (self.device.type == "cuda" and _ops.available()
                 and len(self._params) <= 16):
             _ops.hip.pack_segments([p.grad.reshape(-1)
@@ -314,6 +319,7 @@ class Worker:
         return k * mbs
 
     def _capture_graph(self, key, mbs: int, sub: bool, full_step: bool):
+        self._graph_ws = None
         n = self.x.shape[0]
         xb = torch.empty((mbs,) + tuple(self.x.shape[1:]),
                          dtype=self.x.dtype, device=self.device) \
@@ -353,6 +359,34 @@ class Worker:
         # (the remote fallback gathers eagerly in replay_once)
         gather_inside = sub and (fused_gather or full_step)
 
+        # Fused reduce+apply: backward leaves each layer's wgrad in its
+        # raw split-K workspace and ONE launch sums them all and runs the
+        # device-state Adam on the PS weights — the 4 reduce launches,
+        # pack_segments, and the adam launch collapse into it
+        # (numerically identical; optim.hip reduce_apply_adam).  Probed
+        # on the first warmup call: a model with non-Linear params (conv)
+        # or a bias-less layer fails the coverage check and falls back.
+        fused_apply = (full_step and _ops.available()
+                       and self.device.type == "cuda"
+                       and hasattr(ps.optimizer, "apply_from_workspaces")
+                       and getattr(ps.optimizer, "dev_state", None)
+                       is not None
+                       and os.environ.get("SPARKFLOW_FUSED_APPLY",
+                                          "1") != "0")
+
+        def _fused_tail(xb_, yb_):
+            from sparkflow_amd.ops.functional import defer_wgrad
+
+            recs = []
+            with defer_wgrad(recs):
+                self._grad_step(xb_, yb_)
+            recs.sort(key=lambda r: r[0])
+            base = self.flat_params.data_ptr()
+            offs = [(ptr - base) // 4 for ptr, _ in recs]
+            ws_list = [w for _, w in recs]
+            ps.optimizer.apply_from_workspaces(ps.weights, ws_list, offs)
+            return ws_list
+
         def body():
             if full_step and not fused_gather:
                 self.flat_params.copy_(ps.weights)  # the weight pull
@@ -372,6 +406,9 @@ class Worker:
                     if yb is not None:
                         torch.index_select(self.y, 0, row, out=yb)
                     ctr.add_(1).remainder_(self._IDX_POOL_CAP)
+            if fused_apply:
+                self._graph_ws = _fused_tail(xb, yb)
+                return
             self._grad_step(xb, yb)
             if full_step:
                 ps.optimizer.apply(ps.weights, self.flat_grads)
@@ -380,8 +417,17 @@ class Worker:
         s = torch.cuda.Stream(device=self.device)
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
-            for _ in range(3):
-                body()
+            for i in range(3):
+                # first warmup call doubles as the fused-apply probe so
+                # the warmup step count (and with it the minibatch
+                # cursor / Adam t) is identical on both paths
+                try:
+                    body()
+                except RuntimeError:
+                    if not (fused_apply and i == 0):
+                        raise
+                    fused_apply = False
+                    body()
         torch.cuda.current_stream().wait_stream(s)
 
         graph = torch.cuda.CUDAGraph()
@@ -391,6 +437,7 @@ class Worker:
         self._graph_state = {"graph": graph, "xb": xb, "yb": yb,
                              "idx_pool": idx_pool, "ctr": ctr,
                              "full_step": full_step,
+                             "workspaces": getattr(self, "_graph_ws", None),
                              "gather_inside": gather_inside}
         self._graph_key = key
 

@@ -241,3 +241,44 @@ def test_raw_rccl_comm_bootstrap_gpu():
     h = ops.hip.nccl_comm_init(uid, 1, 0)  # single-rank communicator
     assert h >= 0
     ops.hip.nccl_comm_destroy(h)
+
+
+def test_fused_apply_matches_unfused_training():
+    """SPARKFLOW_FUSED_APPLY=1 (reduce+Adam in one launch) trains
+    identically to the unfused reduce->pack->adam path."""
+    import os
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import mnist_mlp
+    from sparkflow_amd.parallel.ps import OptimizerSpec, PSCore, TrainConfig
+    from sparkflow_amd.parallel.transport import LocalTransport
+    from sparkflow_amd.parallel.worker import Worker
+
+    graph_json = build_graph(mnist_mlp)
+    rng = np.random.default_rng(11)
+    x = rng.random((512, 784), dtype=np.float32)
+    y = rng.integers(0, 10, (512, 1)).astype(np.float32)
+
+    def train(fused):
+        os.environ["SPARKFLOW_FUSED_APPLY"] = "1" if fused else "0"
+        try:
+            ps = PSCore(graph_json, OptimizerSpec("adam", 0.001),
+                        device=DEV)
+            if not hasattr(train, "w0"):
+                train.w0 = ps.weights.clone()
+            ps.weights.copy_(train.w0)  # identical start
+            w = Worker(graph_json, LocalTransport(ps, DEV),
+                       TrainConfig(iters=1, seed=11), device=DEV)
+            w.stage(x, y)
+            w.run_steps(20, 128, use_graph=True)
+            torch.cuda.synchronize()
+            used_ws = w._graph_state.get("workspaces") is not None
+            return ps.weights.clone(), used_ws
+        finally:
+            os.environ.pop("SPARKFLOW_FUSED_APPLY", None)
+
+    w_ref, ws_ref = train(False)
+    w_fused, ws_fused = train(True)
+    assert not ws_ref and ws_fused  # fused path actually engaged
+    assert not torch.equal(w_fused, train.w0)  # training moved weights
+    torch.testing.assert_close(w_fused, w_ref, atol=1e-5, rtol=1e-5)

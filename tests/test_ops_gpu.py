@@ -455,3 +455,60 @@ def test_maxpool_nhwc_gpu(k, s, p, hw, c):
 
     for g, cc in zip(run(DEV), run("cpu")):
         torch.testing.assert_close(g, cc, atol=1e-5, rtol=1e-5)
+
+
+def test_reduce_apply_adam_matches_adam_dev():
+    """One fused reduce+apply launch == split-K reduces + fused_adam_dev
+    (same z-order adds, same Adam math -> tight tolerance)."""
+    torch.manual_seed(123)
+    lr, b1, b2, eps = 0.01, 0.9, 0.999, 1e-8
+    planes = [785 * 256, 257 * 256, 257 * 10]
+    splits = [4, 4, 2]
+    ws = [torch.randn(s, p, device=DEV) for s, p in zip(splits, planes)]
+    n = sum(planes)
+    offs = [0, planes[0], planes[0] + planes[1]]
+
+    p0 = torch.randn(n, device=DEV)
+    m0 = torch.zeros(n, device=DEV)
+    v0 = torch.zeros(n, device=DEV)
+
+    def fresh_state():
+        t_next = 1
+        lr_t = lr * (1.0 - b2 ** t_next) ** 0.5 / (1.0 - b1 ** t_next)
+        return torch.tensor([0.0, lr_t, 0.0, 0.0], device=DEV)
+
+    # reference: reduce each workspace on torch, concat, adam_dev
+    g = torch.cat([w.sum(dim=0) for w in ws])
+    p_ref, m_ref, v_ref = p0.clone(), m0.clone(), v0.clone()
+    st_ref = fresh_state()
+    ops.hip.fused_adam_dev(p_ref, g, m_ref, v_ref, st_ref, lr, b1, b2, eps)
+
+    p_f, m_f, v_f = p0.clone(), m0.clone(), v0.clone()
+    st_f = fresh_state()
+    ops.hip.reduce_apply_adam(ws, offs, p_f, m_f, v_f, st_f,
+                              lr, b1, b2, eps)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(p_f, p_ref, atol=1e-6, rtol=1e-6)
+    torch.testing.assert_close(m_f, m_ref, atol=1e-6, rtol=1e-6)
+    torch.testing.assert_close(v_f, v_ref, atol=1e-6, rtol=1e-6)
+    torch.testing.assert_close(st_f, st_ref)
+
+
+def test_reduce_apply_adam_rejects_gaps():
+    with pytest.raises(RuntimeError):
+        ops.hip.reduce_apply_adam(
+            [torch.zeros(1, 100, device=DEV)], [0],
+            torch.zeros(200, device=DEV), torch.zeros(200, device=DEV),
+            torch.zeros(200, device=DEV),
+            torch.zeros(4, device=DEV), 0.01, 0.9, 0.999, 1e-8)
+
+
+def test_gemm_wgrad_bias_ws_planes_sum_to_wgrad():
+    torch.manual_seed(7)
+    x = torch.randn(300, 784, device=DEV)
+    dy = torch.randn(300, 256, device=DEV)
+    ws = ops.hip.gemm_wgrad_bias_ws(x, dy)
+    dwb = ops.hip.gemm_wgrad_bias(x, dy)
+    torch.cuda.synchronize()
+    assert ws.dim() == 3 and ws.shape[1:] == (785, 256)
+    torch.testing.assert_close(ws.sum(dim=0), dwb, atol=1e-4, rtol=1e-4)
