@@ -224,15 +224,42 @@ __global__ void reduce_apply_adam_kernel(RAArgs a, float* __restrict__ p,
     const float* __restrict__ ws = a.seg[k].ws;
     const long plane = a.seg[k].plane;
     const int splits = a.seg[k].splits;
-    GS_LOOP(off, plane) {
-      float gi = 0.0f;
-      for (int z = 0; z < splits; ++z) gi += ws[(long)z * plane + off];
-      const long i = base + off;
-      const float mi = b1 * m[i] + (1.0f - b1) * gi;
-      const float vi = b2 * v[i] + (1.0f - b2) * gi * gi;
-      m[i] = mi;
-      v[i] = vi;
-      p[i] -= lr_t * mi / (__fsqrt_rn(vi) + eps);
+    if (((plane | base) & 3) == 0) {
+      // float4 path: 4x fewer load/RMW chains (the kernel is latency-
+      // bound, not bandwidth-bound, at ~270k elements on 65k threads)
+      const float4* __restrict__ ws4 = reinterpret_cast<const float4*>(ws);
+      float4* __restrict__ p4 = reinterpret_cast<float4*>(p + base);
+      float4* __restrict__ m4 = reinterpret_cast<float4*>(m + base);
+      float4* __restrict__ v4 = reinterpret_cast<float4*>(v + base);
+      const long plane4 = plane >> 2;
+      GS_LOOP(off, plane4) {
+        float4 g = {0.0f, 0.0f, 0.0f, 0.0f};
+        for (int z = 0; z < splits; ++z) {
+          const float4 w = ws4[(long)z * plane4 + off];
+          g.x += w.x; g.y += w.y; g.z += w.z; g.w += w.w;
+        }
+        float4 mi = m4[off], vi = v4[off], pi = p4[off];
+#define RA_ADAM1(c)                                                           \
+  mi.c = b1 * mi.c + (1.0f - b1) * g.c;                                       \
+  vi.c = b2 * vi.c + (1.0f - b2) * g.c * g.c;                                 \
+  pi.c -= lr_t * mi.c / (__fsqrt_rn(vi.c) + eps);
+        RA_ADAM1(x) RA_ADAM1(y) RA_ADAM1(z) RA_ADAM1(w)
+#undef RA_ADAM1
+        m4[off] = mi;
+        v4[off] = vi;
+        p4[off] = pi;
+      }
+    } else {
+      GS_LOOP(off, plane) {
+        float gi = 0.0f;
+        for (int z = 0; z < splits; ++z) gi += ws[(long)z * plane + off];
+        const long i = base + off;
+        const float mi = b1 * m[i] + (1.0f - b1) * gi;
+        const float vi = b2 * v[i] + (1.0f - b2) * gi * gi;
+        m[i] = mi;
+        v[i] = vi;
+        p[i] -= lr_t * mi / (__fsqrt_rn(vi) + eps);
+      }
     }
     base = a.seg[k].end;
   }
