@@ -135,6 +135,7 @@ void worker_session(int64_t comm_h, int64_t peer, torch::Tensor w_flat,
 void raw_send(int64_t comm_h, int64_t peer, torch::Tensor t);
 void raw_recv(int64_t comm_h, int64_t peer, torch::Tensor t);
 void multi_replay(std::vector<int64_t> execs, int64_t k);
+void replay_streams_init(int64_t n);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sparkflow_amd gfx950 (CDNA4) kernels";
@@ -224,6 +225,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "GIL-free k-step worker loop: recv weights / graph launch / send");
   m.def("raw_send", &raw_send, "one fp32 send on a raw comm");
   m.def("raw_recv", &raw_recv, "one fp32 recv on a raw comm");
+  m.def("replay_streams_init", &replay_streams_init,
+        "pre-create dedicated replay streams (deterministic HW-queue map)");
   m.def("multi_replay", &multi_replay,
         "k round-robin launches of W captured step graphs on W streams");
 }

@@ -185,6 +185,14 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
     the rank's RCCL pair link (driven sequentially by one host thread)."""
     rank, world, device = init_from_env(device=device)
 
+    if device.type == "cuda":
+        from sparkflow_amd import ops as _ops
+
+        if _ops.available():
+            # create the replay streams FIRST so they land on distinct
+            # HW queues (see dist_session.hip replay_streams_init)
+            _ops.hip.replay_streams_init(max(1, workers_per_gpu))
+
     ps = service = None
     if world > 1:
         pairs = DistPairs(workers_per_gpu=max(1, workers_per_gpu))
