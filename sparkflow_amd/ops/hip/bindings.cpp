@@ -136,6 +136,7 @@ void raw_send(int64_t comm_h, int64_t peer, torch::Tensor t);
 void raw_recv(int64_t comm_h, int64_t peer, torch::Tensor t);
 void multi_replay(std::vector<int64_t> execs, int64_t k);
 void replay_streams_init(int64_t n);
+void replay_streams_reroll();
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "sparkflow_amd gfx950 (CDNA4) kernels";
@@ -227,6 +228,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("raw_recv", &raw_recv, "one fp32 recv on a raw comm");
   m.def("replay_streams_init", &replay_streams_init,
         "pre-create dedicated replay streams (deterministic HW-queue map)");
+  m.def("replay_streams_reroll", &replay_streams_reroll,
+        "re-roll replay stream HW-queue placement (device must be idle)");
   m.def("multi_replay", &multi_replay,
         "k round-robin launches of W captured step graphs on W streams");
 }

@@ -192,6 +192,24 @@ void replay_streams_init(int64_t n) {
   }
 }
 
+void replay_streams_reroll() {
+  // Re-roll the replay streams' hardware-queue placement: create the
+  // fresh set FIRST (advancing the runtime's queue assignment) and only
+  // then destroy the old one.  Caller must have synchronized the device
+  // (no replays in flight on the old streams).
+  std::lock_guard<std::mutex> lk(g_rs_mu);
+  const size_t n = g_replay_streams.size();
+  std::vector<hipStream_t> fresh;
+  fresh.reserve(n);
+  for (size_t i = 0; i < n; ++i) {
+    hipStream_t s;
+    HIP_OK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+    fresh.push_back(s);
+  }
+  for (hipStream_t s : g_replay_streams) HIP_OK(hipStreamDestroy(s));
+  g_replay_streams = std::move(fresh);
+}
+
 void multi_replay(std::vector<int64_t> execs, int64_t k) {
   // W full-step graphs round-robin onto W dedicated streams: the whole
   // multi-worker Hogwild loop is enqueued from one GIL-free C++ call.

@@ -322,7 +322,23 @@ int sf_splitk_env_max() {
 // stream contention makes each GEMM slower without removing the reduce
 // launch, so the target stays at 512 regardless of the hint; the env cap
 // remains for experiments.
+// Small-output direct mode: once the wgrad reduces are fused into the
+// optimizer apply (reduce_apply_adam), the per-GEMM split-K reduce
+// launch (~5 us) costs more than the split buys for small-M*N,
+// shallow-K outputs (the 300x256 dgrad / L2-L3 fwd shapes) — below the
+// threshold the GEMM goes direct and its reduce launch disappears.
+static long sf_splitk_direct_mn() {
+  static long cached = -1;
+  if (cached < 0) {
+    const char* e = getenv("SPARKFLOW_SPLITK_DIRECT_MN");
+    cached = e ? atol(e) : 0;  // default off until measured
+    if (cached < 0) cached = 0;
+  }
+  return cached;
+}
+
 static int pick_splitk(int M, int N, int K) {
+  if ((long)M * N <= sf_splitk_direct_mn() && K <= 256) return 1;
   const int base = ceil_div(M, BM) * ceil_div(N, BN);
   const int max_by_k = std::max(1, K / 64);  // keep >=64-deep slices
   const int cap = sf_splitk_env_max();
