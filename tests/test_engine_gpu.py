@@ -282,3 +282,32 @@ def test_fused_apply_matches_unfused_training():
     assert not ws_ref and ws_fused  # fused path actually engaged
     assert not torch.equal(w_fused, train.w0)  # training moved weights
     torch.testing.assert_close(w_fused, w_ref, atol=1e-5, rtol=1e-5)
+
+
+def test_replay_streams_reroll_mid_training():
+    """replay_streams_reroll (bench's warmup health probe) must be safe
+    between run_steps calls: graphs are stream-agnostic, so training
+    continues on the fresh streams."""
+    from sparkflow_amd import ops
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import mnist_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    eng = build_engine(build_graph(mnist_mlp),
+                       OptimizerSpec("adam", 0.001),
+                       TrainConfig(iters=1, seed=13), device=DEV,
+                       workers_per_gpu=4)
+    rng = np.random.default_rng(13)
+    eng.stage_all(rng.random((600, 784), dtype=np.float32),
+                  rng.integers(0, 10, (600, 1)).astype(np.float32))
+    eng.run_steps(20, 150)
+    torch.cuda.synchronize()
+    w_before = eng.ps.weights.clone()
+    for _ in range(3):
+        ops.hip.replay_streams_reroll()
+        eng.run_steps(20, 150)
+        torch.cuda.synchronize()
+    assert torch.isfinite(eng.ps.weights).all()
+    assert not torch.equal(eng.ps.weights, w_before)
+    assert eng.ps.update_count == 4 * 20 * 4  # 4 calls x 20 steps x 4 workers
