@@ -119,18 +119,14 @@ def predict_batch(module: torch.nn.Module, x: np.ndarray,
                   keep_dropout: bool = False) -> np.ndarray:
     """One batched inference forward on the best device.
 
-    ``keep_dropout`` preserves the reference's ``to_keep_dropout``
-    semantics (ml_util.py:70-71: the dropout feed keeps its TRAINING
-    keep-prob at inference instead of 1.0): Dropout modules stay in
-    train mode while everything else runs eval."""
+    ``keep_dropout`` mirrors the reference's ``to_keep_dropout`` flag
+    (ml_util.py:70-71) which only selects WHICH constant is fed to the
+    dropout placeholder — ``1.0`` read as keep-prob when True, ``0.0``
+    read as drop-rate when False.  Dropout is inert at prediction under
+    both values, so the flag is accepted for signature parity but never
+    activates Dropout modules: everything runs in eval mode."""
     device = device or _best_device()
     module = module.to(device).eval()
-    if keep_dropout:
-        from sparkflow_amd.nn.modules import Dropout
-
-        for m in module.modules():
-            if isinstance(m, (Dropout, torch.nn.Dropout)):
-                m.train()
     with torch.no_grad():
         out = module(torch.as_tensor(x, dtype=torch.float32, device=device))
     return out.detach().cpu().numpy()
@@ -143,10 +139,10 @@ def predict_func(rows, graph_json, prediction, graph_weights, inp,
     ml_util.predict_func, ml_util.py:54-83).
 
     Differences by design: the module is rebuilt once per partition and the
-    whole partition runs as ONE batched forward on the GPU; dropout
-    follows the reference keep/drop semantics (``to_keep_dropout=True``
-    keeps Dropout modules active at inference, ml_util.py:70-71;
-    otherwise eval mode makes them inert).  Scalar outputs become floats,
+    whole partition runs as ONE batched forward on the GPU; dropout is
+    inert at prediction for both ``to_keep_dropout`` values, exactly as
+    in the reference (ml_util.py:70-71 feeds 1.0-as-keep-prob or
+    0.0-as-drop-rate — both disable it).  Scalar outputs become floats,
     vector outputs become ``pyspark.ml.linalg.Vectors.dense`` — identical
     row-shape contract.
     """

@@ -124,7 +124,8 @@ std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
 
 // dist_session.hip — native RCCL session loops (GIL-free)
 py::bytes nccl_unique_id();
-int64_t nccl_comm_init(py::bytes id_bytes, int64_t nranks, int64_t rank);
+int64_t nccl_comm_init(py::bytes id_bytes, int64_t nranks, int64_t rank,
+                       int64_t timeout_ms);
 void nccl_comm_destroy(int64_t h);
 void ps_session(int64_t comm_h, int64_t peer, torch::Tensor weights,
                 torch::Tensor grad_buf, int64_t k, int64_t opt_id,
@@ -218,7 +219,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn2d_nhwc_bwd", &bn2d_nhwc_bwd, "NHWC BatchNorm2d bwd");
   m.def("nccl_unique_id", &nccl_unique_id, "RCCL unique id for a pair comm");
   m.def("nccl_comm_init", &nccl_comm_init,
-        "init this rank's side of a raw RCCL communicator");
+        "init this rank's side of a raw RCCL communicator (non-blocking\n"
+        "rendezvous with a bounded wait; aborts the comm on timeout)",
+        py::arg("id_bytes"), py::arg("nranks"), py::arg("rank"),
+        py::arg("timeout_ms") = 30000);
   m.def("nccl_comm_destroy", &nccl_comm_destroy, "destroy a raw comm");
   m.def("ps_session", &ps_session,
         "GIL-free k-step PS loop: send weights / recv grads / fused apply");

@@ -26,7 +26,9 @@
 #include <c10/cuda/CUDAGuard.h>
 #include <rccl/rccl.h>
 
+#include <chrono>
 #include <mutex>
+#include <thread>
 #include <vector>
 
 #include "common.h"
@@ -58,17 +60,45 @@ py::bytes nccl_unique_id() {
   return py::bytes(reinterpret_cast<const char*>(&id), sizeof(id));
 }
 
-int64_t nccl_comm_init(py::bytes id_bytes, int64_t nranks, int64_t rank) {
+int64_t nccl_comm_init(py::bytes id_bytes, int64_t nranks, int64_t rank,
+                       int64_t timeout_ms) {
   std::string s = id_bytes;
   TORCH_CHECK(s.size() == sizeof(ncclUniqueId), "bad unique id size");
   ncclUniqueId id;
   memcpy(&id, s.data(), sizeof(id));
   ncclComm_t comm = nullptr;
+  ncclResult_t st = ncclSuccess;
   {
-    // init rendezvouses over TCP with the peer — release the GIL
+    // Non-blocking init + bounded poll: ncclCommInitRank is a
+    // collective rendezvous, so if the PEER dies before (or while)
+    // calling it, a blocking init would hang this thread forever and
+    // the ok/peer_ok handshake in DistPairs.init_native would never be
+    // reached (the gloo recv after it would deadlock the engine
+    // build).  With blocking=0 we poll ncclCommGetAsyncError and abort
+    // the half-open comm on timeout; the caller then takes the torch
+    // pair fallback and the handshake still completes on both sides.
     py::gil_scoped_release rel;
-    NCCL_OK(ncclCommInitRank(&comm, (int)nranks, id, (int)rank));
+    ncclConfig_t cfg = NCCL_CONFIG_INITIALIZER;
+    cfg.blocking = 0;
+    st = ncclCommInitRankConfig(&comm, (int)nranks, id, (int)rank, &cfg);
+    if (st == ncclSuccess || st == ncclInProgress) {
+      auto dl = std::chrono::steady_clock::now() +
+                std::chrono::milliseconds(timeout_ms > 0 ? timeout_ms
+                                                         : 30000);
+      do {
+        ncclResult_t as = ncclSuccess;
+        st = ncclCommGetAsyncError(comm, &as) == ncclSuccess
+                 ? as
+                 : ncclUnhandledCudaError;
+        if (st != ncclInProgress) break;
+        std::this_thread::sleep_for(std::chrono::milliseconds(2));
+      } while (std::chrono::steady_clock::now() < dl);
+    }
+    if (st != ncclSuccess && comm) ncclCommAbort(comm);
   }
+  TORCH_CHECK(st == ncclSuccess, "raw RCCL comm init failed/timed out: ",
+              st == ncclInProgress ? "timeout waiting for peer"
+                                   : ncclGetErrorString(st));
   std::lock_guard<std::mutex> lk(g_comms_mu);
   g_comms.push_back(comm);
   return (int64_t)g_comms.size() - 1;

@@ -76,9 +76,14 @@ __global__ void adam_dev_kernel(float* __restrict__ p,
   if (threadIdx.x == 0) {
     int* tick = reinterpret_cast<int*>(state + 2);
     if (atomicAdd(tick, 1) == (int)gridDim.x - 1) {
-      *tick = 0;
-      const float tdone = state[0] + 1.0f;  // this step's index
-      state[0] = tdone;
+      // atomicSub (not a plain store) so increments from a CONCURRENT
+      // Hogwild apply sharing this dev_state are never wiped: each +1
+      // is preserved and every crossing of gridDim.x-1 fires exactly
+      // one ticket, so t counts every apply.  atomicAdd on t for the
+      // same reason; the lr_t look-ahead write stays a benign
+      // last-writer race (one step of staleness at most).
+      atomicSub(tick, (int)gridDim.x);
+      const float tdone = atomicAdd(state, 1.0f) + 1.0f;  // this step
       const float tn = tdone + 1.0f;
       state[1] = lr * __fsqrt_rn(1.0f - __powf(b2, tn)) /
                  (1.0f - __powf(b1, tn));
@@ -267,9 +272,10 @@ __global__ void reduce_apply_adam_kernel(RAArgs a, float* __restrict__ p,
   if (threadIdx.x == 0) {
     int* tick = reinterpret_cast<int*>(state + 2);
     if (atomicAdd(tick, 1) == (int)gridDim.x - 1) {
-      *tick = 0;
-      const float tdone = state[0] + 1.0f;
-      state[0] = tdone;
+      // see adam_dev_kernel: atomicSub/atomicAdd so concurrent Hogwild
+      // applies sharing one dev_state never lose a t increment
+      atomicSub(tick, (int)gridDim.x);
+      const float tdone = atomicAdd(state, 1.0f) + 1.0f;
       const float tn = tdone + 1.0f;
       state[1] = lr * __fsqrt_rn(1.0f - __powf(b2, tn)) /
                  (1.0f - __powf(b1, tn));

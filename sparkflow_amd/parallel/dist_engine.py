@@ -113,6 +113,7 @@ class EngineHandle:
 
             for w in self.workers:
                 w.ensure_graph(mini_batch)
+                w.maybe_refresh_pool(k)
             # All-local full-step graphs + extension: the whole W x k
             # Hogwild loop is one GIL-free C++ call (round-robin graph
             # launches onto W pool streams).

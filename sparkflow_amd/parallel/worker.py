@@ -246,16 +246,39 @@ class Worker:
         if getattr(self, "_graph_key", None) != key:
             self._capture_graph(key, mbs, sub, full_step)
             if sub:
-                # Fill the pool ONCE with pre-drawn sample-without-
-                # replacement minibatches; the captured ctr walks it
-                # cyclically.  Keeping the fill out of run_steps keeps
-                # host-side RNG out of the steady-state loop.
-                g = self._graph_state
-                picks = np.stack([self.rng.choice(n, mbs, replace=False)
-                                  for _ in range(self._IDX_POOL_CAP)])
-                g["idx_pool"].copy_(torch.as_tensor(picks,
-                                                    device=self.device))
-                g["ctr"].zero_()
+                self._fill_idx_pool(n, mbs)
+
+    def _fill_idx_pool(self, n: int, mbs: int) -> None:
+        """(Re)draw the device-side pool of sample-without-replacement
+        minibatches and rewind the captured cursor.  Keeping the fill out
+        of the steady-state loop keeps host RNG off the hot path; the
+        pool is REDRAWN once its minibatches have all been consumed (see
+        maybe_refresh_pool) so long runs don't cycle the same
+        _IDX_POOL_CAP batches forever."""
+        g = self._graph_state
+        picks = np.stack([self.rng.choice(n, mbs, replace=False)
+                          for _ in range(self._IDX_POOL_CAP)])
+        g["idx_pool"].copy_(torch.as_tensor(picks, device=self.device))
+        g["ctr"].zero_()
+        self._pool_used = 0
+
+    def maybe_refresh_pool(self, k: int) -> None:
+        """Called before enqueueing k replays: if the pool would wrap,
+        wait for in-flight replays and redraw it.  The device sync is
+        once per _IDX_POOL_CAP steps (~55 ms of GPU work) so the cost is
+        noise; callers must invoke this while no replay is being
+        enqueued concurrently (all run paths do)."""
+        g = getattr(self, "_graph_state", None)
+        if not g or g.get("idx_pool") is None:
+            return
+        used = getattr(self, "_pool_used", 0)
+        if used + k > self._IDX_POOL_CAP:
+            if self.device.type == "cuda":
+                torch.cuda.synchronize(self.device)
+            self._fill_idx_pool(self.x.shape[0], g["idx_pool"].shape[1])
+            if self.device.type == "cuda":
+                torch.cuda.synchronize(self.device)
+        self._pool_used = getattr(self, "_pool_used", 0) + k
 
     def replay_once(self) -> None:
         """Enqueue ONE training step on the current HIP stream (graph must
@@ -298,6 +321,7 @@ class Worker:
         eager.
         """
         self.ensure_graph(mbs)
+        self.maybe_refresh_pool(k)
         g = self._graph_state
         if not g["full_step"] and hasattr(self.transport, "run_session") \
                 and (g["gather_inside"] or g["xb"] is self.x) \

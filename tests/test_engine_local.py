@@ -183,3 +183,44 @@ def test_multi_worker_engine_cpu():
     assert eng.workers[0].rng.integers(0, 1 << 30) != \
         eng.workers[1].rng.integers(0, 1 << 30)
     eng.close()
+
+
+def test_idx_pool_refreshes_on_wrap():
+    """The device-side minibatch index pool is REDRAWN once its
+    _IDX_POOL_CAP pre-drawn minibatches are consumed, so long runs do
+    not cycle the same batches forever (round-1 advisor finding)."""
+    import torch
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import mnist_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    cfg = TrainConfig(iters=1, mini_batch=8, mini_stochastic_iters=1,
+                      seed=3)
+    eng = build_engine(build_graph(mnist_mlp),
+                       OptimizerSpec("sgd", 0.01), cfg, device="cpu")
+    w = eng.worker
+    rng = np.random.default_rng(0)
+    w.stage(rng.random((64, 784), dtype=np.float32),
+            rng.integers(0, 10, (64, 1)).astype(np.float32))
+    cap = w._IDX_POOL_CAP
+    # fabricate the graph-capture state (graphs are GPU-only; the pool
+    # bookkeeping is device-independent)
+    w._graph_state = {
+        "idx_pool": torch.zeros(cap, 8, dtype=torch.int64),
+        "ctr": torch.zeros(1, dtype=torch.int64),
+    }
+    w._fill_idx_pool(64, 8)
+    p0 = w._graph_state["idx_pool"].clone()
+    w.maybe_refresh_pool(100)
+    assert w._pool_used == 100
+    torch.testing.assert_close(w._graph_state["idx_pool"], p0)  # no wrap
+    w._graph_state["ctr"].fill_(37)
+    w.maybe_refresh_pool(cap)  # would wrap -> redraw + rewind
+    assert not torch.equal(w._graph_state["idx_pool"], p0)
+    assert int(w._graph_state["ctr"].item()) == 0
+    # every redrawn row is still a sample-without-replacement minibatch
+    row = w._graph_state["idx_pool"][0]
+    assert len(set(row.tolist())) == 8 and int(row.max()) < 64
+    eng.close()
