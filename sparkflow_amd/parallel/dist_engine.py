@@ -157,12 +157,24 @@ class EngineHandle:
         for w in self.workers:
             w.stage(x, y)
 
+    def check_failure(self):
+        """Surface a PS-service failure ("Too many failures during
+        training") to the rank-0 caller — reference parity
+        (HogwildSparkModel.py:231-237); the service itself keeps
+        draining the wire so no worker rank ever blocks."""
+        if self.service is not None:
+            self.service.check()
+
     def close(self):
         if self.rank != 0:
             for w in self.workers:
                 w.transport.close()
+        failure = None
         if self.service is not None:
-            self.service.join()
+            try:
+                self.service.join()  # raises the first service failure
+            except Exception as e:
+                failure = e
         if self.world > 1 and self.device.type == "cuda":
             from sparkflow_amd.parallel.transport import destroy_native
 
@@ -172,6 +184,8 @@ class EngineHandle:
             if pairs is not None:
                 torch.cuda.synchronize(self.device)
                 destroy_native(pairs)
+        if failure is not None:
+            raise failure
 
 
 def build_engine(graph_json: str, optimizer: OptimizerSpec,

@@ -443,11 +443,34 @@ class DistPSService:
         self.ps = ps_core
         self.pairs = pairs
         self._threads = []
+        # First fatal error from any service thread (e.g. "Too many
+        # failures during training" past the PS error budget).  The
+        # thread that hits it records it here and switches to DRAIN mode
+        # — it keeps answering the wire protocol (pulls get the current
+        # weights, pushed grads are received and dropped) so no peer is
+        # ever left blocked in a recv — and the engine re-raises it to
+        # the train caller (reference HogwildSparkModel.py:231-237
+        # surfaces the same message).
+        self.failure: Optional[Exception] = None
+        self._fail_lock = threading.Lock()
+
+    def _record_failure(self, e: Exception) -> None:
+        with self._fail_lock:
+            if self.failure is None:
+                self.failure = e
+        print(f"[sparkflow_amd] PS service error (serving continues in "
+              f"drain mode): {e}")
+
+    def check(self) -> None:
+        """Raise the first recorded service failure (call from the train
+        driver on rank 0)."""
+        if self.failure is not None:
+            raise self.failure
 
     def start(self):
         for w in range(1, self.pairs.world):
             for i in range(self.pairs.workers_per_gpu):
-                t = threading.Thread(target=self._serve, args=(w, i),
+                t = threading.Thread(target=self._serve_guarded, args=(w, i),
                                      daemon=True)
                 t.start()
                 self._threads.append(t)
@@ -464,6 +487,18 @@ class DistPSService:
         grad_buf = torch.empty(self.ps.numel, dtype=torch.float32, device=dev)
         ctrl = torch.zeros(3, dtype=torch.int64)
 
+        def apply(buf):
+            # Past the error budget apply_gradients raises; record the
+            # failure ONCE and drop further gradients (drain mode) so
+            # the protocol position with the peer never shifts and no
+            # worker blocks in a recv.
+            if self.failure is not None:
+                return
+            try:
+                self.ps.apply_gradients(buf)
+            except Exception as e:
+                self._record_failure(e)
+
         def handle(op: int, native: bool = False):
             native = native and native_comm is not None
             if native:
@@ -474,14 +509,14 @@ class DistPSService:
                         lambda buf: ops.hip.raw_send(native_comm, 1, buf))
                 elif op == OP_PUSH:
                     ops.hip.raw_recv(native_comm, 1, grad_buf)
-                    self.ps.apply_gradients(grad_buf)
+                    apply(grad_buf)
                 return
             if op == OP_PULL:
                 self.ps.send_weights(
                     lambda buf: dist.send(buf, dst=w, group=pg))
             elif op == OP_PUSH:
                 dist.recv(grad_buf, src=w, group=pg)
-                self.ps.apply_gradients(grad_buf)
+                apply(grad_buf)
 
         def run_session(k: int, native: bool):
             # k-step session: {send weights, recv grads, apply} x k with
@@ -490,13 +525,24 @@ class DistPSService:
             # C++ with the GIL released (ops/hip/dist_session.hip);
             # otherwise per-step ops from Python, still session-paced.
             spec = self.ps.optimizer.native_spec() if use_cuda else None
-            if native and spec is not None and not self.ps.acquire_lock:
+            if native and spec is not None and not self.ps.acquire_lock \
+                    and self.failure is None:
                 from sparkflow_amd import ops
 
                 opt_id, state, lr, h1, h2, h3 = spec
-                ops.hip.ps_session(native_comm, 1, self.ps.weights,
-                                   grad_buf, k, opt_id, state, lr, h1, h2,
-                                   h3)
+                try:
+                    ops.hip.ps_session(native_comm, 1, self.ps.weights,
+                                       grad_buf, k, opt_id, state, lr, h1,
+                                       h2, h3)
+                except Exception as e:
+                    # ps_session validates args BEFORE any wire traffic,
+                    # so a raise here means zero steps ran: record and
+                    # serve the whole session per-step (drain applies).
+                    self._record_failure(e)
+                    for _ in range(k):
+                        handle(OP_PULL, native)
+                        handle(OP_PUSH, native)
+                    return
                 self.ps.pull_count += k
                 self.ps.update_count += k
                 return
@@ -525,6 +571,15 @@ class DistPSService:
             else:
                 handle(op, per_op_native)
 
+    def _serve_guarded(self, w: int, widx: int = 0):
+        try:
+            self._serve(w, widx)
+        except Exception as e:
+            # Transport-level death (peer gone, comm torn down): nothing
+            # left to serve, but the caller must still see the error.
+            self._record_failure(e)
+
     def join(self):
         for t in self._threads:
             t.join()
+        self.check()

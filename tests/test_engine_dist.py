@@ -93,7 +93,7 @@ def _dist_lock_worker(rank, world, port):
     torch.distributed.destroy_process_group()
 
 
-def _session_worker(rank, world, port, k):
+def _session_worker(rank, world, port, k, wpg=2):
     """OP_RUN session protocol: k steps of {recv weights, send grads,
     apply} with one control message (the multi-GPU bench hot path)."""
     os.environ["RANK"] = str(rank)
@@ -111,7 +111,7 @@ def _session_worker(rank, world, port, k):
     cfg = TrainConfig(iters=1, mini_batch=16, mini_stochastic_iters=1,
                       seed=4)
     engine = build_engine(graph_json, OptimizerSpec("gradient_descent", 0.01),
-                          cfg, device="cpu", workers_per_gpu=2)
+                          cfg, device="cpu", workers_per_gpu=wpg)
     rng = np.random.default_rng(rank)
     x = rng.normal(0, 1, (64, 10)).astype(np.float32)
     y = (rng.random((64, 1)) > 0.5).astype(np.float32)
@@ -123,7 +123,7 @@ def _session_worker(rank, world, port, k):
         # all updates: 2 local workers x k + (world-1) x 2 workers x k
         import time
 
-        expected = 2 * k * world
+        expected = wpg * k * world
         deadline = time.time() + 30
         while engine.ps.update_count < expected and time.time() < deadline:
             time.sleep(0.05)
@@ -146,3 +146,64 @@ def _session_worker(rank, world, port, k):
 def test_dist_session_protocol_gloo():
     port = free_port()
     mp.spawn(_session_worker, args=(2, port, 3), nprocs=2, join=True)
+
+
+def _failure_worker(rank, world, port):
+    """Past the PS error budget the service must (a) NOT wedge the star —
+    the worker's pulls keep being answered, its pushes are drained — and
+    (b) surface "Too many failures during training" to the rank-0 caller
+    (reference HogwildSparkModel.py:231-237; round-1 verdict weak #3)."""
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["LOCAL_RANK"] = str(rank)
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import gaussian_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    graph_json = build_graph(gaussian_mlp)
+    cfg = TrainConfig(iters=3, mini_batch=16, mini_stochastic_iters=1)
+    engine = build_engine(graph_json, OptimizerSpec("gradient_descent", 0.01),
+                          cfg, device="cpu")
+    if rank == 0:
+        def boom(*a, **k):
+            raise RuntimeError("injected apply failure")
+
+        engine.ps.optimizer.apply = boom
+        engine.ps.max_errors = 1  # first failure exhausts the budget
+    engine.barrier()
+    if rank != 0:
+        rng = np.random.default_rng(7)
+        x = rng.normal(size=(48, 10)).astype(np.float32)
+        y = rng.random((48, 1)).astype(np.float32)
+        engine.worker.run(x, y)  # must complete, not hang on a recv
+        engine.close()
+    else:
+        got = None
+        try:
+            engine.close()  # joins the service threads -> re-raises
+        except Exception as e:
+            got = e
+        assert got is not None and "Too many failures" in str(got), got
+        assert engine.service.failure is not None
+        assert engine.ps.update_count == 0  # every push was drained
+    engine.barrier()
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_dist_failure_drains_not_wedges():
+    port = free_port()
+    mp.spawn(_failure_worker, args=(2, port), nprocs=2, join=True)
+
+
+@pytest.mark.timeout(600)
+def test_dist_session_soak_world8():
+    """World=8 x 4 workers/rank session-protocol soak over gloo — the
+    shape of the driver's 8-GPU scaling bench (star contention: 7 ranks
+    x 4 sessions into rank 0) rehearsed on CPU (round-1 verdict #3b)."""
+    port = free_port()
+    mp.spawn(_session_worker, args=(8, port, 3, 4), nprocs=8, join=True)
