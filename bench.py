@@ -42,7 +42,21 @@ def main():
     ap.add_argument("--workers", type=int, default=4,
                     help="concurrent Hogwild workers per GPU (reference "
                     "canonical config: 4 partitions)")
+    ap.add_argument("--remote-workers", type=int, default=None,
+                    help="workers per NON-PS rank (default: --workers). "
+                    "Star-contention knob for world=8: the PS GPU serves "
+                    "(world-1) x this many sessions, so fewer remote "
+                    "workers trades per-rank throughput for PS headroom")
+    ap.add_argument("--grad-accum", type=int, default=1,
+                    help="worker-side gradient accumulation: push the "
+                    "MEAN of this many minibatch grads per PS round "
+                    "(cuts star traffic by the same factor; 1 = "
+                    "reference semantics).  Remote/TCP transports only")
     args = ap.parse_args()
+    # third knob, env-based because BOTH ends of every RCCL pair must
+    # agree: SPARKFLOW_SESSION_CHUNK_ELEMS=<n> splits session payload
+    # transfers into n-element chunks so concurrent pair transfers
+    # interleave at the PS GPU (see ops/hip/dist_session.hip)
     if args.lock:
         args.workers = 1
 
@@ -59,9 +73,10 @@ def main():
 
     cfg = TrainConfig(iters=args.steps, mini_batch=args.mbs,
                       mini_stochastic_iters=1, shuffle=False,
-                      acquire_lock=args.lock)
+                      acquire_lock=args.lock, grad_accum=args.grad_accum)
     engine = build_engine(graph_json, OptimizerSpec(args.optimizer, args.lr),
-                          cfg, workers_per_gpu=args.workers)
+                          cfg, workers_per_gpu=args.workers,
+                          remote_workers_per_gpu=args.remote_workers)
     device = engine.device
 
     # synthetic data of the benchmark config's shape
@@ -131,17 +146,23 @@ def main():
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
 
-    # MAX elapsed over ranks
+    # MAX elapsed over ranks; SUM of samples (ranks may run different
+    # worker counts under --remote-workers)
+    total_samples = timed_samples
     if world > 1:
         import torch.distributed as dist
 
-        t = torch.tensor([elapsed], dtype=torch.float64,
+        t = torch.tensor([elapsed, float(timed_samples)],
+                         dtype=torch.float64,
                          device=device if device.type == "cuda" else "cpu")
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = float(t.item())
+        e = t[:1].clone()
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        s = t[1:].clone()
+        dist.all_reduce(s, op=dist.ReduceOp.SUM)
+        elapsed = float(e.item())
+        total_samples = int(s.item())
 
     mbs_eff = min(args.mbs, x.shape[0])
-    total_samples = timed_samples * world  # every rank does the same work
     value = total_samples / elapsed
 
     if rank == 0:
@@ -164,9 +185,13 @@ def main():
                           "mnist_cnn_nhwc": "mnist_cnn_5x5x32_3x3x64_nhwc",
                           "tabular_autoencoder": "autoencoder_784_32_784",
                           "resnet18_shape": "resnet18_shape_3x224x224"}[args.model],
-                "global_batch": mbs_eff * world * args.workers,
+                "global_batch": mbs_eff * (args.workers + (world - 1) * (
+                    args.workers if args.remote_workers is None
+                    else args.remote_workers)),
                 "seq_len": None,
                 "workers_per_gpu": args.workers,
+                "remote_workers_per_gpu": args.remote_workers,
+                "grad_accum": args.grad_accum,
                 "parallelism": ("async-ps-%s dp%d x %dw"
                                 % ("lock" if args.lock else "hogwild", world,
                                    args.workers)),

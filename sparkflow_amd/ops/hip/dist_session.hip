@@ -54,6 +54,41 @@ void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
 static std::vector<ncclComm_t> g_comms;
 static std::mutex g_comms_mu;
 
+// SPARKFLOW_SESSION_CHUNK_ELEMS splits every session/raw payload
+// transfer into fixed-size chunks (star-contention knob: with 7 ranks x
+// W sessions landing on the PS GPU, chunking lets RCCL interleave the
+// concurrent pair transfers instead of serializing whole messages,
+// smoothing tail latency).  0 (default) = one transfer per message.
+// MUST be set identically on every rank: NCCL matches send/recv counts
+// per operation, so both ends of a pair have to agree on the split.
+static size_t session_chunk_elems() {
+  static size_t cached = [] {
+    const char* e = getenv("SPARKFLOW_SESSION_CHUNK_ELEMS");
+    long v = e ? atol(e) : 0;
+    return v > 0 ? (size_t)v : (size_t)0;
+  }();
+  return cached;
+}
+
+static void chunked_xfer(bool send, float* p, size_t n, int peer,
+                         ncclComm_t comm, hipStream_t stream) {
+  const size_t c = session_chunk_elems();
+  if (c == 0 || c >= n) {
+    if (send)
+      NCCL_OK(ncclSend(p, n, ncclFloat32, peer, comm, stream));
+    else
+      NCCL_OK(ncclRecv(p, n, ncclFloat32, peer, comm, stream));
+    return;
+  }
+  for (size_t off = 0; off < n; off += c) {
+    const size_t m = std::min(c, n - off);
+    if (send)
+      NCCL_OK(ncclSend(p + off, m, ncclFloat32, peer, comm, stream));
+    else
+      NCCL_OK(ncclRecv(p + off, m, ncclFloat32, peer, comm, stream));
+  }
+}
+
 py::bytes nccl_unique_id() {
   ncclUniqueId id;
   NCCL_OK(ncclGetUniqueId(&id));
@@ -140,8 +175,8 @@ void ps_session(int64_t comm_h, int64_t peer, torch::Tensor weights,
 
   py::gil_scoped_release rel;
   for (int64_t i = 0; i < k; ++i) {
-    NCCL_OK(ncclSend(wp, n, ncclFloat32, (int)peer, comm, stream.stream()));
-    NCCL_OK(ncclRecv(gp, n, ncclFloat32, (int)peer, comm, stream.stream()));
+    chunked_xfer(true, wp, n, (int)peer, comm, stream.stream());
+    chunked_xfer(false, gp, n, (int)peer, comm, stream.stream());
     switch (opt_id) {
       case OPT_SGD:
         fused_sgd(weights, grad_buf, lr);
@@ -177,10 +212,9 @@ void worker_session(int64_t comm_h, int64_t peer, torch::Tensor w_flat,
 
   py::gil_scoped_release rel;
   for (int64_t i = 0; i < k; ++i) {
-    NCCL_OK(ncclRecv(wp, n, ncclFloat32, (int)peer, comm, stream));
+    chunked_xfer(false, wp, n, (int)peer, comm, stream);
     HIP_OK(hipGraphLaunch(exec, stream));
-    NCCL_OK(ncclSend(gp, (size_t)g_flat.numel(), ncclFloat32, (int)peer,
-                     comm, stream));
+    chunked_xfer(true, gp, (size_t)g_flat.numel(), (int)peer, comm, stream);
   }
 }
 
@@ -190,8 +224,8 @@ void raw_send(int64_t comm_h, int64_t peer, torch::Tensor t) {
   ncclComm_t comm = get_comm(comm_h);
   hipStream_t stream = at::cuda::getCurrentCUDAStream().stream();
   py::gil_scoped_release rel;
-  NCCL_OK(ncclSend(t.data_ptr<float>(), t.numel(), ncclFloat32, (int)peer,
-                   comm, stream));
+  chunked_xfer(true, t.data_ptr<float>(), t.numel(), (int)peer, comm,
+               stream);
 }
 
 void raw_recv(int64_t comm_h, int64_t peer, torch::Tensor t) {

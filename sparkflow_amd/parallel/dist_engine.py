@@ -147,10 +147,16 @@ class EngineHandle:
             for s in streams:
                 cur.wait_stream(s)
             return total
+        # CPU: interleave eager steps across workers in grad_accum-sized
+        # chunks (so accumulation rounds stay intact per worker)
         total = 0
-        for _ in range(k):  # CPU: interleave eager single steps
+        chunk = max(1, getattr(self.workers[0].config, "grad_accum", 1))
+        done = 0
+        while done < k:
+            a = min(chunk, k - done)
             for w in self.workers:
-                total += w.run_steps(1, mini_batch, use_graph=False)
+                total += w.run_steps(a, mini_batch, use_graph=False)
+            done += a
         return total
 
     def stage_all(self, x, y) -> None:
@@ -190,15 +196,25 @@ class EngineHandle:
 
 def build_engine(graph_json: str, optimizer: OptimizerSpec,
                  config: TrainConfig, device=None,
-                 loss_callback=None, workers_per_gpu: int = 1) -> EngineHandle:
+                 loss_callback=None, workers_per_gpu: int = 1,
+                 remote_workers_per_gpu: Optional[int] = None) -> EngineHandle:
     """Construct PS + worker(s) for this rank (call on every rank).
 
     ``workers_per_gpu`` > 1 runs that many concurrent Hogwild workers on
     this rank's GPU (the reference's multiple-partitions-per-host model,
     e.g. the canonical README config's 4 partitions on local[4]); on the
     PS rank each gets its own LocalTransport, on worker ranks they share
-    the rank's RCCL pair link (driven sequentially by one host thread)."""
+    the rank's RCCL pair link (driven sequentially by one host thread).
+
+    ``remote_workers_per_gpu`` (default: same as ``workers_per_gpu``)
+    sets the worker count on NON-PS ranks independently — at world=8 the
+    PS GPU serves 7 ranks x W sessions, so fewer remote workers is the
+    first star-contention remedy.  Must be passed identically on every
+    rank (the pair process-groups are created collectively)."""
     rank, world, device = init_from_env(device=device)
+    wpg = max(1, workers_per_gpu)
+    rwpg = wpg if remote_workers_per_gpu is None \
+        else max(1, remote_workers_per_gpu)
 
     if device.type == "cuda":
         from sparkflow_amd import ops as _ops
@@ -206,11 +222,14 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
         if _ops.available():
             # create the replay streams FIRST so they land on distinct
             # HW queues (see dist_session.hip replay_streams_init)
-            _ops.hip.replay_streams_init(max(1, workers_per_gpu))
+            _ops.hip.replay_streams_init(wpg if rank == 0 else rwpg)
 
     ps = service = None
     if world > 1:
-        pairs = DistPairs(workers_per_gpu=max(1, workers_per_gpu))
+        # pair communicators exist only for REMOTE workers (rank 0's
+        # local workers ride LocalTransport), so their per-rank count is
+        # the remote one
+        pairs = DistPairs(workers_per_gpu=rwpg)
         pairs.init_native(device)  # raw RCCL channel (GIL-free sessions)
         _warmup_pairs(pairs, device)
     if rank == 0:
@@ -219,12 +238,11 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
                     max_errors=config.iters)
         if world > 1:
             service = DistPSService(ps, pairs).start()
-        transports = [LocalTransport(ps, device)
-                      for _ in range(max(1, workers_per_gpu))]
+        transports = [LocalTransport(ps, device) for _ in range(wpg)]
     else:
         numel = flat_size(module_from_json(graph_json))
         transports = [DistTransport(pairs, numel, device, widx=i)
-                      for i in range(max(1, workers_per_gpu))]
+                      for i in range(rwpg)]
 
     workers = []
     for i, tr in enumerate(transports):

@@ -67,6 +67,12 @@ class TrainConfig:
     partition_shuffles: int = 1
     port: int = 5000
     seed: Optional[int] = None
+    # Worker-side gradient accumulation (star-contention knob for large
+    # worlds): push the MEAN of grad_accum minibatch grads per transport
+    # round — cuts PS pull/push traffic by that factor.  1 = reference
+    # semantics (push every minibatch).  Applies to remote (RCCL/TCP)
+    # transports; the same-GPU full-step graph path has no wire to save.
+    grad_accum: int = 1
     extra: Dict = field(default_factory=dict)
 
 

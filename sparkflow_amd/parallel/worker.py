@@ -218,15 +218,24 @@ class Worker:
                     raise
                 print(f"[sparkflow_amd] hipGraph capture failed ({e}); "
                       "running eager steps")
-        for _ in range(k):
+        A = max(1, getattr(self.config, "grad_accum", 1))
+        done = 0
+        while done < k:
+            a = min(A, k - done)
             self._pull_and_load()
-            if mbs < n:
-                pick = self._minibatch_indices(n, mbs)
-                xb = self.x[pick]
-                yb = self.y[pick] if self.y is not None else None
-            else:
-                xb, yb = self.x, self.y
-            self._push(self._grad_step(xb, yb))
+            acc = None
+            for _ in range(a):
+                if mbs < n:
+                    pick = self._minibatch_indices(n, mbs)
+                    xb = self.x[pick]
+                    yb = self.y[pick] if self.y is not None else None
+                else:
+                    xb, yb = self.x, self.y
+                g = self._grad_step(xb, yb)
+                acc = g if a == 1 else (g.clone() if acc is None
+                                        else acc.add_(g))
+            self._push(acc if a == 1 else acc.div_(a))
+            done += a
         return k * mbs
 
     _IDX_POOL_CAP = 2048
@@ -323,6 +332,9 @@ class Worker:
         self.ensure_graph(mbs)
         self.maybe_refresh_pool(k)
         g = self._graph_state
+        A = max(1, getattr(self.config, "grad_accum", 1))
+        if A > 1 and not g["full_step"]:
+            return self._run_steps_graphed_accum(k, mbs, A)
         if not g["full_step"] and hasattr(self.transport, "run_session") \
                 and (g["gather_inside"] or g["xb"] is self.x) \
                 and getattr(self.transport, "native_comm", None) is not None:
@@ -340,6 +352,36 @@ class Worker:
         for _ in range(k):
             self.replay_once()
         self.finish_steps(k)
+        return k * mbs
+
+    def _run_steps_graphed_accum(self, k: int, mbs: int, A: int) -> int:
+        """Remote-transport steps with gradient accumulation: each
+        transport round is {pull weights; A x (gather+fwd/bwd replay,
+        accumulate); push mean grad} — k compute steps become
+        ceil(k/A) PS pulls+pushes (star-contention knob; see
+        TrainConfig.grad_accum)."""
+        g = self._graph_state
+        n = self.x.shape[0]
+        rounds = (k + A - 1) // A
+        if hasattr(self.transport, "run_begin"):
+            self.transport.run_begin(rounds)
+        acc = torch.zeros_like(self.flat_grads)
+        done = 0
+        for _ in range(rounds):
+            a = min(A, k - done)
+            self._pull_and_load()
+            acc.zero_()
+            for _ in range(a):
+                if g["xb"] is not self.x and not g["gather_inside"]:
+                    pick = self._minibatch_indices(n, g["xb"].shape[0])
+                    torch.index_select(self.x, 0, pick, out=g["xb"])
+                    if g["yb"] is not None:
+                        torch.index_select(self.y, 0, pick, out=g["yb"])
+                g["graph"].replay()
+                acc.add_(self.flat_grads)
+            self._push(acc.div_(a) if a > 1 else acc)
+            done += a
+        self.steps_done += k
         return k * mbs
 
     def _capture_graph(self, key, mbs: int, sub: bool, full_step: bool):

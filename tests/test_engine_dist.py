@@ -207,3 +207,59 @@ def test_dist_session_soak_world8():
     x 4 sessions into rank 0) rehearsed on CPU (round-1 verdict #3b)."""
     port = free_port()
     mp.spawn(_session_worker, args=(8, port, 3, 4), nprocs=8, join=True)
+
+
+def _knobs_worker(rank, world, port):
+    """Star-contention knobs: remote_workers_per_gpu != workers_per_gpu
+    and grad_accum>1 (push the mean of A minibatch grads per PS round)."""
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["LOCAL_RANK"] = str(rank)
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import gaussian_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    k, A = 6, 3
+    graph_json = build_graph(gaussian_mlp)
+    cfg = TrainConfig(iters=1, mini_batch=16, mini_stochastic_iters=1,
+                      seed=5, grad_accum=A)
+    engine = build_engine(graph_json, OptimizerSpec("gradient_descent", 0.01),
+                          cfg, device="cpu", workers_per_gpu=2,
+                          remote_workers_per_gpu=1)
+    rng = np.random.default_rng(rank)
+    x = rng.normal(0, 1, (64, 10)).astype(np.float32)
+    y = (rng.random((64, 1)) > 0.5).astype(np.float32)
+    engine.stage_all(x, y)
+    engine.barrier()
+    if engine.rank == 0:
+        assert len(engine.workers) == 2
+        samples = engine.run_steps(k, 16)
+        assert samples == 2 * k * 16
+        # local workers: k steps with grad_accum A -> k/A applies each;
+        # remote rank: 1 worker x k/A session pushes
+        import time
+
+        expected = 2 * (k // A) + (world - 1) * 1 * (k // A)
+        deadline = time.time() + 30
+        while engine.ps.update_count < expected and time.time() < deadline:
+            time.sleep(0.05)
+        assert engine.ps.update_count == expected, engine.ps.update_count
+        assert torch.isfinite(engine.ps.weights).all()
+    else:
+        assert len(engine.workers) == 1
+        samples = engine.run_steps(k, 16)
+        assert samples == k * 16
+    engine.barrier()
+    engine.close()
+    engine.barrier()
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_dist_knobs_remote_workers_and_grad_accum():
+    port = free_port()
+    mp.spawn(_knobs_worker, args=(2, port), nprocs=2, join=True)
