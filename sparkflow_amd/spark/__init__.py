@@ -1,9 +1,15 @@
 """Spark-ML Pipeline API layer (parity with reference
 sparkflow/tensorflow_async.py, pipeline_util.py, tensorflow_model_loader.py).
 
-pyspark is imported lazily: the compute engine and kernels work without it
-(local/torchrun modes); installing pyspark enables this layer unchanged.
+When a real pyspark is installed, this layer rides it unchanged.  When
+it is not (the MI355X single-node image has no JVM), the bundled
+``sparkflow_amd.localspark`` runtime registers itself under the pyspark
+names the first time this package is imported, so ``SparkAsyncDL.fit``/
+``transform``/pipeline save-load run for real either way.
 """
+from sparkflow_amd import localspark as _localspark
+
+_localspark.install(quiet=True)
 
 __all__ = [
     "SparkAsyncDL",

@@ -14,6 +14,12 @@ if str(REPO_ROOT) not in sys.path:
 os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
 os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
 
+# Spark layer tests run on the real pyspark when installed, else on the
+# bundled JVM-free localspark runtime — either way they RUN, not skip.
+from sparkflow_amd import localspark as _localspark  # noqa: E402
+
+_localspark.install(quiet=True)
+
 
 def pytest_configure(config):
     config.addinivalue_line(
