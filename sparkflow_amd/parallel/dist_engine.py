@@ -195,7 +195,7 @@ class EngineHandle:
 
 
 def build_engine(graph_json: str, optimizer: OptimizerSpec,
-                 config: TrainConfig, device=None,
+                 config: TrainConfig, device=None, backend=None,
                  loss_callback=None, workers_per_gpu: int = 1,
                  remote_workers_per_gpu: Optional[int] = None) -> EngineHandle:
     """Construct PS + worker(s) for this rank (call on every rank).
@@ -211,7 +211,7 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
     PS GPU serves 7 ranks x W sessions, so fewer remote workers is the
     first star-contention remedy.  Must be passed identically on every
     rank (the pair process-groups are created collectively)."""
-    rank, world, device = init_from_env(device=device)
+    rank, world, device = init_from_env(backend=backend, device=device)
     wpg = max(1, workers_per_gpu)
     rwpg = wpg if remote_workers_per_gpu is None \
         else max(1, remote_workers_per_gpu)

@@ -122,6 +122,111 @@ class HogwildSparkModel:
             return self.worker_device
         return "cuda:0" if torch.cuda.is_available() else "cpu"
 
+    def _dist_fit_eligible(self, rdd) -> bool:
+        """Decide whether fit drives the RCCL star (one rank per
+        partition/executor) instead of the TCP fallback.
+
+        SPARKFLOW_FIT_TRANSPORT=dist forces it, =tcp disables it; "auto"
+        (default) uses it when CUDA is available, there are >= 2
+        partitions, partition_shuffles == 1 (cross-partition reshuffles
+        would need a world rebuild) and the executors are real processes
+        (always true on a Spark cluster; under the bundled localspark it
+        needs the ``localproc[N]`` master)."""
+        import os
+
+        mode = os.environ.get("SPARKFLOW_FIT_TRANSPORT", "auto").lower()
+        if mode == "tcp":
+            return False
+        if rdd.getNumPartitions() < 2 or self.partition_shuffles != 1:
+            if mode == "dist":
+                print("[sparkflow_amd] dist fit forced but unsupported for "
+                      "this job (needs >=2 partitions, partitionShuffles=1)"
+                      "; using TCP")
+            return False
+        sess = getattr(rdd, "_session", None)  # localspark only
+        proc_executors = (sess is None
+                          or sess.sparkContext.process_mode)
+        if mode == "dist":
+            return proc_executors
+        return torch.cuda.is_available() and proc_executors
+
+    def _train_rdd_dist(self, rdd):
+        """One rank per partition over torch.distributed: rank 0 hosts
+        the GPU PS (+ its local worker on partition 0's data), ranks
+        1..N-1 pull/push over their own RCCL pair link — the reference's
+        worker<->PS HTTP hot path (HogwildSparkModel.py:22-35,57-58)
+        replaced by the xGMI star, reachable straight from
+        SparkAsyncDL.fit.  Returns the final weights ndarray list."""
+        from sparkflow_amd.parallel.transport import find_free_port
+
+        world = rdd.getNumPartitions()
+        addr = (self.master_url or "127.0.0.1:0").split(":")[0]
+        if addr in ("localhost",):
+            addr = "127.0.0.1"
+        port = find_free_port()
+        graph_json = self.graph_json
+        cfg = self.config
+        opt = self.optimizer
+        supervised = self.is_supervised
+        cb = self.loss_callback
+
+        def closure(idx, part_iter):
+            import os as _os
+
+            import numpy as _np  # noqa: F401
+            import torch as _torch
+
+            _os.environ["RANK"] = str(idx)
+            _os.environ["WORLD_SIZE"] = str(world)
+            _os.environ["MASTER_ADDR"] = addr
+            _os.environ["MASTER_PORT"] = str(port)
+            _os.environ["LOCAL_RANK"] = str(idx)
+            ngpu = (_torch.cuda.device_count()
+                    if _torch.cuda.is_available() else 0)
+            if ngpu >= world:
+                device, backend = "cuda:%d" % idx, "nccl"  # RCCL star
+            elif ngpu > 0:
+                # more ranks than GPUs: share GPUs for compute, keep the
+                # wire on gloo (multiple RCCL ranks per device deadlock)
+                device, backend = "cuda:%d" % (idx % ngpu), "gloo"
+            else:
+                device, backend = "cpu", "gloo"
+
+            from sparkflow_amd.ml_util import handle_features
+            from sparkflow_amd.parallel.dist_engine import build_engine
+
+            engine = build_engine(graph_json, opt, cfg, device=device,
+                                  backend=backend, loss_callback=cb)
+            info = {"rank": idx, "device": device, "backend": backend,
+                    "transport": type(engine.worker.transport).__name__}
+            try:
+                data = list(part_iter)
+                engine.barrier()
+                if data:
+                    features, labels = handle_features(data, supervised)
+                    engine.worker.run(features, labels)
+                pairs = getattr(engine.worker.transport, "_pairs", None)
+                info["native_pairs"] = len(pairs.native) if pairs else 0
+                engine.close()  # workers: DONE; rank 0: join + failures
+                if idx == 0:
+                    info["weights"] = (engine.ps.weights.detach()
+                                       .cpu().numpy())
+                    info["update_count"] = engine.ps.update_count
+                engine.barrier()
+                return [info]
+            finally:
+                import torch.distributed as dist
+
+                if dist.is_initialized():
+                    dist.destroy_process_group()
+
+        infos = rdd.mapPartitionsWithIndex(closure).collect()
+        by_rank = {i["rank"]: i for i in infos}
+        flat = by_rank[0]["weights"]
+        self.last_dist_fit = by_rank  # introspection for tests/telemetry
+        module = module_from_json(self.graph_json)
+        return flat_to_list(module, torch.as_tensor(flat))
+
     def _train_rdd(self, rdd):
         """Spark path: one worker closure per partition (reference
         HogwildSparkModel.py:258-266)."""
@@ -182,6 +287,10 @@ class HogwildSparkModel:
         (reference train contract, HogwildSparkModel.py:246-272)."""
         try:
             if hasattr(data, "foreachPartition"):
+                if self._dist_fit_eligible(data):
+                    weights = self._train_rdd_dist(data)
+                    self.stop_server()  # TCP PS was never the data plane
+                    return weights
                 self._train_rdd(data)
             else:
                 self._train_local(data)

@@ -213,6 +213,15 @@ def wait_for_server(master_url: str, timeout: float = 60.0) -> None:
 # ---------------------------------------------------------------------------
 
 
+def find_free_port(host: str = "127.0.0.1") -> int:
+    """OS-assigned free TCP port (rendezvous bootstrap)."""
+    s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    s.bind((host, 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 class DistPairs:
     """Pair process-groups between rank 0 (PS) and each worker rank.
 

@@ -248,3 +248,44 @@ def test_transform_keeps_columns(spark):
                        iters=5).fit(df)
     out = model.transform(df)
     assert set(out.columns) == {"label", "features", "predicted"}
+
+
+@pytest.mark.timeout(300)
+def test_dist_fit_traverses_rccl_star(monkeypatch, tmp_path):
+    """VERDICT #1: SparkAsyncDL.fit on process executors drives the
+    torch.distributed star — rank 0 = PS + local worker, rank 1 pulls/
+    pushes through DistTransport (gloo wires on CPU; RCCL pair comms on
+    an MI355X node where world <= n_gpus)."""
+    monkeypatch.setenv("SPARKFLOW_FIT_TRANSPORT", "dist")
+    from sparkflow_amd.parallel.hogwild import HogwildSparkModel
+
+    session = (SparkSession.builder.master("localproc[2]")
+               .appName("sparkflow-dist-fit")
+               .config("spark.driver.host", "127.0.0.1").getOrCreate())
+    try:
+        rng = np.random.default_rng(7)
+        rows = [(i % 2,
+                 Vectors.dense(rng.normal(-2.0 if i % 2 == 0 else 2.0,
+                                          0.7, 10)))
+                for i in range(200)]
+        df = session.createDataFrame(rows, ["label", "features"])
+
+        captured = {}
+        orig = HogwildSparkModel._train_rdd_dist
+
+        def spy(self, rdd):
+            out = orig(self, rdd)
+            captured.update(self.last_dist_fit)
+            return out
+
+        monkeypatch.setattr(HogwildSparkModel, "_train_rdd_dist", spy)
+        est = _estimator(build_graph(gaussian_mlp), free_port(), iters=8)
+        model = est.fit(df)
+        rows10 = model.transform(df).take(10)
+        assert _calculate_errors(rows10) < len(rows10)
+        # the wiring really was the star:
+        assert captured[0]["transport"] == "LocalTransport"  # PS rank
+        assert captured[1]["transport"] == "DistTransport"   # worker rank
+        assert captured[0]["update_count"] >= 2 * 8  # both ranks pushed
+    finally:
+        session.stop()
