@@ -257,11 +257,22 @@ def test_dist_fit_traverses_rccl_star(monkeypatch, tmp_path):
     pushes through DistTransport (gloo wires on CPU; RCCL pair comms on
     an MI355X node where world <= n_gpus)."""
     monkeypatch.setenv("SPARKFLOW_FIT_TRANSPORT", "dist")
+    from sparkflow_amd.localspark import using_localspark
     from sparkflow_amd.parallel.hogwild import HogwildSparkModel
 
-    session = (SparkSession.builder.master("localproc[2]")
-               .appName("sparkflow-dist-fit")
-               .config("spark.driver.host", "127.0.0.1").getOrCreate())
+    if not using_localspark():
+        pytest.skip("localproc executors are a localspark feature; on a "
+                    "real Spark cluster the executors are processes "
+                    "already")
+    # construct directly (getOrCreate would return the module's local[2]
+    # thread-mode session; dist fit needs process executors)
+    from sparkflow_amd.localspark.context import SparkConf, SparkContext
+    from sparkflow_amd.localspark.sql import SparkSession as LSSession
+
+    prev_ctx = SparkContext._active_spark_context
+    session = LSSession(SparkContext(
+        master="localproc[2]", appName="sparkflow-dist-fit",
+        conf=SparkConf({"spark.driver.host": "127.0.0.1"})))
     try:
         rng = np.random.default_rng(7)
         rows = [(i % 2,
@@ -288,4 +299,4 @@ def test_dist_fit_traverses_rccl_star(monkeypatch, tmp_path):
         assert captured[1]["transport"] == "DistTransport"   # worker rank
         assert captured[0]["update_count"] >= 2 * 8  # both ranks pushed
     finally:
-        session.stop()
+        SparkContext._active_spark_context = prev_ctx
