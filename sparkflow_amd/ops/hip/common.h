@@ -21,6 +21,7 @@
 
 // fp32 MFMA accumulator fragment (v_mfma_f32_16x16x4_f32: 4 f32 per lane)
 typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
 
 // Activation codes shared with sparkflow_amd/ops/functional.py
 #define ACT_NONE 0

@@ -164,6 +164,152 @@ __global__ __launch_bounds__(256) void conv_nhwc_fwd_kernel(
   }
 }
 
+// ---------------- forward, large-M tile ----------------
+// ResNet-shape forward convs are LARGE-M (M = N*Ho*Wo is 10^5..10^6)
+// with modest Cout (64..512): the dense-GEMM 64x64/BK16 tile leaves
+// MFMA starved there (1 LDS read per 2048-FLOP mfma_16x16x4).  This
+// kernel uses BM=256 x BN=64 x BK=32 on v_mfma_f32_32x32x2_f32 (the
+// CDNA4 f32 GEMM recipe, cdna_hip_programming.md §3: 2x2 32x32 frags
+// per wave = 1 LDS read per 4096-FLOP instruction, i.e. half the LDS
+// traffic per FLOP) with BN=64 so a Cout=64 layer wastes nothing.
+// Each of the 4 waves owns a 64x64 quadrant of the 256-row stripe.
+// Requires Cin % 4 == 0 (float4 im2col gathers); other shapes take the
+// 64x64 kernel above.
+#define BM2 256
+#define BN2 64
+#define BK2 32
+
+template <bool HAS_BIAS>
+__global__ __launch_bounds__(256) void conv_nhwc_fwd_big_kernel(
+    const float* __restrict__ x, const float* __restrict__ wgt,
+    const float* __restrict__ bias, float* __restrict__ y, ConvShapeHW s,
+    int M, int K, int act) {
+  __shared__ float As[BM2][BK2 + 1];
+  __shared__ float Bs[BK2][BN2 + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = wv * 64;  // wave's 64-row quadrant of the 256-row stripe
+  const int m0 = blockIdx.y * BM2;
+  const int n0 = blockIdx.x * BN2;
+  const int Nc = s.Cout;
+
+  f32x16 acc[2][2] = {};
+
+  // A staging: thread t owns output-pixel row (m0 + t) and ALL BK2 k's
+  // of that row as 8 float4 groups.  ONE pixel decode per thread; the
+  // (khkw, ci) cursor tracks k-group 0 and groups 1..7 derive from it
+  // with at most one Cin-carry each (Cin % 4 == 0 and >= 4).
+  const int gm = m0 + t;
+  const int m_dec = gm < M ? gm : 0;
+  const int a_wo = m_dec % s.Wo;
+  const int a_ho = (m_dec / s.Wo) % s.Ho;
+  const long a_nbase = (long)(m_dec / (s.Wo * s.Ho)) * s.H * s.W * s.Cin;
+  const int h0 = a_ho * s.stride - s.pad;
+  const int w0 = a_wo * s.stride - s.pad;
+  int a_ci = 0, a_khkw = 0;  // cursor at k-group 0 (k = k0)
+
+  auto load_a = [&](int k0, float4 (&r)[8]) {
+#pragma unroll
+    for (int g = 0; g < 8; ++g) r[g] = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gm >= M) return;
+    int ci = a_ci, khkw = a_khkw;
+#pragma unroll
+    for (int g = 0; g < 8; ++g) {
+      const int kh = khkw / s.Kw, kw = khkw - (khkw / s.Kw) * s.Kw;
+      const int h = h0 + kh, w = w0 + kw;
+      if (k0 + g * 4 + 3 < K && h >= 0 && h < s.H && w >= 0 && w < s.W) {
+        r[g] = *reinterpret_cast<const float4*>(
+            &x[a_nbase + ((long)h * s.W + w) * s.Cin + ci]);
+      }
+      ci += 4;
+      if (ci >= s.Cin) { ci -= s.Cin; ++khkw; }
+    }
+  };
+  auto advance_a = [&]() {
+    a_ci += BK2;
+    while (a_ci >= s.Cin) { a_ci -= s.Cin; ++a_khkw; }
+  };
+  // B staging: thread owns co = n0 + (t & 63), k-offsets (t>>6)*8..+7 —
+  // 2 float4 straight reads from the contiguous w row.
+  const int b_co = n0 + (t & 63);
+  const int b_k8 = (t >> 6) * 8;
+  auto load_b = [&](int k0, float4 (&r)[2]) {
+    if (b_co < Nc && k0 + b_k8 + 7 < K) {
+      r[0] = *reinterpret_cast<const float4*>(
+          &wgt[(long)b_co * K + k0 + b_k8]);
+      r[1] = *reinterpret_cast<const float4*>(
+          &wgt[(long)b_co * K + k0 + b_k8 + 4]);
+    } else {
+      float* rr = reinterpret_cast<float*>(&r[0]);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int gk = k0 + b_k8 + i;
+        rr[i] = (b_co < Nc && gk < K) ? wgt[(long)b_co * K + gk] : 0.0f;
+      }
+    }
+  };
+
+  const int fi = lane & 31;   // mfma A row / D col-lane
+  const int fk = lane >> 5;   // mfma k slot (0/1)
+
+  float4 ra[8], rb[2];
+  load_a(0, ra);
+  load_b(0, rb);
+  for (int k0 = 0; k0 < K; k0 += BK2) {
+    {
+#pragma unroll
+      for (int g = 0; g < 8; ++g) {
+        const float* p = reinterpret_cast<const float*>(&ra[g]);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) As[t][g * 4 + i] = p[i];
+      }
+      const float* pb = reinterpret_cast<const float*>(&rb[0]);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) Bs[b_k8 + i][t & 63] = pb[i];
+    }
+    __syncthreads();
+    if (k0 + BK2 < K) {
+      advance_a();
+      load_a(k0 + BK2, ra);
+      load_b(k0 + BK2, rb);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK2 / 2; ++kk) {
+      const float a0 = As[wm + fi][kk * 2 + fk];
+      const float a1 = As[wm + 32 + fi][kk * 2 + fk];
+      const float b0 = Bs[kk * 2 + fk][fi];
+      const float b1 = Bs[kk * 2 + fk][32 + fi];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: D lane mapping col = lane&31, row = (reg&3) + 8*(reg>>2)
+  // + 4*(lane>>5); y[m][co] stores are co-contiguous across lanes.
+#pragma unroll
+  for (int ni = 0; ni < 2; ++ni) {
+    const int co = n0 + ni * 32 + fi;
+    if (co >= Nc) continue;
+    const float bv = HAS_BIAS ? bias[co] : 0.0f;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int row = (reg & 3) + 8 * (reg >> 2) + 4 * fk;
+        const int m = m0 + wm + mi * 32 + row;
+        if (m < M) {
+          y[(long)m * Nc + co] = act_apply(acc[mi][ni][reg] + bv, act);
+        }
+      }
+    }
+  }
+}
+
 // ---------------- wgrad ----------------
 // dw[co][k'] = sum_p dy[p, co] * im2col(x)[p, k'] — A staged transposed
 // (dy contiguous along co), B's im2col contiguous along ci.
@@ -373,12 +519,39 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w,
   const int M = s.N * s.Ho * s.Wo;
   const int K = s.Kh * s.Kw * s.Cin;
   auto y = torch::empty({s.N, s.Ho, s.Wo, s.Cout}, x.options());
-  dim3 grid(ceil_div(s.Cout, BN), ceil_div(M, BM));
   auto stream = at::cuda::getCurrentCUDAStream();
-  if (bias.numel() > 0) {
+  const float* bp = bias.numel() > 0 ? bias.data_ptr<float>() : nullptr;
+  // MEASURED NEGATIVE (round 2, profiles/conv_micro_r02.md): the
+  // 256x64/BK32 mfma_32x32x2 kernel (140 VGPR, 3 waves/SIMD) is ~15%
+  // slower than the 64x64/BK16 kernel on every ResNet-18 layer shape at
+  // mbs=300 — the 64x64 kernel already runs at 87-100 TF (56-64% of the
+  // 155 TF f32 MFMA peak), so halving LDS traffic buys nothing and the
+  // lost occupancy (56 VGPR -> 8 waves/SIMD vs 3) costs latency hiding
+  // on the im2col gathers.  Kept for re-evaluation: SPARKFLOW_CONV_BIG=1
+  // opts in; default is the 64x64 kernel.
+  static const bool big_enabled = [] {
+    const char* e = getenv("SPARKFLOW_CONV_BIG");
+    return e && e[0] == '1';
+  }();
+  const long big_tiles = (long)ceil_div(M, BM2) * ceil_div(s.Cout, BN2);
+  if (big_enabled && (s.Cin & 3) == 0 && K >= BK2 && big_tiles >= 128) {
+    dim3 grid(ceil_div(s.Cout, BN2), ceil_div(M, BM2));
+    if (bp) {
+      hipLaunchKernelGGL((conv_nhwc_fwd_big_kernel<true>), grid, dim3(256),
+                         0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                         bp, y.data_ptr<float>(), s, M, K, (int)act);
+    } else {
+      hipLaunchKernelGGL((conv_nhwc_fwd_big_kernel<false>), grid, dim3(256),
+                         0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                         nullptr, y.data_ptr<float>(), s, M, K, (int)act);
+    }
+    return y;
+  }
+  dim3 grid(ceil_div(s.Cout, BN), ceil_div(M, BM));
+  if (bp) {
     hipLaunchKernelGGL((conv_nhwc_fwd_kernel<true>), grid, dim3(256), 0,
                        stream, x.data_ptr<float>(), w.data_ptr<float>(),
-                       bias.data_ptr<float>(), y.data_ptr<float>(), s, M, K,
+                       bp, y.data_ptr<float>(), s, M, K,
                        (int)act);
   } else {
     hipLaunchKernelGGL((conv_nhwc_fwd_kernel<false>), grid, dim3(256), 0,

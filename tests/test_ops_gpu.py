@@ -352,6 +352,7 @@ def test_batchnorm2d_gpu(relu):
     (2, 28, 28, 4, 8, 3, 2, 1),
     (2, 17, 17, 3, 8, 7, 2, 3),   # Cin=3 scalar path (ResNet stem shape)
     (2, 9, 9, 8, 12, 1, 2, 0),    # 1x1 downsample
+    (12, 56, 56, 8, 64, 3, 1, 1),  # large-M shape
 ])
 def test_conv2d_nhwc_autograd_gpu(conf):
     """NHWC conv fwd/bwd (channel-contiguous kernels) vs the CPU permute
@@ -512,3 +513,48 @@ def test_gemm_wgrad_bias_ws_planes_sum_to_wgrad():
     torch.cuda.synchronize()
     assert ws.dim() == 3 and ws.shape[1:] == (785, 256)
     torch.testing.assert_close(ws.sum(dim=0), dwb, atol=1e-4, rtol=1e-4)
+
+
+def test_conv_nhwc_big_tile_parity_subprocess():
+    """Opt-in 256x64 mfma_32x32x2 conv kernel (SPARKFLOW_CONV_BIG=1 —
+    kept as a measured negative, see conv_nhwc.hip) still has to be
+    CORRECT: parity vs the default 64x64 kernel in a subprocess where
+    the env flag is read fresh."""
+    import os
+    import subprocess
+    import sys
+
+    from tests.conftest import REPO_ROOT
+
+    code = r"""
+import os, sys, torch
+sys.path.insert(0, %r)
+from sparkflow_amd import ops
+assert ops.available()
+torch.manual_seed(99)
+dev = "cuda:0"
+for (N,H,W,Cin,Cout,K,stride,pad) in [(12,56,56,8,64,3,1,1),
+                                      (12,56,56,8,70,3,1,1),
+                                      (6,28,28,20,64,3,2,1)]:
+    x = torch.randn(N,H,W,Cin, device=dev)
+    w = torch.randn(Cout,K,K,Cin, device=dev) * .1
+    b = torch.randn(Cout, device=dev)
+    y = ops.hip.conv_nhwc_fwd(x, w, b, stride, pad, 1)
+    torch.save(y.cpu(), os.environ["OUT"])
+    break
+"""
+    # run the FIRST shape under both flags and compare bitwise-ish
+    outs = {}
+    for flag in ("0", "1"):
+        env = dict(os.environ, SPARKFLOW_CONV_BIG=flag,
+                   OUT=f"/tmp/convbig_{flag}.pt")
+        r = subprocess.run([sys.executable, "-c",
+                            code % str(REPO_ROOT)], env=env,
+                           capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, r.stderr[-2000:]
+        import torch as _t
+
+        outs[flag] = _t.load(f"/tmp/convbig_{flag}.pt")
+    import torch as _t
+
+    _t.testing.assert_close(outs["0"], outs["1"], atol=2e-4, rtol=2e-4)
