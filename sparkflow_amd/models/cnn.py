@@ -61,7 +61,9 @@ class _BasicBlock(nn.Module):
         self.conv2 = snn.Conv2d(cout, cout, 3, stride=1, padding=1,
                                 activation="none", bias=False,
                                 layout=layout)
-        self.bn2 = snn.BatchNorm2d(cout, layout=layout)
+        # bn2 fuses the block boundary: out = relu(bn2(conv2) + identity)
+        # (one kernel instead of bn + add + relu)
+        self.bn2 = snn.BatchNorm2d(cout, fuse_relu=True, layout=layout)
         if stride != 1 or cin != cout:
             self.down_conv = snn.Conv2d(cin, cout, 1, stride=stride,
                                         padding=0, activation="none",
@@ -73,10 +75,9 @@ class _BasicBlock(nn.Module):
     def forward(self, x):
         idt = x
         y = self.bn1(self.conv1(x))  # fused bn+relu
-        y = self.bn2(self.conv2(y))
         if self.down_conv is not None:
             idt = self.down_bn(self.down_conv(x))
-        return torch.relu(y + idt)
+        return self.bn2(self.conv2(y), residual=idt)
 
 
 class ResNet18Shape(nn.Module):

@@ -119,17 +119,23 @@ class BatchNorm2d(nn.Module):
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         if self.layout == "nhwc":
             return F.batchnorm2d_act_nhwc(x, self.weight, self.bias,
                                           self.running_mean,
                                           self.running_var, self.momentum,
                                           self.eps, self.fuse_relu,
-                                          self.training)
-        return F.batchnorm2d_act(x, self.weight, self.bias,
-                                 self.running_mean, self.running_var,
-                                 self.momentum, self.eps, self.fuse_relu,
-                                 self.training)
+                                          self.training, residual=residual)
+        y = F.batchnorm2d_act(x, self.weight, self.bias,
+                              self.running_mean, self.running_var,
+                              self.momentum, self.eps,
+                              self.fuse_relu and residual is None,
+                              self.training)
+        if residual is not None:  # NCHW path: plain add (+relu)
+            y = y + residual
+            if self.fuse_relu:
+                y = torch.relu(y)
+        return y
 
     def extra_repr(self):
         return f"{self.num_features}, fuse_relu={self.fuse_relu}"

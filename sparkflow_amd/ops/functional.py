@@ -415,38 +415,53 @@ def conv2d_act_nhwc(x, w, b=None, stride=1, pad=0, act="none"):
 
 
 class BatchNorm2dActNHWC(torch.autograd.Function):
+    """NHWC training BN with optional fused trailing ReLU AND optional
+    fused ResNet-block residual add: out = relu(bn(x) + res).  Fusing
+    the add+relu into the norm kernel (and the relu-masked d_res into
+    the bwd-dx kernel) removes two full activation-tensor passes per
+    block boundary."""
+
     @staticmethod
     def forward(ctx, x, gamma, beta, run_mean, run_var, momentum, eps,
-                relu):
-        y, mean, invstd = ops.hip.bn2d_nhwc_fwd(x, gamma, beta, run_mean,
-                                                run_var, momentum, eps,
-                                                relu)
+                relu, res=None):
+        empty = torch.empty(0, device=x.device, dtype=x.dtype)
+        y, mean, invstd = ops.hip.bn2d_nhwc_fwd(
+            x, gamma, beta, run_mean, run_var, momentum, eps, relu,
+            res.contiguous() if res is not None else empty)
         ctx.save_for_backward(x, y, gamma, mean, invstd)
         ctx.relu = relu
+        ctx.has_res = res is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x, y, gamma, mean, invstd = ctx.saved_tensors
-        dx, dgamma, dbeta = ops.hip.bn2d_nhwc_bwd(x, dy.contiguous(), y,
-                                                  gamma, mean, invstd,
-                                                  ctx.relu)
-        return dx, dgamma, dbeta, None, None, None, None, None
+        dx, dgamma, dbeta, dres = ops.hip.bn2d_nhwc_bwd(
+            x, dy.contiguous(), y, gamma, mean, invstd, ctx.relu,
+            ctx.has_res)
+        return (dx, dgamma, dbeta, None, None, None, None, None,
+                dres if ctx.has_res else None)
 
 
 def batchnorm2d_act_nhwc(x, gamma, beta, run_mean, run_var, momentum=0.1,
-                         eps=1e-5, relu=False, training=True):
-    """x: [N, H, W, C] channel-last."""
-    if training and ops.use_hip(x):
+                         eps=1e-5, relu=False, training=True,
+                         residual=None):
+    """x: [N, H, W, C] channel-last.  ``residual`` fuses a ResNet block
+    boundary: out = relu(bn(x) + residual)."""
+    if training and ops.use_hip(x) and (residual is None
+                                        or x.shape[-1] % 4 == 0):
         return BatchNorm2dActNHWC.apply(x.contiguous(), gamma, beta,
                                         run_mean, run_var, momentum, eps,
-                                        relu)
+                                        relu, residual)
     xc = x.permute(0, 3, 1, 2)
     y = torch.nn.functional.batch_norm(xc, run_mean, run_var, gamma, beta,
                                        training, momentum, eps)
+    y = y.permute(0, 2, 3, 1)
+    if residual is not None:
+        y = y + residual
     if relu:
         y = torch.relu(y)
-    return y.permute(0, 2, 3, 1).contiguous()
+    return y.contiguous()
 
 
 class MaxPool2d(torch.autograd.Function):
@@ -491,14 +506,15 @@ class MaxPool2dNHWC(torch.autograd.Function):
     def forward(ctx, x, k: int, stride: int, pad: int):
         y, idx = ops.hip.maxpool_nhwc_fwd(x, k, stride, pad)
         ctx.save_for_backward(idx)
-        ctx.conf = (x.shape[1], x.shape[2])
+        ctx.conf = (x.shape[1], x.shape[2], k, stride, pad)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         (idx,) = ctx.saved_tensors
-        H, W = ctx.conf
-        return ops.hip.maxpool_nhwc_bwd(dy.contiguous(), idx, H, W), \
+        H, W, k, stride, pad = ctx.conf
+        return ops.hip.maxpool_nhwc_bwd(dy.contiguous(), idx, H, W, k,
+                                        stride, pad), \
             None, None, None
 
 

@@ -96,7 +96,8 @@ torch::Tensor conv_nhwc_bias_grad(torch::Tensor dy);
 std::vector<torch::Tensor> maxpool_nhwc_fwd(torch::Tensor x, int64_t k,
                                             int64_t stride, int64_t pad);
 torch::Tensor maxpool_nhwc_bwd(torch::Tensor dy, torch::Tensor idx,
-                               int64_t H, int64_t W);
+                               int64_t H, int64_t W, int64_t k,
+                               int64_t stride, int64_t pad);
 
 // norm.hip — training-mode BatchNorm2d (+ fused ReLU)
 std::vector<torch::Tensor> bn2d_fwd(torch::Tensor x, torch::Tensor gamma,
@@ -114,13 +115,13 @@ std::vector<torch::Tensor> bn2d_nhwc_fwd(torch::Tensor x,
                                          torch::Tensor run_mean,
                                          torch::Tensor run_var,
                                          double momentum, double eps,
-                                         bool relu);
+                                         bool relu, torch::Tensor res);
 std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
                                          torch::Tensor y,
                                          torch::Tensor gamma,
                                          torch::Tensor save_mean,
                                          torch::Tensor save_invstd,
-                                         bool relu);
+                                         bool relu, bool want_dres);
 
 // dist_session.hip — native RCCL session loops (GIL-free)
 py::bytes nccl_unique_id();
@@ -210,13 +211,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "NHWC conv bias grad (channel-contiguous column sums)");
   m.def("maxpool_nhwc_fwd", &maxpool_nhwc_fwd,
         "NHWC maxpool fwd with argmax indices");
-  m.def("maxpool_nhwc_bwd", &maxpool_nhwc_bwd, "NHWC maxpool bwd scatter");
+  m.def("maxpool_nhwc_bwd", &maxpool_nhwc_bwd,
+        "NHWC maxpool bwd (float4 window-gather; scalar scatter fallback)");
   m.def("bn2d_fwd", &bn2d_fwd,
         "training BatchNorm2d fwd (+ fused ReLU); returns y, mean, invstd");
   m.def("bn2d_bwd", &bn2d_bwd,
         "BatchNorm2d bwd (+ fused ReLU bwd); returns dx, dgamma, dbeta");
-  m.def("bn2d_nhwc_fwd", &bn2d_nhwc_fwd, "NHWC training BatchNorm2d fwd");
-  m.def("bn2d_nhwc_bwd", &bn2d_nhwc_bwd, "NHWC BatchNorm2d bwd");
+  m.def("bn2d_nhwc_fwd", &bn2d_nhwc_fwd,
+        "NHWC training BatchNorm2d fwd (float4; optional fused residual"
+        " add + relu)");
+  m.def("bn2d_nhwc_bwd", &bn2d_nhwc_bwd,
+        "NHWC BatchNorm2d bwd (optionally emits d_res = relu-masked dy)");
   m.def("nccl_unique_id", &nccl_unique_id, "RCCL unique id for a pair comm");
   m.def("nccl_comm_init", &nccl_comm_init,
         "init this rank's side of a raw RCCL communicator (non-blocking\n"

@@ -242,6 +242,205 @@ std::vector<torch::Tensor> bn2d_bwd(torch::Tensor x, torch::Tensor dy,
 
 #define BNC 64  // channel lanes per block
 
+// float4 variants (C % 4 == 0 — every ResNet/CNN channel count): 16
+// channel-quads x 16 row groups per 256-thread block, so a wave's row
+// read is 16 lanes x float4 = 64 consecutive channels (full-width
+// transactions; the scalar kernels below move one float per lane and
+// measured ~2x off the stream floor on the ResNet shapes).  The norm
+// kernel optionally FUSES the residual add + trailing ReLU of a ResNet
+// block (out = relu(bn(x) + res)), and the bwd-dx kernel then also
+// emits d_res (the relu-masked dy) so no separate torch add/relu
+// kernels run on the block boundary.
+
+__global__ void bn_nhwc_stats_v4_kernel(const float* __restrict__ x,
+                                        float* __restrict__ stats,
+                                        long rows, int C) {
+  const int cl = threadIdx.x & 15;          // channel quad
+  const int rg = threadIdx.x >> 4;          // row group 0..15
+  const int c = blockIdx.x * BNC + cl * 4;
+  const bool ok = c < C;
+  float4 s = {0, 0, 0, 0}, s2 = {0, 0, 0, 0};
+  if (ok) {
+    for (long r = (long)blockIdx.y * 16 + rg; r < rows;
+         r += (long)gridDim.y * 16) {
+      const float4 v = *reinterpret_cast<const float4*>(&x[r * C + c]);
+      s.x += v.x; s.y += v.y; s.z += v.z; s.w += v.w;
+      s2.x += v.x * v.x; s2.y += v.y * v.y;
+      s2.z += v.z * v.z; s2.w += v.w * v.w;
+    }
+  }
+  __shared__ float4 ls[16][16], ls2[16][16];
+  ls[rg][cl] = s;
+  ls2[rg][cl] = s2;
+  __syncthreads();
+  if (rg == 0 && ok) {
+#pragma unroll
+    for (int k = 1; k < 16; ++k) {
+      const float4 a = ls[k][cl], b = ls2[k][cl];
+      s.x += a.x; s.y += a.y; s.z += a.z; s.w += a.w;
+      s2.x += b.x; s2.y += b.y; s2.z += b.z; s2.w += b.w;
+    }
+    atomicAdd(&stats[c + 0], s.x);
+    atomicAdd(&stats[c + 1], s.y);
+    atomicAdd(&stats[c + 2], s.z);
+    atomicAdd(&stats[c + 3], s.w);
+    atomicAdd(&stats[C + c + 0], s2.x);
+    atomicAdd(&stats[C + c + 1], s2.y);
+    atomicAdd(&stats[C + c + 2], s2.z);
+    atomicAdd(&stats[C + c + 3], s2.w);
+  }
+}
+
+__global__ void bn_nhwc_norm_v4_kernel(
+    const float* __restrict__ x, float* __restrict__ y,
+    const float* __restrict__ stats, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ save_mean,
+    float* __restrict__ save_invstd, float* __restrict__ run_mean,
+    float* __restrict__ run_var, const float* __restrict__ res,
+    float momentum, float eps, long rows, int C, int relu) {
+  const int cl = threadIdx.x & 15;
+  const int rg = threadIdx.x >> 4;
+  const int c = blockIdx.x * BNC + cl * 4;
+  if (c >= C) return;
+  const float cnt = (float)rows;
+  float4 scale, shift;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const float mean = stats[c + j] / cnt;
+    const float var = stats[C + c + j] / cnt - mean * mean;
+    const float invstd = __frsqrt_rn(var + eps);
+    const float g = gamma ? gamma[c + j] : 1.0f;
+    const float b = beta ? beta[c + j] : 0.0f;
+    reinterpret_cast<float*>(&scale)[j] = g * invstd;
+    reinterpret_cast<float*>(&shift)[j] = b - mean * (g * invstd);
+    if (blockIdx.y == 0 && rg == 0) {
+      save_mean[c + j] = mean;
+      save_invstd[c + j] = invstd;
+      if (run_mean) {
+        run_mean[c + j] =
+            (1.0f - momentum) * run_mean[c + j] + momentum * mean;
+        const float ub = var * cnt / fmaxf(cnt - 1.0f, 1.0f);
+        run_var[c + j] = (1.0f - momentum) * run_var[c + j] + momentum * ub;
+      }
+    }
+  }
+  for (long r = (long)blockIdx.y * 16 + rg; r < rows;
+       r += (long)gridDim.y * 16) {
+    float4 v = *reinterpret_cast<const float4*>(&x[r * C + c]);
+    v.x = v.x * scale.x + shift.x;
+    v.y = v.y * scale.y + shift.y;
+    v.z = v.z * scale.z + shift.z;
+    v.w = v.w * scale.w + shift.w;
+    if (res) {
+      const float4 rv = *reinterpret_cast<const float4*>(&res[r * C + c]);
+      v.x += rv.x; v.y += rv.y; v.z += rv.z; v.w += rv.w;
+    }
+    if (relu) {
+      v.x = fmaxf(v.x, 0.0f);
+      v.y = fmaxf(v.y, 0.0f);
+      v.z = fmaxf(v.z, 0.0f);
+      v.w = fmaxf(v.w, 0.0f);
+    }
+    *reinterpret_cast<float4*>(&y[r * C + c]) = v;
+  }
+}
+
+__global__ void bn_nhwc_bwd_reduce_v4_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    const float* __restrict__ y, const float* __restrict__ sm,
+    const float* __restrict__ si, float* __restrict__ red, long rows,
+    int C, int relu) {
+  const int cl = threadIdx.x & 15;
+  const int rg = threadIdx.x >> 4;
+  const int c = blockIdx.x * BNC + cl * 4;
+  const bool ok = c < C;
+  float4 sd = {0, 0, 0, 0}, sdx = {0, 0, 0, 0};
+  if (ok) {
+    const float4 mean = *reinterpret_cast<const float4*>(&sm[c]);
+    const float4 invstd = *reinterpret_cast<const float4*>(&si[c]);
+    for (long r = (long)blockIdx.y * 16 + rg; r < rows;
+         r += (long)gridDim.y * 16) {
+      float4 d = *reinterpret_cast<const float4*>(&dy[r * C + c]);
+      if (relu) {
+        const float4 yv = *reinterpret_cast<const float4*>(&y[r * C + c]);
+        if (yv.x <= 0.0f) d.x = 0.0f;
+        if (yv.y <= 0.0f) d.y = 0.0f;
+        if (yv.z <= 0.0f) d.z = 0.0f;
+        if (yv.w <= 0.0f) d.w = 0.0f;
+      }
+      const float4 xv = *reinterpret_cast<const float4*>(&x[r * C + c]);
+      sd.x += d.x; sd.y += d.y; sd.z += d.z; sd.w += d.w;
+      sdx.x += d.x * (xv.x - mean.x) * invstd.x;
+      sdx.y += d.y * (xv.y - mean.y) * invstd.y;
+      sdx.z += d.z * (xv.z - mean.z) * invstd.z;
+      sdx.w += d.w * (xv.w - mean.w) * invstd.w;
+    }
+  }
+  __shared__ float4 ls[16][16], ls2[16][16];
+  ls[rg][cl] = sd;
+  ls2[rg][cl] = sdx;
+  __syncthreads();
+  if (rg == 0 && ok) {
+#pragma unroll
+    for (int k = 1; k < 16; ++k) {
+      const float4 a = ls[k][cl], b = ls2[k][cl];
+      sd.x += a.x; sd.y += a.y; sd.z += a.z; sd.w += a.w;
+      sdx.x += b.x; sdx.y += b.y; sdx.z += b.z; sdx.w += b.w;
+    }
+    atomicAdd(&red[c + 0], sd.x);
+    atomicAdd(&red[c + 1], sd.y);
+    atomicAdd(&red[c + 2], sd.z);
+    atomicAdd(&red[c + 3], sd.w);
+    atomicAdd(&red[C + c + 0], sdx.x);
+    atomicAdd(&red[C + c + 1], sdx.y);
+    atomicAdd(&red[C + c + 2], sdx.z);
+    atomicAdd(&red[C + c + 3], sdx.w);
+  }
+}
+
+__global__ void bn_nhwc_bwd_dx_v4_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    const float* __restrict__ y, float* __restrict__ dx,
+    const float* __restrict__ sm, const float* __restrict__ si,
+    const float* __restrict__ gamma, const float* __restrict__ red,
+    float* __restrict__ dres, long rows, int C, int relu) {
+  const int cl = threadIdx.x & 15;
+  const int rg = threadIdx.x >> 4;
+  const int c = blockIdx.x * BNC + cl * 4;
+  if (c >= C) return;
+  const float4 mean = *reinterpret_cast<const float4*>(&sm[c]);
+  const float4 invstd = *reinterpret_cast<const float4*>(&si[c]);
+  const float cnt = (float)rows;
+  float4 scale, mdy, mdyx;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const float g = gamma ? gamma[c + j] : 1.0f;
+    reinterpret_cast<float*>(&scale)[j] =
+        g * reinterpret_cast<const float*>(&invstd)[j];
+    reinterpret_cast<float*>(&mdy)[j] = red[c + j] / cnt;
+    reinterpret_cast<float*>(&mdyx)[j] = red[C + c + j] / cnt;
+  }
+  for (long r = (long)blockIdx.y * 16 + rg; r < rows;
+       r += (long)gridDim.y * 16) {
+    float4 d = *reinterpret_cast<const float4*>(&dy[r * C + c]);
+    if (relu) {
+      const float4 yv = *reinterpret_cast<const float4*>(&y[r * C + c]);
+      if (yv.x <= 0.0f) d.x = 0.0f;
+      if (yv.y <= 0.0f) d.y = 0.0f;
+      if (yv.z <= 0.0f) d.z = 0.0f;
+      if (yv.w <= 0.0f) d.w = 0.0f;
+    }
+    if (dres) *reinterpret_cast<float4*>(&dres[r * C + c]) = d;
+    const float4 xv = *reinterpret_cast<const float4*>(&x[r * C + c]);
+    float4 o;
+    o.x = scale.x * (d.x - mdy.x - (xv.x - mean.x) * invstd.x * mdyx.x);
+    o.y = scale.y * (d.y - mdy.y - (xv.y - mean.y) * invstd.y * mdyx.y);
+    o.z = scale.z * (d.z - mdy.z - (xv.z - mean.z) * invstd.z * mdyx.z);
+    o.w = scale.w * (d.w - mdy.w - (xv.w - mean.w) * invstd.w * mdyx.w);
+    *reinterpret_cast<float4*>(&dx[r * C + c]) = o;
+  }
+}
+
 __global__ void bn_nhwc_stats_kernel(const float* __restrict__ x,
                                      float* __restrict__ stats, long rows,
                                      int C) {
@@ -378,32 +577,55 @@ static inline dim3 bn_nhwc_grid(long rows, int C) {
   return dim3(cgrids, (int)chunks);
 }
 
+static inline dim3 bn_nhwc_grid_v4(long rows, int C) {
+  const int cgrids = (C + BNC - 1) / BNC;
+  long chunks = (rows + 15) / 16;
+  chunks = std::min(chunks, (long)std::max(1, 2048 / cgrids));
+  return dim3(cgrids, (int)chunks);
+}
+
 std::vector<torch::Tensor> bn2d_nhwc_fwd(torch::Tensor x,
                                          torch::Tensor gamma,
                                          torch::Tensor beta,
                                          torch::Tensor run_mean,
                                          torch::Tensor run_var,
                                          double momentum, double eps,
-                                         bool relu) {
+                                         bool relu, torch::Tensor res) {
   check_f32n(x, "x");
   const int C = x.size(-1);
   const long rows = x.numel() / C;
+  const bool v4 = (C & 3) == 0;
+  TORCH_CHECK(res.numel() == 0 || (v4 && res.numel() == x.numel()),
+              "fused residual needs C%4==0 and matching shape");
   auto y = torch::empty_like(x);
   auto stats = torch::zeros({2 * C}, x.options());
   auto save_mean = torch::empty({C}, x.options());
   auto save_invstd = torch::empty({C}, x.options());
   auto stream = at::cuda::getCurrentCUDAStream();
+  const float* gp = gamma.numel() ? gamma.data_ptr<float>() : nullptr;
+  const float* bp = beta.numel() ? beta.data_ptr<float>() : nullptr;
+  float* rmp = run_mean.numel() ? run_mean.data_ptr<float>() : nullptr;
+  float* rvp = run_var.numel() ? run_var.data_ptr<float>() : nullptr;
+  if (v4) {
+    const dim3 grid = bn_nhwc_grid_v4(rows, C);
+    hipLaunchKernelGGL(bn_nhwc_stats_v4_kernel, grid, dim3(256), 0, stream,
+                       x.data_ptr<float>(), stats.data_ptr<float>(), rows,
+                       C);
+    hipLaunchKernelGGL(
+        bn_nhwc_norm_v4_kernel, grid, dim3(256), 0, stream,
+        x.data_ptr<float>(), y.data_ptr<float>(), stats.data_ptr<float>(),
+        gp, bp, save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
+        rmp, rvp, res.numel() ? res.data_ptr<float>() : nullptr,
+        (float)momentum, (float)eps, rows, C, (int)relu);
+    return {y, save_mean, save_invstd};
+  }
   const dim3 grid = bn_nhwc_grid(rows, C);
   hipLaunchKernelGGL(bn_nhwc_stats_kernel, grid, dim3(256), 0, stream,
                      x.data_ptr<float>(), stats.data_ptr<float>(), rows, C);
   hipLaunchKernelGGL(
       bn_nhwc_norm_kernel, grid, dim3(256), 0, stream, x.data_ptr<float>(),
-      y.data_ptr<float>(), stats.data_ptr<float>(),
-      gamma.numel() ? gamma.data_ptr<float>() : nullptr,
-      beta.numel() ? beta.data_ptr<float>() : nullptr,
-      save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
-      run_mean.numel() ? run_mean.data_ptr<float>() : nullptr,
-      run_var.numel() ? run_var.data_ptr<float>() : nullptr,
+      y.data_ptr<float>(), stats.data_ptr<float>(), gp, bp,
+      save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(), rmp, rvp,
       (float)momentum, (float)eps, rows, C, (int)relu);
   return {y, save_mean, save_invstd};
 }
@@ -413,16 +635,35 @@ std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
                                          torch::Tensor gamma,
                                          torch::Tensor save_mean,
                                          torch::Tensor save_invstd,
-                                         bool relu) {
+                                         bool relu, bool want_dres) {
   check_f32n(x, "x");
   check_f32n(dy, "dy");
   const int C = x.size(-1);
   const long rows = x.numel() / C;
+  const bool v4 = (C & 3) == 0;
+  TORCH_CHECK(!want_dres || v4, "dres needs C%4==0");
   auto red = torch::zeros({2 * C}, x.options());
   auto dx = torch::empty_like(x);
+  auto dres = want_dres ? torch::empty_like(x) : torch::empty({0}, x.options());
   auto stream = at::cuda::getCurrentCUDAStream();
-  const dim3 grid = bn_nhwc_grid(rows, C);
   const float* yp = relu ? y.data_ptr<float>() : nullptr;
+  const float* gp = gamma.numel() ? gamma.data_ptr<float>() : nullptr;
+  if (v4) {
+    const dim3 grid = bn_nhwc_grid_v4(rows, C);
+    hipLaunchKernelGGL(bn_nhwc_bwd_reduce_v4_kernel, grid, dim3(256), 0,
+                       stream, x.data_ptr<float>(), dy.data_ptr<float>(),
+                       yp, save_mean.data_ptr<float>(),
+                       save_invstd.data_ptr<float>(), red.data_ptr<float>(),
+                       rows, C, (int)relu);
+    hipLaunchKernelGGL(
+        bn_nhwc_bwd_dx_v4_kernel, grid, dim3(256), 0, stream,
+        x.data_ptr<float>(), dy.data_ptr<float>(), yp, dx.data_ptr<float>(),
+        save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(), gp,
+        red.data_ptr<float>(),
+        want_dres ? dres.data_ptr<float>() : nullptr, rows, C, (int)relu);
+    return {dx, red.narrow(0, C, C), red.narrow(0, 0, C), dres};
+  }
+  const dim3 grid = bn_nhwc_grid(rows, C);
   hipLaunchKernelGGL(bn_nhwc_bwd_reduce_kernel, grid, dim3(256), 0, stream,
                      x.data_ptr<float>(), dy.data_ptr<float>(), yp,
                      save_mean.data_ptr<float>(),
@@ -431,8 +672,7 @@ std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
   hipLaunchKernelGGL(
       bn_nhwc_bwd_dx_kernel, grid, dim3(256), 0, stream,
       x.data_ptr<float>(), dy.data_ptr<float>(), yp, dx.data_ptr<float>(),
-      save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
-      gamma.numel() ? gamma.data_ptr<float>() : nullptr,
+      save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(), gp,
       red.data_ptr<float>(), rows, C, (int)relu);
-  return {dx, red.narrow(0, C, C), red.narrow(0, 0, C)};
+  return {dx, red.narrow(0, C, C), red.narrow(0, 0, C), dres};
 }

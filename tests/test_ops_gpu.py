@@ -558,3 +558,39 @@ for (N,H,W,Cin,Cout,K,stride,pad) in [(12,56,56,8,64,3,1,1),
     import torch as _t
 
     _t.testing.assert_close(outs["0"], outs["1"], atol=2e-4, rtol=2e-4)
+
+
+def test_batchnorm2d_nhwc_fused_residual_gpu():
+    """out = relu(bn(x) + res) fused into the norm kernel; d_res is the
+    relu-masked dy (the ResNet block boundary, models/cnn.py)."""
+    torch.manual_seed(17)
+    N, H, W, C = 5, 9, 9, 32
+    x = torch.randn(N, H, W, C)
+    res = torch.randn(N, H, W, C)
+    g = torch.rand(C) + 0.5
+    b = torch.randn(C)
+    proj = torch.randn(N, H, W, C)
+
+    def run(device):
+        from sparkflow_amd.ops.functional import batchnorm2d_act_nhwc
+
+        xi = x.clone().to(device).requires_grad_()
+        ri = res.clone().to(device).requires_grad_()
+        gi = g.clone().to(device).requires_grad_()
+        bi = b.clone().to(device).requires_grad_()
+        rm = torch.zeros(C, device=device)
+        rv = torch.ones(C, device=device)
+        if device == "cpu":  # plain-torch reference of the same math
+            xc = xi.permute(0, 3, 1, 2)
+            y = torch.nn.functional.batch_norm(
+                xc, rm, rv, gi, bi, True, 0.1, 1e-5)
+            y = torch.relu(y.permute(0, 2, 3, 1) + ri)
+        else:
+            y = batchnorm2d_act_nhwc(xi, gi, bi, rm, rv, 0.1, 1e-5,
+                                     relu=True, training=True, residual=ri)
+        (y * proj.to(device)).sum().backward()
+        return (y.detach().cpu(), xi.grad.cpu(), ri.grad.cpu(),
+                gi.grad.cpu(), bi.grad.cpu(), rm.cpu(), rv.cpu())
+
+    for gg, cc in zip(run(DEV), run("cpu")):
+        torch.testing.assert_close(gg, cc, atol=2e-4, rtol=2e-4)
