@@ -376,18 +376,26 @@ class Conv2dActNHWC(torch.autograd.Function):
             if not need_dx:
                 dx = None
             else:
-                # dgrad IS a stride-1 conv of (zero-stuffed) dpre with
-                # spatially-flipped, in/out-swapped weights
+                # dgrad IS a stride-1 conv of dpre with spatially-
+                # flipped, in/out-swapped weights.  stride > 1 uses the
+                # direct s^2-sub-convolution kernel (no zero-stuffed
+                # dilation: the dilated form ran s^2 x the useful MFMA
+                # work); the dilate path stays as the Cout%4!=0 fallback.
                 wt = w.flip(1, 2).permute(3, 1, 2, 0).contiguous()
                 if stride == 1:
-                    src = dpre
+                    dx = ops.hip.conv_nhwc_fwd(
+                        dpre, wt, torch.empty(0, device=x.device), 1,
+                        kk - 1 - pad, ACT_NONE)
+                elif w.shape[0] % 4 == 0:
+                    dx = ops.hip.conv_nhwc_dgrad_strided(
+                        dpre, wt, x.shape[1], x.shape[2], stride, pad)
                 else:
                     rh = (x.shape[1] + 2 * pad - kk) % stride
                     rw = (x.shape[2] + 2 * pad - kk) % stride
                     src = ops.hip.conv_nhwc_dilate(dpre, stride, rh, rw)
-                dx = ops.hip.conv_nhwc_fwd(
-                    src, wt, torch.empty(0, device=x.device), 1,
-                    kk - 1 - pad, ACT_NONE)
+                    dx = ops.hip.conv_nhwc_fwd(
+                        src, wt, torch.empty(0, device=x.device), 1,
+                        kk - 1 - pad, ACT_NONE)
             dw = ops.hip.conv_nhwc_wgrad(x, dpre, kk, w.shape[2], stride,
                                          pad)
             db = ops.hip.conv_nhwc_bias_grad(dpre) if has_bias else None

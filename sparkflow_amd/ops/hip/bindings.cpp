@@ -98,6 +98,9 @@ std::vector<torch::Tensor> maxpool_nhwc_fwd(torch::Tensor x, int64_t k,
 torch::Tensor maxpool_nhwc_bwd(torch::Tensor dy, torch::Tensor idx,
                                int64_t H, int64_t W, int64_t k,
                                int64_t stride, int64_t pad);
+torch::Tensor conv_nhwc_dgrad_strided(torch::Tensor dy, torch::Tensor wt,
+                                      int64_t H, int64_t W, int64_t stride,
+                                      int64_t pad);
 
 // norm.hip — training-mode BatchNorm2d (+ fused ReLU)
 std::vector<torch::Tensor> bn2d_fwd(torch::Tensor x, torch::Tensor gamma,
@@ -211,6 +214,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "NHWC conv bias grad (channel-contiguous column sums)");
   m.def("maxpool_nhwc_fwd", &maxpool_nhwc_fwd,
         "NHWC maxpool fwd with argmax indices");
+  m.def("conv_nhwc_dgrad_strided", &conv_nhwc_dgrad_strided,
+        "direct stride>1 NHWC dgrad (s^2 sub-convolutions, no dilation)");
   m.def("maxpool_nhwc_bwd", &maxpool_nhwc_bwd,
         "NHWC maxpool bwd (float4 window-gather; scalar scatter fallback)");
   m.def("bn2d_fwd", &bn2d_fwd,

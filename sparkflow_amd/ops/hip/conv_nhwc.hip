@@ -478,6 +478,164 @@ __global__ __launch_bounds__(256) void conv_nhwc_wgrad_kernel(
   }
 }
 
+// ---------------- direct strided dgrad ----------------
+// dgrad for stride s > 1 WITHOUT zero-stuffing: the dilated-dy
+// formulation (dilate + stride-1 fwd conv) runs s^2 = 4x the useful
+// MFMA work on zeros (plus a dilate kernel and a big fill).  Here dx is
+// computed directly as s^2 sub-convolutions: output pixels of class
+// (h%s, w%s) see only the filter taps kh ≡ (p'-h) (mod s) — the K
+// dimension per class is nkh*nkw*Cout ≈ K/s^2, and dy is read
+// unstrided.  blockIdx.z picks the class; same 64x64/BK16
+// mfma_16x16x4 scheme as the fwd kernel (measured best, see
+// profiles/conv_micro_r02.md).  Requires Cout % 4 == 0.
+//
+//   dx[n, h, w, ci] = sum_{khi,kwi,co} dy[n, hh+khi+oh_off,
+//                        ww+kwi+ow_off, co] * wt[ci, kh0+khi*s,
+//                        kw0+kwi*s, co]
+// with h = a + s*hh, kh0 = (p'-a) mod s, oh_off = (a+kh0-p')/s and the
+// same for w/b/kw0/ow_off; wt = w.flip(1,2).permute(3,1,2,0) exactly as
+// the dilated path's operand (ops/functional.py Conv2dActNHWC.backward).
+__global__ __launch_bounds__(256) void conv_nhwc_dgrad_s_kernel(
+    const float* __restrict__ dy, const float* __restrict__ wt,
+    float* __restrict__ dx, ConvShapeHW s_, int pprime) {
+  // geometry for this class --------------------------------------------
+  const int S = s_.stride;
+  const int a = blockIdx.z / S, b = blockIdx.z % S;
+  const int kh0 = ((pprime - a) % S + S) % S;
+  const int kw0 = ((pprime - b) % S + S) % S;
+  const int nkh = s_.Kh > kh0 ? (s_.Kh - kh0 + S - 1) / S : 0;
+  const int nkw = s_.Kw > kw0 ? (s_.Kw - kw0 + S - 1) / S : 0;
+  const int Ha = s_.H > a ? (s_.H - a + S - 1) / S : 0;
+  const int Wb = s_.W > b ? (s_.W - b + S - 1) / S : 0;
+  const int oh_off = (a + kh0 - pprime) / S;
+  const int ow_off = (b + kw0 - pprime) / S;
+  const int M = s_.N * Ha * Wb;
+  const int K = nkh * nkw * s_.Cout;  // may be 0 -> nothing to write
+  const int Nc = s_.Cin;              // GEMM columns = input channels
+
+  __shared__ float As[BM][BK + 1];
+  __shared__ float Bs[BK][BN + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = (wv >> 1) * 32;
+  const int wn = (wv & 1) * 32;
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+  if (m0 >= M || K == 0) return;  // classes have different M grids
+
+  f32x4 acc[2][2] = {};
+
+  const int st_row = t >> 2, st_c4 = (t & 3) * 4;
+  // A: thread owns dx pixel gm = m0 + st_row, k-slot st_c4 (4
+  // consecutive co inside one (khi, kwi) — Cout % 4 == 0)
+  const int gm = m0 + st_row;
+  const int m_dec = gm < M ? gm : 0;
+  const int a_ww = m_dec % Wb;
+  const int a_hh = (m_dec / Wb) % Ha;
+  const int a_n = m_dec / (Wb * Ha);
+  const long dy_nbase = (long)a_n * s_.Ho * s_.Wo * s_.Cout;
+  // (khi, kwi, co) cursor for k-slot st_c4, advanced by BK per step
+  int a_co = st_c4 % s_.Cout;
+  int a_kk = st_c4 / s_.Cout;  // khi*nkw + kwi
+
+  auto load_a = [&](int k0) -> float4 {
+    // slots never straddle K: slot starts and K are both %4 == 0
+    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gm >= M || k0 + st_c4 >= K) return r;
+    const int khi = a_kk / nkw, kwi = a_kk - (a_kk / nkw) * nkw;
+    const int oh = a_hh + khi + oh_off;
+    const int ow = a_ww + kwi + ow_off;
+    if (oh >= 0 && oh < s_.Ho && ow >= 0 && ow < s_.Wo) {
+      r = *reinterpret_cast<const float4*>(
+          &dy[dy_nbase + ((long)oh * s_.Wo + ow) * s_.Cout + a_co]);
+    }
+    return r;
+  };
+  auto advance_a = [&]() {
+    a_co += BK;
+    while (a_co >= s_.Cout) { a_co -= s_.Cout; ++a_kk; }
+  };
+  // B: thread owns wt row ci = n0 + st_row, k-slot st_c4; same cursor
+  int b_co = st_c4 % s_.Cout;
+  int b_kk = st_c4 / s_.Cout;
+  const int gci = n0 + st_row;
+  const long wt_row = (long)(gci < Nc ? gci : 0) * s_.Kh * s_.Kw * s_.Cout;
+
+  auto load_b = [&](int k0) -> float4 {
+    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gci >= Nc || k0 + st_c4 >= K) return r;
+    const int khi = b_kk / nkw, kwi = b_kk - (b_kk / nkw) * nkw;
+    const int kh = kh0 + khi * S, kw = kw0 + kwi * S;
+    r = *reinterpret_cast<const float4*>(
+        &wt[wt_row + ((long)kh * s_.Kw + kw) * s_.Cout + b_co]);
+    return r;
+  };
+  auto advance_b = [&]() {
+    b_co += BK;
+    while (b_co >= s_.Cout) { b_co -= s_.Cout; ++b_kk; }
+  };
+
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  float4 ra = load_a(0);
+  float4 rb = load_b(0);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    {
+      const float* pa = reinterpret_cast<const float*>(&ra);
+      const float* pb = reinterpret_cast<const float*>(&rb);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) As[st_row][st_c4 + i] = pa[i];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) Bs[st_c4 + i][st_row] = pb[i];
+    }
+    __syncthreads();
+    if (k0 + BK < K) {
+      advance_a();
+      advance_b();
+      ra = load_a(k0 + BK);
+      rb = load_b(k0 + BK);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      const float a0 = As[wm + fr][kk * 4 + fk];
+      const float a1 = As[wm + 16 + fr][kk * 4 + fk];
+      const float b0 = Bs[kk * 4 + fk][wn + fr];
+      const float b1 = Bs[kk * 4 + fk][wn + 16 + fr];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: dx[(n, a + s*hh, b + s*ww), ci] — ci-contiguous stores
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int ci = n0 + wn + ni * 16 + fcol;
+      if (ci >= Nc) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int m = m0 + wm + mi * 16 + frow + i;
+        if (m < M) {
+          const int ww = m % Wb;
+          const int hh = (m / Wb) % Ha;
+          const int n = m / (Wb * Ha);
+          dx[(((long)n * s_.H + a + (long)S * hh) * s_.W + b +
+              (long)S * ww) * s_.Cin + ci] = acc[mi][ni][i];
+        }
+      }
+    }
+  }
+}
+
 // Zero-stuff dy (NHWC) for strided dgrad-as-convolution: rows stay
 // channel-contiguous (a row copy is C contiguous floats).
 __global__ void dilate_nhwc_kernel(const float* __restrict__ dy,
@@ -601,6 +759,42 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
                        dw.data_ptr<float>(), s, M, Nc, K, K);
   }
   return dw;
+}
+
+// dy: [N, Ho, Wo, Cout] (act-bwd'ed), wt: [Cin, Kh, Kw, Cout]
+// (= w.flip(1,2).permute(3,1,2,0)) -> dx: [N, H, W, Cin].
+torch::Tensor conv_nhwc_dgrad_strided(torch::Tensor dy, torch::Tensor wt,
+                                      int64_t H, int64_t W, int64_t stride,
+                                      int64_t pad) {
+  check_f32w(dy, "dy");
+  check_f32w(wt, "wt");
+  ConvShapeHW s;
+  s.N = dy.size(0); s.Ho = dy.size(1); s.Wo = dy.size(2);
+  s.Cout = dy.size(3);
+  s.Cin = wt.size(0); s.Kh = wt.size(1); s.Kw = wt.size(2);
+  TORCH_CHECK(wt.size(3) == s.Cout, "Cout mismatch");
+  TORCH_CHECK((s.Cout & 3) == 0, "strided dgrad needs Cout % 4 == 0");
+  s.H = H; s.W = W; s.stride = stride; s.pad = pad;
+  const int S = stride;
+  const int pprime = s.Kh - 1 - pad;
+  // any class with zero taps leaves its dx pixels unwritten -> zeros
+  bool any_empty = false;
+  for (int a = 0; a < S && !any_empty; ++a) {
+    const int kh0 = ((pprime - a) % S + S) % S;
+    if (s.Kh <= kh0) any_empty = true;
+  }
+  auto dx = any_empty
+                ? torch::zeros({s.N, (long)H, (long)W, s.Cin},
+                               dy.options())
+                : torch::empty({s.N, (long)H, (long)W, s.Cin},
+                               dy.options());
+  const long maxM = (long)s.N * ((H + S - 1) / S) * ((W + S - 1) / S);
+  dim3 grid(ceil_div(s.Cin, BN), ceil_div((int)maxM, BM), S * S);
+  hipLaunchKernelGGL(conv_nhwc_dgrad_s_kernel, grid, dim3(256), 0,
+                     at::cuda::getCurrentCUDAStream(),
+                     dy.data_ptr<float>(), wt.data_ptr<float>(),
+                     dx.data_ptr<float>(), s, pprime);
+  return dx;
 }
 
 torch::Tensor conv_nhwc_dilate(torch::Tensor dy, int64_t s, int64_t extra_h,
