@@ -92,6 +92,43 @@ class PSCore:
         self.error_count = 0
         self.update_count = 0
         self.pull_count = 0
+        # Lock-mode stream-ordered consistency (SPARKFLOW_LOCK_EVENTS=0
+        # falls back to device syncs): instead of torch.cuda.synchronize
+        # inside every read/write critical section (full host
+        # serialization), readers wait on the last apply's EVENT and
+        # writers wait on all outstanding readers' events — the RWLock
+        # orders the ENQUEUES, the event chain orders the device work,
+        # and the host never blocks.
+        import os
+
+        self._lock_events = (self.device.type == "cuda"
+                             and os.environ.get("SPARKFLOW_LOCK_EVENTS",
+                                                "1") != "0")
+        self._apply_ev = None
+        self._read_evs = []
+        self._ev_mu = threading.Lock()
+
+    def _reader_sync_in(self):
+        if self._apply_ev is not None:
+            torch.cuda.current_stream(self.device).wait_event(self._apply_ev)
+
+    def _reader_sync_out(self):
+        ev = torch.cuda.Event()
+        ev.record(torch.cuda.current_stream(self.device))
+        with self._ev_mu:
+            self._read_evs.append(ev)
+
+    def _writer_sync_in(self):
+        with self._ev_mu:
+            evs, self._read_evs = self._read_evs, []
+        s = torch.cuda.current_stream(self.device)
+        for ev in evs:
+            s.wait_event(ev)
+
+    def _writer_sync_out(self):
+        ev = torch.cuda.Event()
+        ev.record(torch.cuda.current_stream(self.device))
+        self._apply_ev = ev
 
     # -- pulls ---------------------------------------------------------------
 
@@ -100,6 +137,11 @@ class PSCore:
         self.pull_count += 1
         if self.acquire_lock:
             with self.lock.reading():
+                if self._lock_events:
+                    self._reader_sync_in()
+                    w = self.weights.clone()
+                    self._reader_sync_out()
+                    return w
                 w = self.weights.clone()
                 if self.device.type == "cuda":
                     torch.cuda.synchronize(self.device)
@@ -116,6 +158,11 @@ class PSCore:
         self.pull_count += 1
         if self.acquire_lock:
             with self.lock.reading():
+                if self._lock_events:
+                    self._reader_sync_in()
+                    send_fn(self.weights)
+                    self._reader_sync_out()
+                    return
                 send_fn(self.weights)
                 if self.device.type == "cuda":
                     torch.cuda.synchronize(self.device)
@@ -131,9 +178,14 @@ class PSCore:
                     f"gradient size {grad_flat.numel()} != model size {self.numel}")
             if self.acquire_lock:
                 with self.lock.writing():
-                    self.optimizer.apply(self.weights, grad_flat)
-                    if self.device.type == "cuda":
-                        torch.cuda.synchronize(self.device)
+                    if self._lock_events:
+                        self._writer_sync_in()
+                        self.optimizer.apply(self.weights, grad_flat)
+                        self._writer_sync_out()
+                    else:
+                        self.optimizer.apply(self.weights, grad_flat)
+                        if self.device.type == "cuda":
+                            torch.cuda.synchronize(self.device)
             else:
                 self.optimizer.apply(self.weights, grad_flat)
             self.update_count += 1
