@@ -128,7 +128,7 @@ def main():
     rng = np.random.default_rng(1000 + rank)
     ev_rng = np.random.default_rng(999)
     if args.task == "teacher":
-        noise = 0.4 if args.noise is None else args.noise
+        noise = 0.15 if args.noise is None else args.noise
         teacher = make_teacher()
         x, y = make_teacher_task(rng, args.train_rows, teacher, noise)
         if rank == 0:
