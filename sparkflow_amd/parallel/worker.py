@@ -269,6 +269,8 @@ class Worker:
                           for _ in range(self._IDX_POOL_CAP)])
         g["idx_pool"].copy_(torch.as_tensor(picks, device=self.device))
         g["ctr"].zero_()
+        if g.get("done") is not None:
+            g["done"].zero_()  # ticket is 0 at quiescence; re-assert
         self._pool_used = 0
 
     def maybe_refresh_pool(self, k: int) -> None:
@@ -500,8 +502,16 @@ class Worker:
         with torch.cuda.graph(graph):
             body()
 
+        # `done` (the gather's done-ticket word) MUST stay referenced:
+        # the captured kernel keeps using its address forever, and if
+        # the tensor were garbage-collected the allocator would hand the
+        # block to ANY later allocation (an eval forward, a checkpoint
+        # copy), whose writes then corrupt the ticket and silently
+        # freeze the minibatch cursor — the round-2 time-to-accuracy
+        # plateau bug (every replay re-training the same minibatch).
         self._graph_state = {"graph": graph, "xb": xb, "yb": yb,
                              "idx_pool": idx_pool, "ctr": ctr,
+                             "done": done,
                              "full_step": full_step,
                              "workspaces": getattr(self, "_graph_ws", None),
                              "gather_inside": gather_inside}
