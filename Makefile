@@ -4,7 +4,7 @@
 
 PY ?= python
 
-.PHONY: build test test-gpu bench clean
+.PHONY: build test test-gpu bench ci cluster-demo clean
 
 build:
 	PYTORCH_ROCM_ARCH=gfx950 $(PY) -c 'import __graft_entry__ as g; g.build()'
@@ -17,6 +17,18 @@ test-gpu:
 
 bench:
 	$(PY) bench.py --gpus 1 --steps 500 --warmup 100
+
+# CI entry (parity: reference .travis.yml:16-18 runs the Spark test
+# suite in docker): full CPU test run INCLUDING the executed Spark-ML
+# suite (real pyspark if installed, bundled localspark otherwise).
+ci: build test
+
+# one-command multi-process cluster demo (parity: reference
+# docker-compose.yml:1-22 Spark master + worker cluster) — N spawned
+# executor processes run fit/transform/pipeline-save-load end to end,
+# with fit driving the torch.distributed PS star.
+cluster-demo:
+	$(PY) examples/cluster_demo.py --executors 2
 
 clean:
 	rm -rf sparkflow_amd/ops/_build build *.egg-info
