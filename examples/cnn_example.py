@@ -5,18 +5,31 @@ from pathlib import Path
 
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
+from sparkflow_amd import localspark
+
+localspark.install(quiet=True)  # no-op when a real pyspark is installed
+
 from pyspark.ml.feature import VectorAssembler
 from pyspark.sql import SparkSession
 
+from examples._data import load_mnist_df
 from sparkflow_amd.graph_utils import build_graph
 from sparkflow_amd.models import mnist_cnn
 from sparkflow_amd.spark import SparkAsyncDL
 
 
 def main():
-    spark = SparkSession.builder.appName("sparkflow-amd cnn").getOrCreate()
-    df = (spark.read.option("inferSchema", "true")
-          .csv("examples/mnist_train.csv"))
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rows", type=int, default=2000)
+    ap.add_argument("--iters", type=int, default=None)
+    args = ap.parse_args()
+
+    spark = (SparkSession.builder.master("local[2]")
+             .appName("sparkflow-amd cnn")
+             .config("spark.driver.host", "127.0.0.1").getOrCreate())
+    df = load_mnist_df(spark, rows=args.rows)
     va = VectorAssembler(inputCols=df.columns[1:785], outputCol="features")
 
     spark_model = SparkAsyncDL(
@@ -26,7 +39,7 @@ def main():
         tfLabel="_c0",
         tfOutput="out:0",
         tfLearningRate=.001,
-        iters=20,
+        iters=args.iters or 20,
         predictionCol="predicted",
         labelCol="_c0",
         partitions=8,

@@ -9,10 +9,15 @@ from pathlib import Path
 
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
+from sparkflow_amd import localspark
+
+localspark.install(quiet=True)  # no-op when a real pyspark is installed
+
 from pyspark.ml.feature import OneHotEncoder, VectorAssembler
 from pyspark.ml.pipeline import Pipeline, PipelineModel
 from pyspark.sql import SparkSession
 
+from examples._data import load_mnist_df
 from sparkflow_amd.graph_utils import build_graph
 from sparkflow_amd.models import mnist_mlp
 from sparkflow_amd.spark import SparkAsyncDL
@@ -20,9 +25,17 @@ from sparkflow_amd.spark.pipeline_util import PysparkPipelineWrapper
 
 
 def main():
-    spark = SparkSession.builder.appName("sparkflow-amd simple_dnn").getOrCreate()
-    df = (spark.read.option("inferSchema", "true")
-          .csv("examples/mnist_train.csv"))
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rows", type=int, default=2000)
+    ap.add_argument("--iters", type=int, default=None)
+    args = ap.parse_args()
+
+    spark = (SparkSession.builder.master("local[2]")
+             .appName("sparkflow-amd simple_dnn")
+             .config("spark.driver.host", "127.0.0.1").getOrCreate())
+    df = load_mnist_df(spark, rows=args.rows)
 
     graph_json = build_graph(mnist_mlp)
     va = VectorAssembler(inputCols=df.columns[1:785], outputCol="features")
@@ -35,7 +48,7 @@ def main():
         tfLabel="_c0",          # class-index labels
         tfOutput="out:0",
         tfLearningRate=.001,
-        iters=50,
+        iters=args.iters or 50,
         predictionCol="predicted",
         labelCol="_c0",
         miniBatchSize=300,
