@@ -257,39 +257,75 @@ class Worker:
             if sub:
                 self._fill_idx_pool(n, mbs)
 
-    def _fill_idx_pool(self, n: int, mbs: int) -> None:
-        """(Re)draw the device-side pool of sample-without-replacement
-        minibatches and rewind the captured cursor.  Keeping the fill out
-        of the steady-state loop keeps host RNG off the hot path; the
-        pool is REDRAWN once its minibatches have all been consumed (see
-        maybe_refresh_pool) so long runs don't cycle the same
-        _IDX_POOL_CAP batches forever."""
+    def _draw_pool_picks(self, n: int, mbs: int) -> np.ndarray:
+        """CAP sample-without-replacement minibatches (the reference's
+        np.random.choice(n, mbs, replace=False), ml_util.py:118).  The
+        dedicated rng keeps background refills race-free vs the eager
+        paths' self.rng."""
+        rng = getattr(self, "_pool_rng", None)
+        if rng is None:
+            seed = None if self.config.seed is None \
+                else self.config.seed + 7919
+            rng = self._pool_rng = np.random.default_rng(seed)
+        return np.stack([rng.choice(n, mbs, replace=False)
+                         for _ in range(self._IDX_POOL_CAP)])
+
+    def _install_pool(self, picks: np.ndarray) -> None:
         g = self._graph_state
-        picks = np.stack([self.rng.choice(n, mbs, replace=False)
-                          for _ in range(self._IDX_POOL_CAP)])
         g["idx_pool"].copy_(torch.as_tensor(picks, device=self.device))
         g["ctr"].zero_()
         if g.get("done") is not None:
             g["done"].zero_()  # ticket is 0 at quiescence; re-assert
         self._pool_used = 0
+        self._pool_next = None
+
+    def _fill_idx_pool(self, n: int, mbs: int) -> None:
+        """(Re)draw the device-side pool of minibatch index rows and
+        rewind the captured cursor.  The pool is REDRAWN once consumed
+        (see maybe_refresh_pool) so long runs don't cycle the same
+        _IDX_POOL_CAP batches forever."""
+        self._install_pool(self._draw_pool_picks(n, mbs))
 
     def maybe_refresh_pool(self, k: int) -> None:
-        """Called before enqueueing k replays: if the pool would wrap,
-        wait for in-flight replays and redraw it.  The device sync is
-        once per _IDX_POOL_CAP steps (~55 ms of GPU work) so the cost is
-        noise; callers must invoke this while no replay is being
-        enqueued concurrently (all run paths do)."""
+        """Called before enqueueing k replays: once the pool is half
+        consumed a BACKGROUND thread draws the next CAP minibatches (the
+        host RNG overlaps GPU compute instead of stalling the step loop
+        — the per-row choice() loop costs ~100 ms that a round-1-style
+        synchronous refill would put straight into training time); when
+        the pool would wrap, join the draw, wait for in-flight replays
+        and install it (one H2D copy).  Callers must invoke this while
+        no replay is being enqueued concurrently (all run paths do)."""
         g = getattr(self, "_graph_state", None)
         if not g or g.get("idx_pool") is None:
             return
         used = getattr(self, "_pool_used", 0)
-        if used + k > self._IDX_POOL_CAP:
+        cap = self._IDX_POOL_CAP
+        n, mbs = self.x.shape[0], g["idx_pool"].shape[1]
+        if used + k > cap // 2 and getattr(self, "_pool_thread", None) \
+                is None and getattr(self, "_pool_next", None) is None:
+            import threading
+
+            def _draw():
+                self._pool_next = self._draw_pool_picks(n, mbs)
+
+            t = threading.Thread(target=_draw, daemon=True)
+            t.start()
+            self._pool_thread = t
+        if used + k > cap:
+            t = getattr(self, "_pool_thread", None)
+            if t is not None:
+                t.join()
+                self._pool_thread = None
+            picks = getattr(self, "_pool_next", None)
+            if picks is None:  # fallback: synchronous draw
+                picks = self._draw_pool_picks(n, mbs)
             if self.device.type == "cuda":
                 torch.cuda.synchronize(self.device)
-            self._fill_idx_pool(self.x.shape[0], g["idx_pool"].shape[1])
+            self._install_pool(picks)
             if self.device.type == "cuda":
                 torch.cuda.synchronize(self.device)
-        self._pool_used = getattr(self, "_pool_used", 0) + k
+            used = 0
+        self._pool_used = used + k
 
     def replay_once(self) -> None:
         """Enqueue ONE training step on the current HIP stream (graph must
