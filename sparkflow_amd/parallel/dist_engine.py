@@ -113,15 +113,23 @@ class EngineHandle:
 
             for w in self.workers:
                 w.ensure_graph(mini_batch)
-                w.maybe_refresh_pool(k)
             # All-local full-step graphs + extension: the whole W x k
-            # Hogwild loop is one GIL-free C++ call (round-robin graph
-            # launches onto W pool streams).
+            # Hogwild loop runs as GIL-free C++ calls (round-robin graph
+            # launches onto W pool streams), in half-pool chunks so each
+            # worker's background pool draw overlaps the previous
+            # chunk's GPU work (Worker.maybe_refresh_pool).
             if _ops.available() and all(
                     w._graph_state["full_step"] for w in self.workers):
                 execs = [w._graph_state["graph"].raw_cuda_graph_exec()
                          for w in self.workers]
-                _ops.hip.multi_replay(execs, k)
+                chunk = max(1, self.workers[0]._IDX_POOL_CAP // 2)
+                done = 0
+                while done < k:
+                    c = min(chunk, k - done)
+                    for w in self.workers:
+                        w.maybe_refresh_pool(c)
+                    _ops.hip.multi_replay(execs, c)
+                    done += c
                 total = 0
                 for w in self.workers:
                     w.finish_steps(k)

@@ -368,15 +368,16 @@ class Worker:
         eager.
         """
         self.ensure_graph(mbs)
-        self.maybe_refresh_pool(k)
         g = self._graph_state
         A = max(1, getattr(self.config, "grad_accum", 1))
         if A > 1 and not g["full_step"]:
+            self.maybe_refresh_pool(k)
             return self._run_steps_graphed_accum(k, mbs, A)
         if not g["full_step"] and hasattr(self.transport, "run_session") \
                 and (g["gather_inside"] or g["xb"] is self.x) \
                 and getattr(self.transport, "native_comm", None) is not None:
             # whole session in one GIL-free C++ call over the raw comm
+            self.maybe_refresh_pool(k)
             try:
                 exec_h = g["graph"].raw_cuda_graph_exec()
             except Exception:
@@ -387,8 +388,16 @@ class Worker:
                 return k * mbs
         if not g["full_step"] and hasattr(self.transport, "run_begin"):
             self.transport.run_begin(k)  # zero per-step control traffic
-        for _ in range(k):
-            self.replay_once()
+        # replay in half-pool chunks so the background pool draw always
+        # has a chunk of GPU time to overlap with (maybe_refresh_pool)
+        chunk = max(1, self._IDX_POOL_CAP // 2)
+        done = 0
+        while done < k:
+            c = min(chunk, k - done)
+            self.maybe_refresh_pool(c)
+            for _ in range(c):
+                self.replay_once()
+            done += c
         self.finish_steps(k)
         return k * mbs
 
