@@ -127,7 +127,16 @@ class EngineHandle:
                 while done < k:
                     c = min(chunk, k - done)
                     for w in self.workers:
-                        w.maybe_refresh_pool(c)
+                        w.pool_prefetch(c)
+                    need = [w for w in self.workers
+                            if w.pool_would_wrap(c)]
+                    if need:  # ONE device drain covers every worker
+                        torch.cuda.synchronize(self.device)
+                        for w in need:
+                            w.pool_install_now()
+                        torch.cuda.synchronize(self.device)
+                    for w in self.workers:
+                        w.pool_account(c)
                     _ops.hip.multi_replay(execs, c)
                     done += c
                 total = 0

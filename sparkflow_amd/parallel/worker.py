@@ -238,7 +238,7 @@ class Worker:
             done += a
         return k * mbs
 
-    _IDX_POOL_CAP = 2048
+    _IDX_POOL_CAP = 8192
 
     def ensure_graph(self, mini_batch: int) -> None:
         """Capture (once) the hipGraph for steps of ``mini_batch`` and fill
@@ -286,24 +286,22 @@ class Worker:
         _IDX_POOL_CAP batches forever."""
         self._install_pool(self._draw_pool_picks(n, mbs))
 
-    def maybe_refresh_pool(self, k: int) -> None:
-        """Called before enqueueing k replays: once the pool is half
-        consumed a BACKGROUND thread draws the next CAP minibatches (the
-        host RNG overlaps GPU compute instead of stalling the step loop
-        — the per-row choice() loop costs ~100 ms that a round-1-style
-        synchronous refill would put straight into training time); when
-        the pool would wrap, join the draw, wait for in-flight replays
-        and install it (one H2D copy).  Callers must invoke this while
-        no replay is being enqueued concurrently (all run paths do)."""
+    def _has_pool(self) -> bool:
         g = getattr(self, "_graph_state", None)
-        if not g or g.get("idx_pool") is None:
+        return bool(g) and g.get("idx_pool") is not None
+
+    def pool_prefetch(self, k: int) -> None:
+        """Kick the background draw of the next pool once this one is
+        half consumed (the ~50-100 ms host RNG overlaps GPU compute)."""
+        if not self._has_pool():
             return
         used = getattr(self, "_pool_used", 0)
-        cap = self._IDX_POOL_CAP
-        n, mbs = self.x.shape[0], g["idx_pool"].shape[1]
-        if used + k > cap // 2 and getattr(self, "_pool_thread", None) \
-                is None and getattr(self, "_pool_next", None) is None:
+        if used + k > self._IDX_POOL_CAP // 2 \
+                and getattr(self, "_pool_thread", None) is None \
+                and getattr(self, "_pool_next", None) is None:
             import threading
+
+            n, mbs = self.x.shape[0], self._graph_state["idx_pool"].shape[1]
 
             def _draw():
                 self._pool_next = self._draw_pool_picks(n, mbs)
@@ -311,21 +309,45 @@ class Worker:
             t = threading.Thread(target=_draw, daemon=True)
             t.start()
             self._pool_thread = t
-        if used + k > cap:
-            t = getattr(self, "_pool_thread", None)
-            if t is not None:
-                t.join()
-                self._pool_thread = None
-            picks = getattr(self, "_pool_next", None)
-            if picks is None:  # fallback: synchronous draw
-                picks = self._draw_pool_picks(n, mbs)
+
+    def pool_would_wrap(self, k: int) -> bool:
+        return self._has_pool() and \
+            getattr(self, "_pool_used", 0) + k > self._IDX_POOL_CAP
+
+    def pool_install_now(self) -> None:
+        """Join the background draw and install it (one H2D copy + ctr
+        rewind).  The caller must have drained in-flight replays of THIS
+        worker first (a device sync covers all workers at once)."""
+        t = getattr(self, "_pool_thread", None)
+        if t is not None:
+            t.join()
+            self._pool_thread = None
+        picks = getattr(self, "_pool_next", None)
+        if picks is None:  # fallback: synchronous draw
+            g = self._graph_state
+            picks = self._draw_pool_picks(self.x.shape[0],
+                                          g["idx_pool"].shape[1])
+        self._install_pool(picks)
+
+    def pool_account(self, k: int) -> None:
+        if self._has_pool():
+            self._pool_used = getattr(self, "_pool_used", 0) + k
+
+    def maybe_refresh_pool(self, k: int) -> None:
+        """Single-worker wrapper: prefetch at half consumption, install
+        on wrap (with a device drain), account k replays.  Callers must
+        invoke this while no replay is being enqueued concurrently (all
+        run paths do)."""
+        if not self._has_pool():
+            return
+        self.pool_prefetch(k)
+        if self.pool_would_wrap(k):
             if self.device.type == "cuda":
                 torch.cuda.synchronize(self.device)
-            self._install_pool(picks)
+            self.pool_install_now()
             if self.device.type == "cuda":
                 torch.cuda.synchronize(self.device)
-            used = 0
-        self._pool_used = used + k
+        self.pool_account(k)
 
     def replay_once(self) -> None:
         """Enqueue ONE training step on the current HIP stream (graph must
