@@ -278,6 +278,25 @@ class Worker:
             g["done"].zero_()  # ticket is 0 at quiescence; re-assert
         self._pool_used = 0
         self._pool_next = None
+        # kick the NEXT draw right away: it then has the whole pool
+        # lifetime (~CAP steps of GPU work) to finish, so installs are
+        # always a join-of-a-finished-thread + one H2D copy
+        self._start_pool_draw()
+
+    def _start_pool_draw(self) -> None:
+        if getattr(self, "_pool_thread", None) is not None \
+                or getattr(self, "_pool_next", None) is not None:
+            return
+        import threading
+
+        n, mbs = self.x.shape[0], self._graph_state["idx_pool"].shape[1]
+
+        def _draw():
+            self._pool_next = self._draw_pool_picks(n, mbs)
+
+        t = threading.Thread(target=_draw, daemon=True)
+        t.start()
+        self._pool_thread = t
 
     def _fill_idx_pool(self, n: int, mbs: int) -> None:
         """(Re)draw the device-side pool of minibatch index rows and
@@ -291,24 +310,12 @@ class Worker:
         return bool(g) and g.get("idx_pool") is not None
 
     def pool_prefetch(self, k: int) -> None:
-        """Kick the background draw of the next pool once this one is
-        half consumed (the ~50-100 ms host RNG overlaps GPU compute)."""
-        if not self._has_pool():
-            return
-        used = getattr(self, "_pool_used", 0)
-        if used + k > self._IDX_POOL_CAP // 2 \
-                and getattr(self, "_pool_thread", None) is None \
-                and getattr(self, "_pool_next", None) is None:
-            import threading
-
-            n, mbs = self.x.shape[0], self._graph_state["idx_pool"].shape[1]
-
-            def _draw():
-                self._pool_next = self._draw_pool_picks(n, mbs)
-
-            t = threading.Thread(target=_draw, daemon=True)
-            t.start()
-            self._pool_thread = t
+        """Ensure a background draw of the next pool is running (it
+        normally already is — _install_pool kicks one immediately, so
+        the ~50-200 ms host RNG has a whole pool lifetime of GPU work to
+        hide behind)."""
+        if self._has_pool():
+            self._start_pool_draw()
 
     def pool_would_wrap(self, k: int) -> bool:
         return self._has_pool() and \
