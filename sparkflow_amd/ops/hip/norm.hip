@@ -594,8 +594,13 @@ std::vector<torch::Tensor> bn2d_nhwc_fwd(torch::Tensor x,
   check_f32n(x, "x");
   const int C = x.size(-1);
   const long rows = x.numel() / C;
-  const bool v4 = (C & 3) == 0;
-  TORCH_CHECK(res.numel() == 0 || (v4 && res.numel() == x.numel()),
+  static const bool v4_enabled = [] {
+    const char* e = getenv("SPARKFLOW_BN_V4");
+    return !(e && e[0] == '0');
+  }();
+  const bool v4 = (C & 3) == 0 && (v4_enabled || res.numel() > 0);
+  TORCH_CHECK(res.numel() == 0 || ((C & 3) == 0
+              && res.numel() == x.numel()),
               "fused residual needs C%4==0 and matching shape");
   auto y = torch::empty_like(x);
   auto stats = torch::zeros({2 * C}, x.options());
@@ -640,8 +645,12 @@ std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
   check_f32n(dy, "dy");
   const int C = x.size(-1);
   const long rows = x.numel() / C;
-  const bool v4 = (C & 3) == 0;
-  TORCH_CHECK(!want_dres || v4, "dres needs C%4==0");
+  static const bool v4_enabled = [] {
+    const char* e = getenv("SPARKFLOW_BN_V4");
+    return !(e && e[0] == '0');
+  }();
+  const bool v4 = (C & 3) == 0 && (v4_enabled || want_dres);
+  TORCH_CHECK(!want_dres || (C & 3) == 0, "dres needs C%4==0");
   auto red = torch::zeros({2 * C}, x.options());
   auto dx = torch::empty_like(x);
   auto dres = want_dres ? torch::empty_like(x) : torch::empty({0}, x.options());
