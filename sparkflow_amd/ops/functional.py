@@ -461,8 +461,7 @@ def batchnorm2d_act_nhwc(x, gamma, beta, run_mean, run_var, momentum=0.1,
                          residual=None):
     """x: [N, H, W, C] channel-last.  ``residual`` fuses a ResNet block
     boundary: out = relu(bn(x) + residual)."""
-    if training and ops.use_hip(x) and (residual is None
-                                        or x.shape[-1] % 4 == 0):
+    if training and ops.use_hip(x):
         return BatchNorm2dActNHWC.apply(x.contiguous(), gamma, beta,
                                         run_mean, run_var, momentum, eps,
                                         relu, residual)

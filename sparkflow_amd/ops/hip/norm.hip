@@ -478,6 +478,7 @@ __global__ void bn_nhwc_norm_kernel(const float* __restrict__ x,
                                     float* __restrict__ save_invstd,
                                     float* __restrict__ run_mean,
                                     float* __restrict__ run_var,
+                                    const float* __restrict__ res,
                                     float momentum, float eps, long rows,
                                     int C, int relu) {
   const int cl = threadIdx.x & (BNC - 1);
@@ -503,7 +504,8 @@ __global__ void bn_nhwc_norm_kernel(const float* __restrict__ x,
   }
   for (long r = (long)blockIdx.y * 4 + rg; r < rows;
        r += (long)gridDim.y * 4) {
-    const float v = x[r * C + c] * scale + shift;
+    float v = x[r * C + c] * scale + shift;
+    if (res) v += res[r * C + c];
     y[r * C + c] = relu ? fmaxf(v, 0.0f) : v;
   }
 }
@@ -550,6 +552,7 @@ __global__ void bn_nhwc_bwd_dx_kernel(const float* __restrict__ x,
                                       const float* __restrict__ si,
                                       const float* __restrict__ gamma,
                                       const float* __restrict__ red,
+                                      float* __restrict__ dres,
                                       long rows, int C, int relu) {
   const int cl = threadIdx.x & (BNC - 1);
   const int rg = threadIdx.x >> 6;
@@ -565,6 +568,7 @@ __global__ void bn_nhwc_bwd_dx_kernel(const float* __restrict__ x,
        r += (long)gridDim.y * 4) {
     float d = dy[r * C + c];
     if (relu && y[r * C + c] <= 0.0f) d = 0.0f;
+    if (dres) dres[r * C + c] = d;
     const float xhat = (x[r * C + c] - mean) * invstd;
     dx[r * C + c] = scale * (d - mean_dy - xhat * mean_dyx);
   }
@@ -594,14 +598,19 @@ std::vector<torch::Tensor> bn2d_nhwc_fwd(torch::Tensor x,
   check_f32n(x, "x");
   const int C = x.size(-1);
   const long rows = x.numel() / C;
+  // MEASURED (round 2, benchmarks/bn_micro.py): the scalar per-lane
+  // kernels stream at 3.0-3.5 TB/s effective on the ResNet shapes, the
+  // float4 v4 variants only ~2.0 — the scalar layout (one wave = 64
+  // consecutive channels of one row) wins; v4 kept behind
+  // SPARKFLOW_BN_V4=1 for re-evaluation.  The scalar kernels carry the
+  // fused residual too.
   static const bool v4_enabled = [] {
     const char* e = getenv("SPARKFLOW_BN_V4");
-    return !(e && e[0] == '0');
+    return e && e[0] == '1';
   }();
-  const bool v4 = (C & 3) == 0 && (v4_enabled || res.numel() > 0);
-  TORCH_CHECK(res.numel() == 0 || ((C & 3) == 0
-              && res.numel() == x.numel()),
-              "fused residual needs C%4==0 and matching shape");
+  const bool v4 = (C & 3) == 0 && v4_enabled;
+  TORCH_CHECK(res.numel() == 0 || res.numel() == x.numel(),
+              "fused residual needs matching shape");
   auto y = torch::empty_like(x);
   auto stats = torch::zeros({2 * C}, x.options());
   auto save_mean = torch::empty({C}, x.options());
@@ -631,6 +640,7 @@ std::vector<torch::Tensor> bn2d_nhwc_fwd(torch::Tensor x,
       bn_nhwc_norm_kernel, grid, dim3(256), 0, stream, x.data_ptr<float>(),
       y.data_ptr<float>(), stats.data_ptr<float>(), gp, bp,
       save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(), rmp, rvp,
+      res.numel() ? res.data_ptr<float>() : nullptr,
       (float)momentum, (float)eps, rows, C, (int)relu);
   return {y, save_mean, save_invstd};
 }
@@ -647,10 +657,9 @@ std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
   const long rows = x.numel() / C;
   static const bool v4_enabled = [] {
     const char* e = getenv("SPARKFLOW_BN_V4");
-    return !(e && e[0] == '0');
+    return e && e[0] == '1';
   }();
-  const bool v4 = (C & 3) == 0 && (v4_enabled || want_dres);
-  TORCH_CHECK(!want_dres || (C & 3) == 0, "dres needs C%4==0");
+  const bool v4 = (C & 3) == 0 && v4_enabled;
   auto red = torch::zeros({2 * C}, x.options());
   auto dx = torch::empty_like(x);
   auto dres = want_dres ? torch::empty_like(x) : torch::empty({0}, x.options());
@@ -682,6 +691,7 @@ std::vector<torch::Tensor> bn2d_nhwc_bwd(torch::Tensor x, torch::Tensor dy,
       bn_nhwc_bwd_dx_kernel, grid, dim3(256), 0, stream,
       x.data_ptr<float>(), dy.data_ptr<float>(), yp, dx.data_ptr<float>(),
       save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(), gp,
-      red.data_ptr<float>(), rows, C, (int)relu);
+      red.data_ptr<float>(),
+      want_dres ? dres.data_ptr<float>() : nullptr, rows, C, (int)relu);
   return {dx, red.narrow(0, C, C), red.narrow(0, 0, C), dres};
 }
