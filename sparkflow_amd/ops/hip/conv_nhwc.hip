@@ -815,15 +815,31 @@ torch::Tensor conv_nhwc_dilate(torch::Tensor dy, int64_t s, int64_t extra_h,
 }
 
 // db[co] = sum over (n, ho, wo) of dy[..., co] — rows are co-contiguous.
+// Same access shape as the BN stats kernel (64 channel lanes x 4 row
+// groups per 256-thread block, LDS combine, one atomic per channel per
+// block): the round-1 version strided one thread per channel down the
+// rows and measured 26% of the MNIST-CNN step at 0.43 TB/s.
+#define CBG 64
+
 __global__ void conv_nhwc_bias_grad_kernel(const float* __restrict__ dy,
                                            float* __restrict__ db, long rows,
                                            int C) {
-  const long r0 = blockIdx.x;
+  const int cl = threadIdx.x & (CBG - 1);
+  const int rg = threadIdx.x >> 6;
+  const int c = blockIdx.x * CBG + cl;
   float s = 0.0f;
-  const int c = threadIdx.x;
-  if (c >= C) return;
-  for (long r = r0; r < rows; r += gridDim.x) s += dy[r * C + c];
-  atomicAdd(&db[c], s);
+  if (c < C) {
+    for (long r = (long)blockIdx.y * 4 + rg; r < rows;
+         r += (long)gridDim.y * 4) {
+      s += dy[r * C + c];
+    }
+  }
+  __shared__ float ls[4][CBG];
+  ls[rg][cl] = s;
+  __syncthreads();
+  if (rg == 0 && c < C) {
+    atomicAdd(&db[c], ls[0][cl] + ls[1][cl] + ls[2][cl] + ls[3][cl]);
+  }
 }
 
 torch::Tensor conv_nhwc_bias_grad(torch::Tensor dy) {
@@ -831,12 +847,12 @@ torch::Tensor conv_nhwc_bias_grad(torch::Tensor dy) {
   const int C = dy.size(3);
   const long rows = dy.numel() / C;
   auto db = torch::zeros({C}, dy.options());
-  TORCH_CHECK(C <= 1024, "bias grad: C too large for one block");
-  const int blocks = (int)std::min<long>(512, rows);
-  hipLaunchKernelGGL(conv_nhwc_bias_grad_kernel, dim3(blocks),
-                     dim3(std::min(1024, ((C + 63) / 64) * 64)), 0,
-                     at::cuda::getCurrentCUDAStream(), dy.data_ptr<float>(),
-                     db.data_ptr<float>(), rows, C);
+  const int cgrids = (C + CBG - 1) / CBG;
+  long chunks = (rows + 3) / 4;
+  chunks = std::min(chunks, (long)std::max(1, 2048 / cgrids));
+  hipLaunchKernelGGL(conv_nhwc_bias_grad_kernel, dim3(cgrids, (int)chunks),
+                     dim3(256), 0, at::cuda::getCurrentCUDAStream(),
+                     dy.data_ptr<float>(), db.data_ptr<float>(), rows, C);
   return db;
 }
 
