@@ -300,3 +300,60 @@ def test_dist_fit_traverses_rccl_star(monkeypatch, tmp_path):
         assert captured[0]["update_count"] >= 2 * 8  # both ranks pushed
     finally:
         SparkContext._active_spark_context = prev_ctx
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(420)
+def test_dist_fit_gpu_compute():
+    """SparkAsyncDL.fit with process executors on a GPU box: both ranks
+    compute on the GPU (world=2 > n_gpus=1 here, so the wire backend is
+    gloo per the closure's rule; on an 8-GPU node the same code takes
+    nccl/RCCL pair links)."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    import os
+    import subprocess
+    import sys
+
+    from tests.conftest import REPO_ROOT
+
+    code = r"""
+import os, sys
+os.environ["SPARKFLOW_FIT_TRANSPORT"] = "dist"
+sys.path.insert(0, %r)
+import numpy as np
+from sparkflow_amd import localspark
+localspark.install(quiet=True)
+from sparkflow_amd.localspark.context import SparkConf, SparkContext
+from sparkflow_amd.localspark.sql import SparkSession as LS
+from pyspark.ml.linalg import Vectors
+from sparkflow_amd.graph_utils import build_graph
+from sparkflow_amd.models import gaussian_mlp
+from sparkflow_amd.spark import SparkAsyncDL
+from sparkflow_amd.parallel.transport import find_free_port
+
+s = LS(SparkContext("localproc[2]", "gpufit", SparkConf()))
+rng = np.random.default_rng(3)
+rows = [(i %% 2, Vectors.dense(rng.normal(-2.0 if i %% 2 == 0 else 2.0,
+                                          0.7, 10))) for i in range(200)]
+df = s.createDataFrame(rows, ["label", "features"])
+est = SparkAsyncDL(inputCol="features",
+                   tensorflowGraph=build_graph(gaussian_mlp),
+                   tfInput="x:0", tfLabel="y:0", tfOutput="out:0",
+                   tfLearningRate=.01, iters=8, predictionCol="predicted",
+                   labelCol="label", partitions=2, miniBatchSize=-1,
+                   port=find_free_port())
+model = est.fit(df)
+preds = model.transform(df).collect()
+errs = sum(1 for r in preds
+           if round(min(max(float(r["predicted"]), 0.0), 1.0)) != r["label"])
+assert errs < len(preds), errs
+print("GPU_DIST_FIT_OK", errs)
+"""
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", GLOO_SOCKET_IFNAME="lo")
+    r = subprocess.run([sys.executable, "-c", code % str(REPO_ROOT)],
+                       env=env, capture_output=True, text=True, timeout=390)
+    assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-3000:]
+    assert "GPU_DIST_FIT_OK" in r.stdout
