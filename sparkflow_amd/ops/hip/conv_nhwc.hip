@@ -819,26 +819,29 @@ torch::Tensor conv_nhwc_dilate(torch::Tensor dy, int64_t s, int64_t extra_h,
 // groups per 256-thread block, LDS combine, one atomic per channel per
 // block): the round-1 version strided one thread per channel down the
 // rows and measured 26% of the MNIST-CNN step at 0.43 TB/s.
-#define CBG 64
-
 __global__ void conv_nhwc_bias_grad_kernel(const float* __restrict__ dy,
                                            float* __restrict__ db, long rows,
-                                           int C) {
-  const int cl = threadIdx.x & (CBG - 1);
-  const int rg = threadIdx.x >> 6;
-  const int c = blockIdx.x * CBG + cl;
+                                           int C, int lg_lanes) {
+  // lane width adapts to C (e.g. C=32 -> 32 lanes x 8 row groups, so a
+  // wave reads 2 FULL consecutive rows — no idle lanes, contiguous)
+  const int lanes = 1 << lg_lanes;
+  const int rgs = 256 >> lg_lanes;
+  const int cl = threadIdx.x & (lanes - 1);
+  const int rg = threadIdx.x >> lg_lanes;
+  const int c = blockIdx.x * lanes + cl;
   float s = 0.0f;
   if (c < C) {
-    for (long r = (long)blockIdx.y * 4 + rg; r < rows;
-         r += (long)gridDim.y * 4) {
+    for (long r = (long)blockIdx.y * rgs + rg; r < rows;
+         r += (long)gridDim.y * rgs) {
       s += dy[r * C + c];
     }
   }
-  __shared__ float ls[4][CBG];
-  ls[rg][cl] = s;
+  __shared__ float ls[256];
+  ls[threadIdx.x] = s;
   __syncthreads();
   if (rg == 0 && c < C) {
-    atomicAdd(&db[c], ls[0][cl] + ls[1][cl] + ls[2][cl] + ls[3][cl]);
+    for (int k = 1; k < rgs; ++k) s += ls[k * lanes + cl];
+    atomicAdd(&db[c], s);
   }
 }
 
@@ -847,12 +850,17 @@ torch::Tensor conv_nhwc_bias_grad(torch::Tensor dy) {
   const int C = dy.size(3);
   const long rows = dy.numel() / C;
   auto db = torch::zeros({C}, dy.options());
-  const int cgrids = (C + CBG - 1) / CBG;
-  long chunks = (rows + 3) / 4;
+  int lg = 4;  // >= 16 lanes
+  while ((1 << (lg + 1)) <= C && lg < 6) ++lg;
+  const int lanes = 1 << lg;
+  const int cgrids = (C + lanes - 1) / lanes;
+  const int rgs = 256 >> lg;
+  long chunks = (rows + rgs - 1) / rgs;
   chunks = std::min(chunks, (long)std::max(1, 2048 / cgrids));
   hipLaunchKernelGGL(conv_nhwc_bias_grad_kernel, dim3(cgrids, (int)chunks),
                      dim3(256), 0, at::cuda::getCurrentCUDAStream(),
-                     dy.data_ptr<float>(), db.data_ptr<float>(), rows, C);
+                     dy.data_ptr<float>(), db.data_ptr<float>(), rows, C,
+                     lg);
   return db;
 }
 
