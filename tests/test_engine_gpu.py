@@ -311,3 +311,20 @@ def test_replay_streams_reroll_mid_training():
     assert torch.isfinite(eng.ps.weights).all()
     assert not torch.equal(eng.ps.weights, w_before)
     assert eng.ps.update_count == 4 * 20 * 4  # 4 calls x 20 steps x 4 workers
+
+
+def test_raw_rccl_comm_init_times_out_without_peer():
+    """ADVICE round-1 #4: ncclCommInitRank is a collective rendezvous —
+    with a dead peer the old blocking init hung forever and wedged the
+    pair handshake.  The non-blocking init must RAISE within the bound
+    instead, so init_native can fall back to torch pairs."""
+    import time
+
+    from sparkflow_amd import ops
+
+    uid = ops.hip.nccl_unique_id()
+    t0 = time.perf_counter()
+    with pytest.raises(RuntimeError, match="timed out|timeout|failed"):
+        ops.hip.nccl_comm_init(uid, 2, 0, 2000)  # nobody joins rank 1
+    dt = time.perf_counter() - t0
+    assert dt < 30, f"timeout not bounded: {dt:.1f}s"
