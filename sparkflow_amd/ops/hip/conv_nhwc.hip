@@ -856,7 +856,10 @@ torch::Tensor conv_nhwc_bias_grad(torch::Tensor dy) {
   const int cgrids = (C + lanes - 1) / lanes;
   const int rgs = 256 >> lg;
   long chunks = (rows + rgs - 1) / rgs;
-  chunks = std::min(chunks, (long)std::max(1, 2048 / cgrids));
+  // cap the per-address atomic chain (chunks atomics per channel):
+  // CNN-sized tensors are small enough that 2048 blocks made the
+  // same-address atomics the bound, not the streaming reads
+  chunks = std::min(chunks, (long)std::max(1, 512 / cgrids));
   hipLaunchKernelGGL(conv_nhwc_bias_grad_kernel, dim3(cgrids, (int)chunks),
                      dim3(256), 0, at::cuda::getCurrentCUDAStream(),
                      dy.data_ptr<float>(), db.data_ptr<float>(), rows, C,
