@@ -742,6 +742,13 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
          splitk * 2 <= max_by_k) {
     splitk *= 2;
   }
+  // tiny-dw wgrads (base = 1 tile, e.g. MNIST conv1 dw[32][25]) stall at
+  // splitk=256 = 256 workgroups = 1 per CU with poor latency hiding;
+  // keep doubling toward the 512-workgroup fill target
+  while (base * splitk < 512 && splitk < cap && splitk * 2 <= max_by_k &&
+         splitk < 1024) {
+    splitk *= 2;
+  }
   if (splitk > 1) {
     const int k_chunk = ceil_div(K, splitk * BK) * BK;
     auto ws = torch::empty({splitk, M, Nc}, x.options());
