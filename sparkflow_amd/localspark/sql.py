@@ -208,6 +208,10 @@ class SparkSession:
         self.sparkContext = sparkContext
         self._sc = sparkContext
 
+    @property
+    def read(self):
+        return DataFrameReader(self)
+
     def createDataFrame(self, data, schema=None) -> DataFrame:
         rows = list(data)
         if schema is not None and not isinstance(schema, (list, tuple)):
@@ -233,6 +237,45 @@ class SparkSession:
     @property
     def conf(self):
         return self.sparkContext.getConf()
+
+
+class DataFrameReader:
+    """Minimal spark.read: headerless CSV with optional type inference
+    (the reference examples' `spark.read.option("inferSchema", "true")
+    .csv(path)` — columns named _c0.._cN like Spark)."""
+
+    def __init__(self, session):
+        self._session = session
+        self._options = {}
+
+    def option(self, key, value):
+        self._options[str(key)] = value
+        return self
+
+    def csv(self, path) -> DataFrame:
+        import csv as _csv
+
+        infer = str(self._options.get("inferSchema", "false")).lower() \
+            == "true"
+        rows = []
+        with open(path, newline="") as f:
+            for rec in _csv.reader(f):
+                if infer:
+                    vals = []
+                    for v in rec:
+                        try:
+                            vals.append(int(v))
+                        except ValueError:
+                            try:
+                                vals.append(float(v))
+                            except ValueError:
+                                vals.append(v)
+                    rows.append(tuple(vals))
+                else:
+                    rows.append(tuple(rec))
+        ncol = len(rows[0]) if rows else 0
+        cols = ["_c%d" % i for i in range(ncol)]
+        return self._session.createDataFrame(rows, cols)
 
 
 class _BuilderAccessor:

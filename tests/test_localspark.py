@@ -162,3 +162,12 @@ def test_stopwords_carrier_roundtrip_bytes(tmp_path):
     back = load_stage(path)
     assert back.getStopWords()[-1] == PY_OBJ_GUID
     assert decode_py_obj(back.getStopWords()) == obj
+
+
+def test_spark_read_csv(spark, tmp_path):
+    p = tmp_path / "d.csv"
+    p.write_text("1,0.5,a\n2,1.5,b\n")
+    df = spark.read.option("inferSchema", "true").csv(str(p))
+    assert df.columns == ["_c0", "_c1", "_c2"]
+    r = df.collect()
+    assert r[0]["_c0"] == 1 and r[0]["_c1"] == 0.5 and r[1]["_c2"] == "b"
