@@ -310,6 +310,151 @@ __global__ __launch_bounds__(256) void conv_nhwc_fwd_big_kernel(
   }
 }
 
+// ---------------- forward, 128x128 tile (Cout >= 128) ----------------
+// The guide's canonical f32 recipe (cdna_hip_programming.md §3:
+// 128x128xBK32 block, 2x2 32x32 frags per wave on mfma_f32_32x32x2) —
+// applicable where Cout >= 128 so the BN=128 tile wastes nothing
+// (ResNet layers 2-4).  Lighter staging than the 256x64 attempt
+// (4+4 float4 per thread instead of 8+2), ~110 VGPR -> 4 waves/SIMD.
+#define BM3 128
+#define BN3 128
+
+template <bool HAS_BIAS>
+__global__ __launch_bounds__(256) void conv_nhwc_fwd_sq_kernel(
+    const float* __restrict__ x, const float* __restrict__ wgt,
+    const float* __restrict__ bias, float* __restrict__ y, ConvShapeHW s,
+    int M, int K, int act) {
+  __shared__ float As[BM3][BK2 + 1];
+  __shared__ float Bs[BK2][BN3 + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = (wv >> 1) * 64;
+  const int wn = (wv & 1) * 64;
+  const int m0 = blockIdx.y * BM3;
+  const int n0 = blockIdx.x * BN3;
+  const int Nc = s.Cout;
+
+  f32x16 acc[2][2] = {};
+
+  // A: thread owns pixel row m0 + (t>>1) and k-slots (t&1)*16 + 4i
+  const int a_row = t >> 1, a_k16 = (t & 1) * 16;
+  const int gm = m0 + a_row;
+  const int m_dec = gm < M ? gm : 0;
+  const int a_wo = m_dec % s.Wo;
+  const int a_ho = (m_dec / s.Wo) % s.Ho;
+  const long a_nbase = (long)(m_dec / (s.Wo * s.Ho)) * s.H * s.W * s.Cin;
+  const int h0 = a_ho * s.stride - s.pad;
+  const int w0 = a_wo * s.stride - s.pad;
+  int a_ci = a_k16 % s.Cin;          // cursor at k-group (t&1)*16
+  int a_khkw = a_k16 / s.Cin;
+
+  auto load_a = [&](int k0, float4 (&r)[4]) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g) r[g] = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gm >= M) return;
+    int ci = a_ci, khkw = a_khkw;
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int kh = khkw / s.Kw, kw = khkw - (khkw / s.Kw) * s.Kw;
+      const int h = h0 + kh, w = w0 + kw;
+      if (k0 + a_k16 + g * 4 + 3 < K && h >= 0 && h < s.H && w >= 0 &&
+          w < s.W) {
+        r[g] = *reinterpret_cast<const float4*>(
+            &x[a_nbase + ((long)h * s.W + w) * s.Cin + ci]);
+      }
+      ci += 4;
+      if (ci >= s.Cin) { ci -= s.Cin; ++khkw; }
+    }
+  };
+  auto advance_a = [&]() {
+    a_ci += BK2;
+    while (a_ci >= s.Cin) { a_ci -= s.Cin; ++a_khkw; }
+  };
+  // B: thread owns co = n0 + (t & 127), k-offsets (t>>7)*16 + 4i
+  const int b_co = n0 + (t & 127);
+  const int b_k16 = (t >> 7) * 16;
+  auto load_b = [&](int k0, float4 (&r)[4]) {
+    if (b_co < Nc && k0 + b_k16 + 15 < K) {
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        r[g] = *reinterpret_cast<const float4*>(
+            &wgt[(long)b_co * K + k0 + b_k16 + g * 4]);
+      }
+    } else {
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        float* rr = reinterpret_cast<float*>(&r[g]);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int gk = k0 + b_k16 + g * 4 + i;
+          rr[i] = (b_co < Nc && gk < K) ? wgt[(long)b_co * K + gk] : 0.0f;
+        }
+      }
+    }
+  };
+
+  const int fi = lane & 31;
+  const int fk = lane >> 5;
+
+  float4 ra[4], rb[4];
+  load_a(0, ra);
+  load_b(0, rb);
+  for (int k0 = 0; k0 < K; k0 += BK2) {
+    {
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        const float* p = reinterpret_cast<const float*>(&ra[g]);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) As[a_row][a_k16 + g * 4 + i] = p[i];
+      }
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        const float* p = reinterpret_cast<const float*>(&rb[g]);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) Bs[b_k16 + g * 4 + i][t & 127] = p[i];
+      }
+    }
+    __syncthreads();
+    if (k0 + BK2 < K) {
+      advance_a();
+      load_a(k0 + BK2, ra);
+      load_b(k0 + BK2, rb);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK2 / 2; ++kk) {
+      const float a0 = As[wm + fi][kk * 2 + fk];
+      const float a1 = As[wm + 32 + fi][kk * 2 + fk];
+      const float b0 = Bs[kk * 2 + fk][wn + fi];
+      const float b1 = Bs[kk * 2 + fk][wn + 32 + fi];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int ni = 0; ni < 2; ++ni) {
+    const int co = n0 + wn + ni * 32 + fi;
+    if (co >= Nc) continue;
+    const float bv = HAS_BIAS ? bias[co] : 0.0f;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int row = (reg & 3) + 8 * (reg >> 2) + 4 * fk;
+        const int m = m0 + wm + mi * 32 + row;
+        if (m < M) {
+          y[(long)m * Nc + co] = act_apply(acc[mi][ni][reg] + bv, act);
+        }
+      }
+    }
+  }
+}
+
 // ---------------- wgrad ----------------
 // dw[co][k'] = sum_p dy[p, co] * im2col(x)[p, k'] — A staged transposed
 // (dy contiguous along co), B's im2col contiguous along ci.
@@ -691,6 +836,27 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w,
     const char* e = getenv("SPARKFLOW_CONV_BIG");
     return e && e[0] == '1';
   }();
+  // 128x128/BK32 mfma_32x32x2 tile for Cout >= 128 shapes (the guide's
+  // canonical f32 recipe; SPARKFLOW_CONV_SQ=0 disables)
+  static const bool sq_enabled = [] {
+    const char* e = getenv("SPARKFLOW_CONV_SQ");
+    return !(e && e[0] == '0');
+  }();
+  const long sq_tiles = (long)ceil_div(M, BM3) * ceil_div(s.Cout, BN3);
+  if (sq_enabled && (s.Cin & 3) == 0 && s.Cout >= BN3 && K >= BK2 &&
+      sq_tiles >= 128) {
+    dim3 grid(ceil_div(s.Cout, BN3), ceil_div(M, BM3));
+    if (bp) {
+      hipLaunchKernelGGL((conv_nhwc_fwd_sq_kernel<true>), grid, dim3(256),
+                         0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                         bp, y.data_ptr<float>(), s, M, K, (int)act);
+    } else {
+      hipLaunchKernelGGL((conv_nhwc_fwd_sq_kernel<false>), grid, dim3(256),
+                         0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                         nullptr, y.data_ptr<float>(), s, M, K, (int)act);
+    }
+    return y;
+  }
   const long big_tiles = (long)ceil_div(M, BM2) * ceil_div(s.Cout, BN2);
   if (big_enabled && (s.Cin & 3) == 0 && K >= BK2 && big_tiles >= 128) {
     dim3 grid(ceil_div(s.Cout, BN2), ceil_div(M, BM2));
