@@ -455,6 +455,191 @@ __global__ __launch_bounds__(256) void conv_nhwc_fwd_sq_kernel(
   }
 }
 
+// ---------------- wgrad, 128x128 tile (Cout >= 128) ----------------
+// Same mfma_f32_32x32x2 recipe as the sq forward, for the wgrad GEMM
+// dw[co][k'] = sum_p dy[p, co] * im2col(x)[p, k'] on the Cout >= 128
+// ResNet layers (the 64x64 kernel stays for smaller dw).  BK2=32 deep;
+// A (dy, transposed at staging) and B (im2col) both float4.
+template <bool SPLITK>
+__global__ __launch_bounds__(256) void conv_nhwc_wgrad_sq_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    float* __restrict__ dw, ConvShapeHW s, int M, int Nc, int K,
+    int k_chunk) {
+  __shared__ float As[BM3][BK2 + 1];
+  __shared__ float Bs[BK2][BN3 + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = (wv >> 1) * 64;
+  const int wn = (wv & 1) * 64;
+  const int m0 = blockIdx.y * BM3;  // co tile
+  const int n0 = blockIdx.x * BN3;  // k' tile
+
+  f32x16 acc[2][2] = {};
+
+  int k_begin = 0, k_end = K;
+  if (SPLITK) {
+    k_begin = blockIdx.z * k_chunk;
+    k_end = min(K, k_begin + k_chunk);
+  }
+
+  // A: As[co][p] = dy[p][co] — thread owns pixel-row (t>>3) and co-slots
+  // (t&7)*16 + 4g (4 float4 per step; dy rows are co-contiguous)
+  const int a_kk = t >> 3, a_m16 = (t & 7) * 16;
+  // B: Bs[p][k'] — thread owns pixel-row (t>>3) and k'-slots
+  // (t&7)*16 + 4g
+  const int b_kk = t >> 3, b_n16 = (t & 7) * 16;
+  const int HoWo = s.Ho * s.Wo;
+
+  int pa_n = (k_begin + a_kk) / HoWo;
+  int pa_hw = (k_begin + a_kk) % HoWo;
+  int pb_n = (k_begin + b_kk) / HoWo;
+  int pb_ho = ((k_begin + b_kk) % HoWo) / s.Wo;
+  int pb_wo = ((k_begin + b_kk) % HoWo) % s.Wo;
+  // decode the 16 k' columns once (fixed for the whole K loop)
+  int kw_i[16], kh_i[16], ci_i[16];
+  bool gk_ok[16], gvec[4];
+  const bool cin4 = (s.Cin & 3) == 0;
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    const int gk = n0 + b_n16 + i;
+    gk_ok[i] = gk < Nc;
+    const int g = gk_ok[i] ? gk : 0;
+    ci_i[i] = g % s.Cin;
+    const int khkw = g / s.Cin;
+    kw_i[i] = khkw % s.Kw;
+    kh_i[i] = khkw / s.Kw;
+  }
+#pragma unroll
+  for (int g = 0; g < 4; ++g) {
+    gvec[g] = cin4 && gk_ok[g * 4 + 3] && kh_i[g * 4] == kh_i[g * 4 + 3] &&
+              kw_i[g * 4] == kw_i[g * 4 + 3];
+  }
+  const long n_stride = (long)s.H * s.W * s.Cin;
+
+  auto load_a = [&](int gp0, float4 (&r)[4]) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g) r[g] = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gp0 >= k_end) return;
+    const long rowbase = ((long)pa_n * HoWo + pa_hw) * s.Cout;
+    if (m0 + a_m16 + 15 < s.Cout && (s.Cout & 3) == 0) {
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        r[g] = *reinterpret_cast<const float4*>(
+            &dy[rowbase + m0 + a_m16 + g * 4]);
+      }
+    } else {
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        float* rr = reinterpret_cast<float*>(&r[g]);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int co = m0 + a_m16 + g * 4 + i;
+          rr[i] = co < s.Cout ? dy[rowbase + co] : 0.0f;
+        }
+      }
+    }
+  };
+  auto load_b = [&](int gp0, float4 (&r)[4]) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g) r[g] = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gp0 >= k_end) return;
+    const long nbase = (long)pb_n * n_stride;
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      if (gvec[g]) {
+        const int h = pb_ho * s.stride - s.pad + kh_i[g * 4];
+        const int w = pb_wo * s.stride - s.pad + kw_i[g * 4];
+        if (h >= 0 && h < s.H && w >= 0 && w < s.W) {
+          r[g] = *reinterpret_cast<const float4*>(
+              &x[nbase + ((long)h * s.W + w) * s.Cin + ci_i[g * 4]]);
+        }
+      } else {
+        float* rr = reinterpret_cast<float*>(&r[g]);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int j = g * 4 + i;
+          if (gk_ok[j]) {
+            const int h = pb_ho * s.stride - s.pad + kh_i[j];
+            const int w = pb_wo * s.stride - s.pad + kw_i[j];
+            if (h >= 0 && h < s.H && w >= 0 && w < s.W)
+              rr[i] = x[nbase + ((long)h * s.W + w) * s.Cin + ci_i[j]];
+          }
+        }
+      }
+    }
+  };
+  auto advance = [&]() {  // both pixel cursors += BK2
+    pa_hw += BK2;
+    while (pa_hw >= HoWo) { pa_hw -= HoWo; ++pa_n; }
+    pb_wo += BK2;
+    while (pb_wo >= s.Wo) { pb_wo -= s.Wo; ++pb_ho; }
+    while (pb_ho >= s.Ho) { pb_ho -= s.Ho; ++pb_n; }
+  };
+
+  const int fi = lane & 31;
+  const int fk = lane >> 5;
+
+  float4 ra[4], rb[4];
+  load_a(k_begin + a_kk, ra);
+  load_b(k_begin + b_kk, rb);
+  for (int k0 = k_begin; k0 < k_end; k0 += BK2) {
+    {
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        const float* pa = reinterpret_cast<const float*>(&ra[g]);
+        const float* pb = reinterpret_cast<const float*>(&rb[g]);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          As[a_m16 + g * 4 + i][a_kk] = pa[i];
+          Bs[b_kk][b_n16 + g * 4 + i] = pb[i];
+        }
+      }
+    }
+    __syncthreads();
+    if (k0 + BK2 < k_end) {
+      advance();
+      load_a(k0 + BK2 + a_kk, ra);
+      load_b(k0 + BK2 + b_kk, rb);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK2 / 2; ++kk) {
+      const float a0 = As[wm + fi][kk * 2 + fk];
+      const float a1 = As[wm + 32 + fi][kk * 2 + fk];
+      const float b0 = Bs[kk * 2 + fk][wn + fi];
+      const float b1 = Bs[kk * 2 + fk][wn + 32 + fi];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int ni = 0; ni < 2; ++ni) {
+    const int kcol = n0 + wn + ni * 32 + fi;
+    if (kcol >= Nc) continue;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int row = (reg & 3) + 8 * (reg >> 2) + 4 * fk;
+        const int co = m0 + wm + mi * 32 + row;
+        if (co < M) {
+          if (SPLITK) {
+            dw[(long)blockIdx.z * M * Nc + (long)co * Nc + kcol] =
+                acc[mi][ni][reg];
+          } else {
+            dw[(long)co * Nc + kcol] = acc[mi][ni][reg];
+          }
+        }
+      }
+    }
+  }
+}
+
 // ---------------- wgrad ----------------
 // dw[co][k'] = sum_p dy[p, co] * im2col(x)[p, k'] — A staged transposed
 // (dy contiguous along co), B's im2col contiguous along ci.
@@ -900,6 +1085,48 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
   const int K = s.N * s.Ho * s.Wo;
   auto dw = torch::empty({s.Cout, (long)Kh, (long)Kw, s.Cin}, x.options());
   auto stream = at::cuda::getCurrentCUDAStream();
+  // 128x128 mfma_32x32x2 wgrad tile for Cout >= 128 (SPARKFLOW_CONV_SQ=0
+  // disables, same flag as the sq forward)
+  static const bool sq_enabled = [] {
+    const char* e = getenv("SPARKFLOW_CONV_SQ");
+    return !(e && e[0] == '0');
+  }();
+  // measured (benchmarks/wgrad_micro.py): the sq tile wins at
+  // Cout >= 256 (l3 +10%, l4 +2%) and loses slightly at 128 — dispatch
+  // at M >= 2*BM3
+  if (sq_enabled && M >= 2 * BM3 && (s.Cout & 3) == 0) {
+    const int base_sq = ceil_div(Nc, BN3) * ceil_div(M, BM3);
+    int splitk = 1;
+    const int max_by_k = std::max(1, K / 256);
+    const int cap = sf_splitk_env_max();
+    while (splitk < 256 && splitk < cap && base_sq * splitk <= 1024 &&
+           splitk * 2 <= max_by_k) {
+      splitk *= 2;
+    }
+    while (base_sq * splitk < 512 && splitk < cap &&
+           splitk * 2 <= max_by_k && splitk < 1024) {
+      splitk *= 2;
+    }
+    if (splitk > 1) {
+      const int k_chunk = ceil_div(K, splitk * BK2) * BK2;
+      auto ws = torch::empty({splitk, M, Nc}, x.options());
+      dim3 grid(ceil_div(Nc, BN3), ceil_div(M, BM3), splitk);
+      hipLaunchKernelGGL((conv_nhwc_wgrad_sq_kernel<true>), grid,
+                         dim3(256), 0, stream, x.data_ptr<float>(),
+                         dy.data_ptr<float>(), ws.data_ptr<float>(), s, M,
+                         Nc, K, k_chunk);
+      const long n = (long)M * Nc;
+      launch_conv_splitk_reduce(ws.data_ptr<float>(), dw.data_ptr<float>(),
+                                n, splitk, stream);
+    } else {
+      dim3 grid(ceil_div(Nc, BN3), ceil_div(M, BM3));
+      hipLaunchKernelGGL((conv_nhwc_wgrad_sq_kernel<false>), grid,
+                         dim3(256), 0, stream, x.data_ptr<float>(),
+                         dy.data_ptr<float>(), dw.data_ptr<float>(), s, M,
+                         Nc, K, K);
+    }
+    return dw;
+  }
   const int base = ceil_div(Nc, BN) * ceil_div(M, BM);
   int splitk = 1;
   const int max_by_k = std::max(1, K / 256);
