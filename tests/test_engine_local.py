@@ -224,3 +224,36 @@ def test_idx_pool_refreshes_on_wrap():
     row = w._graph_state["idx_pool"][0]
     assert len(set(row.tolist())) == 8 and int(row.max()) < 64
     eng.close()
+
+
+def test_idx_pool_background_draw():
+    """Installing a pool kicks the NEXT draw on a background thread, so
+    a wrap joins a finished thread instead of paying the host RNG in
+    the step loop."""
+    import torch
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import mnist_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+
+    cfg = TrainConfig(iters=1, mini_batch=8, mini_stochastic_iters=1,
+                      seed=9)
+    eng = build_engine(build_graph(mnist_mlp),
+                       OptimizerSpec("sgd", 0.01), cfg, device="cpu")
+    w = eng.worker
+    rng = np.random.default_rng(1)
+    w.stage(rng.random((64, 784), dtype=np.float32),
+            rng.integers(0, 10, (64, 1)).astype(np.float32))
+    cap = w._IDX_POOL_CAP
+    w._graph_state = {"idx_pool": torch.zeros(cap, 8, dtype=torch.int64),
+                      "ctr": torch.zeros(1, dtype=torch.int64)}
+    w._fill_idx_pool(64, 8)
+    t = getattr(w, "_pool_thread", None)
+    assert t is not None  # draw already running
+    t.join(timeout=60)
+    assert w._pool_next is not None
+    p0 = w._graph_state["idx_pool"].clone()
+    w.maybe_refresh_pool(cap + 1)  # wrap -> installs the prefetched pool
+    assert not torch.equal(w._graph_state["idx_pool"], p0)
+    eng.close()

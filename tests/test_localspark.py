@@ -171,3 +171,23 @@ def test_spark_read_csv(spark, tmp_path):
     assert df.columns == ["_c0", "_c1", "_c2"]
     r = df.collect()
     assert r[0]["_c0"] == 1 and r[0]["_c1"] == 0.5 and r[1]["_c2"] == "b"
+
+
+def test_nested_pipeline_unwrap(spark, tmp_path):
+    """unwrap recurses into sub-pipelines (reference
+    pipeline_util.py:62-64)."""
+    from sparkflow_amd.spark.pipeline_util import (
+        PysparkPipelineWrapper,
+        encode_py_obj,
+    )
+
+    inner_payload = {"kind": "python-stage"}
+    sw = StopWordsRemover(inputCol="a", outputCol="b")
+    sw.setStopWords(encode_py_obj(inner_payload))
+    inner = PipelineModel([sw])
+    outer = PipelineModel([inner])
+    path = str(tmp_path / "nested")
+    outer.write().overwrite().save(path)
+    loaded = PysparkPipelineWrapper.unwrap(PipelineModel.load(path))
+    assert isinstance(loaded.stages[0], PipelineModel)
+    assert loaded.stages[0].stages[0] == inner_payload
