@@ -249,6 +249,14 @@ def build_engine(graph_json: str, optimizer: OptimizerSpec,
         pairs = DistPairs(workers_per_gpu=rwpg)
         pairs.init_native(device)  # raw RCCL channel (GIL-free sessions)
         _warmup_pairs(pairs, device)
+        if rank == 0:
+            total = (world - 1) * rwpg
+            print(f"[sparkflow_amd] engine world={world} device={device} "
+                  f"local_workers={wpg} remote_workers/rank={rwpg} "
+                  f"raw_rccl_pairs={len(pairs.native)}/{total}"
+                  + ("" if len(pairs.native) == total else
+                     " (torch-pair fallback on the rest)"),
+                  flush=True)
     if rank == 0:
         ps = PSCore(graph_json, optimizer, device=device,
                     acquire_lock=config.acquire_lock,
