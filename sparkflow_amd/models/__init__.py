@@ -1,4 +1,9 @@
-from sparkflow_amd.models.mlp import ClassifierMLP, Autoencoder, RegressorMLP  # noqa: F401
+from sparkflow_amd.models.mlp import (  # noqa: F401
+    Autoencoder,
+    ClassifierMLP,
+    DropoutRegressorMLP,
+    RegressorMLP,
+)
 from sparkflow_amd.models.cnn import MNISTCNN, ResNet18Shape  # noqa: F401
 from sparkflow_amd.models.zoo import (  # noqa: F401
     mnist_mlp,
@@ -9,4 +14,5 @@ from sparkflow_amd.models.zoo import (  # noqa: F401
     resnet18_shape,
     xor_mlp,
     gaussian_mlp,
+    dropout_mlp,
 )

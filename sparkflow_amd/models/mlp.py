@@ -60,6 +60,33 @@ class RegressorMLP(nn.Module):
         return self.loss_fn(out, target.reshape(out.shape).to(out.dtype))
 
 
+class DropoutRegressorMLP(nn.Module):
+    """RegressorMLP with an inter-layer Dropout — exercises the
+    reference's tf_dropout / to_keep_dropout surface (ml_util.py:70-71):
+    dropout is active in training and inert at prediction for both
+    flag values."""
+
+    def __init__(self, dims: List[int], p: float = 0.3):
+        super().__init__()
+        layers = []
+        for i in range(len(dims) - 1):
+            act = "relu" if i < len(dims) - 2 else "sigmoid"
+            layers.append(snn.Linear(dims[i], dims[i + 1], activation=act))
+        self.layers = nn.ModuleList(layers)
+        self.drop = snn.Dropout(p)
+        self.loss_fn = snn.MSE()
+
+    def forward(self, x):
+        for i, layer in enumerate(self.layers):
+            x = layer(x)
+            if i < len(self.layers) - 1:
+                x = self.drop(x)
+        return x
+
+    def compute_loss(self, out, target):
+        return self.loss_fn(out, target.reshape(out.shape).to(out.dtype))
+
+
 class Autoencoder(nn.Module):
     """Unsupervised dense AE; loss is MSE against the input batch
     (reference examples/autoencoder_example.py:9-15 — ``tfLabel=None``)."""

@@ -6,7 +6,12 @@ BASELINE.json's five benchmark configs.
 from __future__ import annotations
 
 from sparkflow_amd.models.cnn import MNISTCNN, ResNet18Shape
-from sparkflow_amd.models.mlp import Autoencoder, ClassifierMLP, RegressorMLP
+from sparkflow_amd.models.mlp import (
+    Autoencoder,
+    ClassifierMLP,
+    DropoutRegressorMLP,
+    RegressorMLP,
+)
 
 
 def mnist_mlp():
@@ -48,3 +53,9 @@ def xor_mlp(in_dim: int = 2, hidden: int = 256):
 def gaussian_mlp(in_dim: int = 10, hidden: int = 256):
     """10-in MLP (reference tests/dl_runner.py:56-63 create_random_model)."""
     return RegressorMLP([in_dim, hidden, 1])
+
+
+def dropout_mlp(in_dim: int = 10, hidden: int = 64, p: float = 0.3):
+    """gaussian_mlp + inter-layer Dropout (the reference's tf_dropout
+    surface, ml_util.py:70-71)."""
+    return DropoutRegressorMLP([in_dim, hidden, 1], p=p)

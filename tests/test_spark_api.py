@@ -364,27 +364,7 @@ def test_dropout_model_through_api(spark):
     tfDropout / toKeepDropout Params flow to transform, where dropout
     is inert for BOTH flag values (reference ml_util.py:70-71 feed
     semantics — 1.0-as-keep-prob / 0.0-as-drop-rate both disable it)."""
-    import torch
-
-    from sparkflow_amd import nn as snn
-
-    def dropout_mlp():
-        class M(torch.nn.Module):
-            def __init__(self):
-                super().__init__()
-                self.l1 = snn.Linear(10, 64, activation="relu")
-                self.drop = snn.Dropout(0.3)
-                self.l2 = snn.Linear(64, 1, activation="sigmoid")
-                self.loss_fn = snn.MSE()
-
-            def forward(self, x):
-                return self.l2(self.drop(self.l1(x)))
-
-            def compute_loss(self, out, target):
-                return self.loss_fn(out,
-                                    target.reshape(out.shape).to(out.dtype))
-
-        return M()
+    from sparkflow_amd.models import dropout_mlp
 
     df = _gaussian_df(spark, 200)
     est = _estimator(build_graph(dropout_mlp), free_port(), iters=12,
