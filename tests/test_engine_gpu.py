@@ -328,3 +328,50 @@ def test_raw_rccl_comm_init_times_out_without_peer():
         ops.hip.nccl_comm_init(uid, 2, 0, 2000)  # nobody joins rank 1
     dt = time.perf_counter() - t0
     assert dt < 30, f"timeout not bounded: {dt:.1f}s"
+
+
+def test_interleaved_eval_does_not_freeze_sampling():
+    """Regression for the round-2 done-ticket bug: allocations between
+    replay batches (an eval forward) used to reuse the gather ticket's
+    freed block and silently freeze the minibatch cursor — every replay
+    then re-trained one minibatch.  The cursor must keep advancing and
+    the eval accuracy must keep responding across chunks."""
+    import numpy as np
+
+    from sparkflow_amd.graph_utils import build_graph
+    from sparkflow_amd.models import mnist_mlp
+    from sparkflow_amd.parallel.dist_engine import build_engine
+    from sparkflow_amd.parallel.ps import OptimizerSpec, TrainConfig
+    from sparkflow_amd.utils.serialization import (
+        module_from_json,
+        unpack_params,
+    )
+
+    graph_json = build_graph(mnist_mlp)
+    cfg = TrainConfig(iters=300, mini_batch=64, mini_stochastic_iters=1,
+                      shuffle=False, seed=21)
+    engine = build_engine(graph_json, OptimizerSpec("adam", 0.003), cfg)
+    rng = np.random.default_rng(0)
+    # learnable task: class templates + noise
+    templ = rng.random((10, 784), dtype=np.float32)
+    lab = rng.integers(0, 10, 4096)
+    x = (templ[lab] + rng.normal(0, 0.3, (4096, 784))).astype(np.float32)
+    engine.worker.stage(x, lab.astype(np.float32).reshape(-1, 1))
+    em = module_from_json(graph_json).to(engine.device).eval()
+    xt = torch.as_tensor(x[:512], device=engine.device)
+    yt = torch.as_tensor(lab[:512], device=engine.device)
+
+    ctrs, accs = [], []
+    for _ in range(4):
+        engine.worker.run_steps(80, 64)
+        torch.cuda.synchronize()
+        g = engine.worker._graph_state
+        ctrs.append(int(g["ctr"].item()) if g.get("ctr") is not None
+                    else -1)
+        unpack_params(em, engine.ps.weights)
+        with torch.no_grad():
+            accs.append(float((em(xt).argmax(1) == yt).float().mean()))
+    # cursor advanced between every chunk despite the eval allocations
+    assert len(set(ctrs)) == len(ctrs), ctrs
+    assert accs[-1] > 0.9, accs
+    engine.close()
