@@ -164,6 +164,139 @@ __global__ __launch_bounds__(256) void conv_nhwc_fwd_kernel(
   }
 }
 
+// ---------------- forward, narrow tile (Cout <= 32) ----------------
+// The MNIST CNN's convs have Cout=32 (and its dgrad "Cout"=Cin=32): a
+// BN=64 tile wastes half its MFMA work there.  64x32 tile, 4 waves
+// each covering a 32x16 quadrant (2x1 16x16x4 frags).
+template <bool HAS_BIAS>
+__global__ __launch_bounds__(256) void conv_nhwc_fwd_n32_kernel(
+    const float* __restrict__ x, const float* __restrict__ wgt,
+    const float* __restrict__ bias, float* __restrict__ y, ConvShapeHW s,
+    int M, int K, int act) {
+  __shared__ float As[BM][BK + 1];
+  __shared__ float Bs[BK][32 + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = (wv & 1) * 32;   // wave rows quadrant
+  const int wn = (wv >> 1) * 16;  // wave cols half
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * 32;
+  const int Nc = s.Cout;
+
+  f32x4 acc[2] = {};
+
+  const int st_row = t >> 2, st_c4 = (t & 3) * 4;
+  const int gm = m0 + st_row;
+  const int m_dec = gm < M ? gm : 0;
+  const int a_wo = m_dec % s.Wo;
+  const int a_ho = (m_dec / s.Wo) % s.Ho;
+  const long a_nbase = (long)(m_dec / (s.Wo * s.Ho)) * s.H * s.W * s.Cin;
+  const int h0 = a_ho * s.stride - s.pad;
+  const int w0 = a_wo * s.stride - s.pad;
+  int a_ci = st_c4 % s.Cin;
+  int a_khkw = st_c4 / s.Cin;
+  const bool vec = (s.Cin & 3) == 0;
+
+  auto load_a = [&](int k0) -> float4 {
+    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gm >= M) return r;
+    if (vec) {
+      const int kh = a_khkw / s.Kw, kw = a_khkw % s.Kw;
+      const int h = h0 + kh, w = w0 + kw;
+      if (k0 + st_c4 + 3 < K && h >= 0 && h < s.H && w >= 0 && w < s.W) {
+        r = *reinterpret_cast<const float4*>(
+            &x[a_nbase + ((long)h * s.W + w) * s.Cin + a_ci]);
+      }
+      return r;
+    }
+    float* rr = reinterpret_cast<float*>(&r);
+    int ci = a_ci, khkw = a_khkw;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int kh = khkw / s.Kw, kw = khkw % s.Kw;
+      const int h = h0 + kh, w = w0 + kw;
+      if (k0 + st_c4 + i < K && h >= 0 && h < s.H && w >= 0 && w < s.W) {
+        rr[i] = x[a_nbase + ((long)h * s.W + w) * s.Cin + ci];
+      }
+      if (++ci == s.Cin) { ci = 0; ++khkw; }
+    }
+    return r;
+  };
+  auto advance_a = [&]() {
+    a_ci += BK;
+    while (a_ci >= s.Cin) { a_ci -= s.Cin; ++a_khkw; }
+  };
+  // B tile is 16k x 32co: threads with st_row < 32 stage it
+  auto load_b = [&](int k0) -> float4 {
+    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
+    const int gn = n0 + st_row;
+    if (st_row >= 32) return r;
+    if (gn < Nc && k0 + BK <= K) {
+      r = *reinterpret_cast<const float4*>(&wgt[(long)gn * K + k0 + st_c4]);
+    } else {
+      float* rr = reinterpret_cast<float*>(&r);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int gk = k0 + st_c4 + i;
+        rr[i] = (gn < Nc && gk < K) ? wgt[(long)gn * K + gk] : 0.0f;
+      }
+    }
+    return r;
+  };
+
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  float4 ra = load_a(0);
+  float4 rb = load_b(0);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    {
+      const float* pa = reinterpret_cast<const float*>(&ra);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) As[st_row][st_c4 + i] = pa[i];
+      if (st_row < 32) {
+        const float* pb = reinterpret_cast<const float*>(&rb);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) Bs[st_c4 + i][st_row] = pb[i];
+      }
+    }
+    __syncthreads();
+    if (k0 + BK < K) {
+      advance_a();
+      ra = load_a(k0 + BK);
+      rb = load_b(k0 + BK);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      const float a0 = As[wm + fr][kk * 4 + fk];
+      const float a1 = As[wm + 16 + fr][kk * 4 + fk];
+      const float b0 = Bs[kk * 4 + fk][wn + fr];
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0], 0, 0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+  const int co = n0 + wn + fcol;
+  if (co < Nc) {
+    const float bv = HAS_BIAS ? bias[co] : 0.0f;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int m = m0 + wm + mi * 16 + frow + i;
+        if (m < M) {
+          y[(long)m * Nc + co] = act_apply(acc[mi][i] + bv, act);
+        }
+      }
+    }
+  }
+}
+
 // ---------------- forward, large-M tile ----------------
 // ResNet-shape forward convs are LARGE-M (M = N*Ho*Wo is 10^5..10^6)
 // with modest Cout (64..512): the dense-GEMM 64x64/BK16 tile leaves
@@ -1051,6 +1184,19 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w,
                          bp, y.data_ptr<float>(), s, M, K, (int)act);
     } else {
       hipLaunchKernelGGL((conv_nhwc_fwd_big_kernel<false>), grid, dim3(256),
+                         0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                         nullptr, y.data_ptr<float>(), s, M, K, (int)act);
+    }
+    return y;
+  }
+  if (s.Cout <= 32) {  // narrow tile: BN=64 would waste half its MFMAs
+    dim3 grid(ceil_div(s.Cout, 32), ceil_div(M, BM));
+    if (bp) {
+      hipLaunchKernelGGL((conv_nhwc_fwd_n32_kernel<true>), grid, dim3(256),
+                         0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                         bp, y.data_ptr<float>(), s, M, K, (int)act);
+    } else {
+      hipLaunchKernelGGL((conv_nhwc_fwd_n32_kernel<false>), grid, dim3(256),
                          0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
                          nullptr, y.data_ptr<float>(), s, M, K, (int)act);
     }
