@@ -773,6 +773,165 @@ __global__ __launch_bounds__(256) void conv_nhwc_wgrad_sq_kernel(
   }
 }
 
+// ---------------- wgrad, narrow tile (Cout <= 32) ----------------
+// dw for the MNIST CNN's 32-channel convs: BM=64 wastes half the MFMA
+// work when M = Cout <= 32.  32x64 tile; 4 waves each 16 co x 32 k'
+// (1x2 16x16x4 frags).
+template <bool SPLITK>
+__global__ __launch_bounds__(256) void conv_nhwc_wgrad_m32_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    float* __restrict__ dw, ConvShapeHW s, int M, int Nc, int K,
+    int k_chunk) {
+  __shared__ float As[32][BK + 1];
+  __shared__ float Bs[BK][BN + 4];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wv = t >> 6;
+  const int wm = (wv & 1) * 16;   // wave's co half
+  const int wn = (wv >> 1) * 32;  // wave's k' quadrant
+  const int m0 = blockIdx.y * 32;  // co tile
+  const int n0 = blockIdx.x * BN;  // k' tile
+
+  f32x4 acc[2] = {};
+
+  int k_begin = 0, k_end = K;
+  if (SPLITK) {
+    k_begin = blockIdx.z * k_chunk;
+    k_end = min(K, k_begin + k_chunk);
+  }
+
+  // A: As[co][p] = dy[p][co] — threads with (t&15)*4 < 32 stage it
+  const int a_kk = t >> 4, a_m4 = (t & 15) * 4;
+  const int b_kk = t >> 4, b_n4 = (t & 15) * 4;
+  const int HoWo = s.Ho * s.Wo;
+
+  int pa_n = (k_begin + a_kk) / HoWo;
+  int pa_hw = (k_begin + a_kk) % HoWo;
+  int pb_n = (k_begin + b_kk) / HoWo;
+  int pb_ho = ((k_begin + b_kk) % HoWo) / s.Wo;
+  int pb_wo = ((k_begin + b_kk) % HoWo) % s.Wo;
+  int kw_i[4], kh_i[4], ci_i[4];
+  bool gk_ok[4];
+  bool bvec = (s.Cin & 3) == 0;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int gk = n0 + b_n4 + i;
+    gk_ok[i] = gk < Nc;
+    const int g = gk_ok[i] ? gk : 0;
+    ci_i[i] = g % s.Cin;
+    const int khkw = g / s.Cin;
+    kw_i[i] = khkw % s.Kw;
+    kh_i[i] = khkw / s.Kw;
+  }
+  bvec = bvec && gk_ok[3] && kh_i[0] == kh_i[3] && kw_i[0] == kw_i[3];
+  const long n_stride = (long)s.H * s.W * s.Cin;
+
+  auto load_a = [&](int gp0) -> float4 {
+    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gp0 >= k_end || a_m4 >= 32) return r;
+    int n = pa_n, hw = pa_hw;
+    if ((s.Cout & 3) == 0 && m0 + a_m4 + 3 < s.Cout) {
+      r = *reinterpret_cast<const float4*>(
+          &dy[((long)n * HoWo + hw) * s.Cout + m0 + a_m4]);
+    } else {
+      float* rr = reinterpret_cast<float*>(&r);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int co = m0 + a_m4 + i;
+        rr[i] = co < s.Cout
+                    ? dy[((long)n * HoWo + hw) * s.Cout + co] : 0.0f;
+      }
+    }
+    return r;
+  };
+  auto load_b = [&](int gp0) -> float4 {
+    float4 r = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (gp0 >= k_end) return r;
+    const long nbase = (long)pb_n * n_stride;
+    if (bvec) {
+      const int h = pb_ho * s.stride - s.pad + kh_i[0];
+      const int w = pb_wo * s.stride - s.pad + kw_i[0];
+      if (h >= 0 && h < s.H && w >= 0 && w < s.W) {
+        r = *reinterpret_cast<const float4*>(
+            &x[nbase + ((long)h * s.W + w) * s.Cin + ci_i[0]]);
+      }
+      return r;
+    }
+    float* rr = reinterpret_cast<float*>(&r);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      if (gk_ok[i]) {
+        const int h = pb_ho * s.stride - s.pad + kh_i[i];
+        const int w = pb_wo * s.stride - s.pad + kw_i[i];
+        if (h >= 0 && h < s.H && w >= 0 && w < s.W)
+          rr[i] = x[nbase + ((long)h * s.W + w) * s.Cin + ci_i[i]];
+      }
+    }
+    return r;
+  };
+  auto advance = [&]() {
+    pa_hw += BK;
+    while (pa_hw >= HoWo) { pa_hw -= HoWo; ++pa_n; }
+    pb_wo += BK;
+    while (pb_wo >= s.Wo) { pb_wo -= s.Wo; ++pb_ho; }
+    while (pb_ho >= s.Ho) { pb_ho -= s.Ho; ++pb_n; }
+  };
+
+  const int fr = lane & 15;
+  const int fk = lane >> 4;
+
+  float4 ra = load_a(k_begin + a_kk);
+  float4 rb = load_b(k_begin + b_kk);
+  for (int k0 = k_begin; k0 < k_end; k0 += BK) {
+    {
+      if (a_m4 < 32) {
+        const float* pa = reinterpret_cast<const float*>(&ra);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) As[a_m4 + i][a_kk] = pa[i];
+      }
+      const float* pb = reinterpret_cast<const float*>(&rb);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) Bs[b_kk][b_n4 + i] = pb[i];
+    }
+    __syncthreads();
+    if (k0 + BK < k_end) {
+      advance();
+      ra = load_a(k0 + BK + a_kk);
+      rb = load_b(k0 + BK + b_kk);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      const float a0 = As[wm + fr][kk * 4 + fk];
+      const float b0 = Bs[kk * 4 + fk][wn + fr];
+      const float b1 = Bs[kk * 4 + fk][wn + 16 + fr];
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0], 0, 0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+#pragma unroll
+  for (int ni = 0; ni < 2; ++ni) {
+    const int kcol = n0 + wn + ni * 16 + fcol;
+    if (kcol >= Nc) continue;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int co = m0 + wm + frow + i;
+      if (co < M) {
+        if (SPLITK) {
+          dw[(long)blockIdx.z * M * Nc + (long)co * Nc + kcol] =
+              acc[ni][i];
+        } else {
+          dw[(long)co * Nc + kcol] = acc[ni][i];
+        }
+      }
+    }
+  }
+}
+
 // ---------------- wgrad ----------------
 // dw[co][k'] = sum_p dy[p, co] * im2col(x)[p, k'] — A staged transposed
 // (dy contiguous along co), B's im2col contiguous along ci.
@@ -1267,6 +1426,40 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor x, torch::Tensor dy, int64_t Kh,
     } else {
       dim3 grid(ceil_div(Nc, BN3), ceil_div(M, BM3));
       hipLaunchKernelGGL((conv_nhwc_wgrad_sq_kernel<false>), grid,
+                         dim3(256), 0, stream, x.data_ptr<float>(),
+                         dy.data_ptr<float>(), dw.data_ptr<float>(), s, M,
+                         Nc, K, K);
+    }
+    return dw;
+  }
+  // narrow-M tile for Cout <= 32 (same rationale as the fwd n32 tile)
+  if (M <= 32) {
+    const int base32 = ceil_div(Nc, BN) * ceil_div(M, 32);
+    int splitk = 1;
+    const int max_by_k = std::max(1, K / 256);
+    const int cap = sf_splitk_env_max();
+    while (splitk < 256 && splitk < cap && base32 * splitk <= 1024 &&
+           splitk * 2 <= max_by_k) {
+      splitk *= 2;
+    }
+    while (base32 * splitk < 512 && splitk < cap &&
+           splitk * 2 <= max_by_k && splitk < 1024) {
+      splitk *= 2;
+    }
+    if (splitk > 1) {
+      const int k_chunk = ceil_div(K, splitk * BK) * BK;
+      auto ws = torch::empty({splitk, M, Nc}, x.options());
+      dim3 grid(ceil_div(Nc, BN), ceil_div(M, 32), splitk);
+      hipLaunchKernelGGL((conv_nhwc_wgrad_m32_kernel<true>), grid,
+                         dim3(256), 0, stream, x.data_ptr<float>(),
+                         dy.data_ptr<float>(), ws.data_ptr<float>(), s, M,
+                         Nc, K, k_chunk);
+      const long n = (long)M * Nc;
+      launch_conv_splitk_reduce(ws.data_ptr<float>(), dw.data_ptr<float>(),
+                                n, splitk, stream);
+    } else {
+      dim3 grid(ceil_div(Nc, BN), ceil_div(M, 32));
+      hipLaunchKernelGGL((conv_nhwc_wgrad_m32_kernel<false>), grid,
                          dim3(256), 0, stream, x.data_ptr<float>(),
                          dy.data_ptr<float>(), dw.data_ptr<float>(), s, M,
                          Nc, K, K);
