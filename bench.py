@@ -28,8 +28,8 @@ import torch
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=200)
-    ap.add_argument("--warmup", type=int, default=50)
+    ap.add_argument("--steps", type=int, default=500)
+    ap.add_argument("--warmup", type=int, default=100)
     ap.add_argument("--mbs", type=int, default=300,
                     help="minibatch per step (reference config: 300)")
     ap.add_argument("--model", default="mnist_mlp",
