@@ -1516,6 +1516,9 @@ torch::Tensor conv_nhwc_dgrad_strided(torch::Tensor dy, torch::Tensor wt,
   s.H = H; s.W = W; s.stride = stride; s.pad = pad;
   const int S = stride;
   const int pprime = s.Kh - 1 - pad;
+  // the NHWC conv family takes one kernel size / one pad (square);
+  // the class decomposition below relies on it
+  TORCH_CHECK(s.Kh == s.Kw, "strided dgrad expects square kernels");
   // any class with zero taps leaves its dx pixels unwritten -> zeros
   bool any_empty = false;
   for (int a = 0; a < S && !any_empty; ++a) {
