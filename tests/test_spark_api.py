@@ -375,3 +375,28 @@ def test_dropout_model_through_api(spark):
     assert p1 == p2  # dropout inert at inference (deterministic)
     rows = fitted.transform(df).take(10)
     assert _calculate_errors(rows) < len(rows)
+
+
+def test_dist_fit_falls_back_on_thread_executors(monkeypatch, spark):
+    """SPARKFLOW_FIT_TRANSPORT=dist on a thread-executor (local[N])
+    session cannot host one torch.distributed rank per partition in one
+    process — eligibility must say no and fit must ride TCP."""
+    monkeypatch.setenv("SPARKFLOW_FIT_TRANSPORT", "dist")
+    from sparkflow_amd.localspark import using_localspark
+    from sparkflow_amd.parallel.hogwild import HogwildSparkModel
+
+    if not using_localspark():
+        pytest.skip("thread executors are the localspark local[N] mode")
+    df = _gaussian_df(spark, 80)
+    called = {"dist": False}
+    orig = HogwildSparkModel._train_rdd_dist
+
+    def spy(self, rdd):
+        called["dist"] = True
+        return orig(self, rdd)
+
+    monkeypatch.setattr(HogwildSparkModel, "_train_rdd_dist", spy)
+    model = _estimator(build_graph(gaussian_mlp), free_port(),
+                       iters=4).fit(df)
+    assert not called["dist"]  # TCP fallback took it
+    assert model.getOrDefault(model.modelWeights)
