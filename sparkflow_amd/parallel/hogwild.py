@@ -156,7 +156,14 @@ class HogwildSparkModel:
         1..N-1 pull/push over their own RCCL pair link — the reference's
         worker<->PS HTTP hot path (HogwildSparkModel.py:22-35,57-58)
         replaced by the xGMI star, reachable straight from
-        SparkAsyncDL.fit.  Returns the final weights ndarray list."""
+        SparkAsyncDL.fit.  Returns the final weights ndarray list.
+
+        Deployment constraint: the rendezvous address is the DRIVER
+        host, so rank 0's executor must be reachable there — true for
+        the target deployment (one 8-GPU node, executors colocated with
+        the driver) and for localspark's ``localproc``.  A multi-HOST
+        executor set should use the TCP path
+        (SPARKFLOW_FIT_TRANSPORT=tcp)."""
         from sparkflow_amd.parallel.transport import find_free_port
 
         world = rdd.getNumPartitions()
