@@ -354,6 +354,7 @@ def test_batchnorm2d_gpu(relu):
     (2, 9, 9, 8, 12, 1, 2, 0),    # 1x1 downsample
     (12, 56, 56, 8, 64, 3, 1, 1),  # large-M shape
     (24, 28, 28, 16, 128, 3, 1, 1),  # Cout>=128: 128x128 mfma_32x32x2
+    (24, 28, 28, 128, 32, 3, 1, 1),  # dgrad hits the sq tile (Cin=128)
 ])
 def test_conv2d_nhwc_autograd_gpu(conf):
     """NHWC conv fwd/bwd (channel-contiguous kernels) vs the CPU permute
