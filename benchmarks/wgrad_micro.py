@@ -2,7 +2,9 @@ import json, os, sys, time
 sys.path.insert(0, "/root/repo")
 import torch
 from sparkflow_amd import ops
-SHAPES = [("l2_3x3", 28, 28, 128, 128, 3, 1, 1),
+SHAPES = [("l1_3x3", 56, 56, 64, 64, 3, 1, 1),
+          ("l2_in_s2", 56, 56, 64, 128, 3, 2, 1),
+          ("l2_3x3", 28, 28, 128, 128, 3, 1, 1),
           ("l3_3x3", 14, 14, 256, 256, 3, 1, 1),
           ("l4_3x3", 7, 7, 512, 512, 3, 1, 1)]
 mode = os.environ.get("SPARKFLOW_CONV_SQ", "1")
