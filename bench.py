@@ -126,10 +126,10 @@ def main():
                 return time.perf_counter() - c0
 
             best = _probe()
-            for _ in range(4):
+            for _ in range(8):
                 _ops.hip.replay_streams_reroll()
                 t = _probe()
-                ok = t <= best * 1.2
+                ok = t <= best * 1.1
                 best = min(best, t)
                 if ok:
                     break
